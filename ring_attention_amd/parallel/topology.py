@@ -1,0 +1,83 @@
+"""Ring topology math: rank/world helpers and sub-ring neighbor arithmetic.
+
+Capability parity with the reference's distributed helpers and neighbor math
+(/root/reference/ring_attention_pytorch/distributed.py:31-41,
+ /root/reference/ring_attention_pytorch/ring.py:35-47), re-designed for an
+MI355X node: world is one process per GPU over RCCL ("nccl" backend on ROCm);
+``ring_size < world_size`` splits the world into independent sub-rings so one
+node can run several data-parallel rings over xGMI simultaneously.
+"""
+
+from __future__ import annotations
+
+import functools
+
+import torch.distributed as dist
+
+
+def is_distributed() -> bool:
+    return dist.is_initialized() and dist.get_world_size() > 1
+
+
+@functools.lru_cache(maxsize=None)
+def _cached_rank_world() -> tuple[int, int]:
+    if not dist.is_initialized():
+        return 0, 1
+    return dist.get_rank(), dist.get_world_size()
+
+
+def get_rank() -> int:
+    # NOTE: not lru_cached on its own so tests that tear down/re-init process
+    # groups of different sizes in one process stay correct.
+    if not dist.is_initialized():
+        return 0
+    return dist.get_rank()
+
+
+def get_world_size() -> int:
+    if not dist.is_initialized():
+        return 1
+    return dist.get_world_size()
+
+
+class RingTopology:
+    """Neighbor math for a (sub-)ring.
+
+    The world of size W is partitioned into ``W // ring_size`` independent
+    rings of ``ring_size`` consecutive ranks; ring ``i`` owns global ranks
+    ``[i*ring_size, (i+1)*ring_size)``.  ``ring_rank`` is the position within
+    the ring.  With ``ring_size == world_size`` this is the plain full ring.
+    """
+
+    def __init__(self, ring_size: int | None = None, rank: int | None = None, world_size: int | None = None):
+        world = world_size if world_size is not None else get_world_size()
+        rank = rank if rank is not None else get_rank()
+        ring_size = ring_size if ring_size is not None else world
+        assert world % ring_size == 0, f"world size {world} not divisible by ring size {ring_size}"
+        self.world_size = world
+        self.rank = rank
+        self.ring_size = ring_size
+        self.ring_index = rank // ring_size          # which sub-ring this rank belongs to
+        self.ring_rank = rank % ring_size            # position within the sub-ring
+        self.ring_base = self.ring_index * ring_size  # global rank of ring position 0
+
+    def global_rank_of(self, ring_rank: int) -> int:
+        return self.ring_base + (ring_rank % self.ring_size)
+
+    @property
+    def right(self) -> int:
+        """Global rank of the next rank around the ring (receives what we send)."""
+        return self.global_rank_of(self.ring_rank + 1)
+
+    @property
+    def left(self) -> int:
+        """Global rank of the previous rank around the ring (sends what we receive)."""
+        return self.global_rank_of(self.ring_rank - 1)
+
+    def source_of_hop(self, hop: int) -> int:
+        """Ring rank whose original shard this rank holds after ``hop`` passes.
+
+        Shards travel rightward, so after ``hop`` passes rank r holds the
+        shard that originated at ring rank ``r - hop``.
+        """
+        return (self.ring_rank - hop) % self.ring_size
