@@ -35,57 +35,81 @@ from torch.autograd import Function
 from ..parallel import RingAccumulator, RingTopology, all_ring_pass, is_distributed
 from .reference import MASK_VALUE, softclamp
 
-# bucket-pair mask modes
-SKIP = 0
-FULL = 1
-TRI_INCL = 2   # mask cols j > i           (diagonal attends)
-TRI_STRICT = 3  # mask cols j >= i          (strict: diagonal masked)
+# bucket-pair mask decisions
+SKIP = "skip"
+FULL = "full"
+
+
+def bucket_positions(b: int, bucket_size: int, rank: int, shard_len: int,
+                     ring_size: int, striped: bool, device) -> Tensor:
+    """Global token positions of bucket ``b`` of the shard held by ring ``rank``."""
+    local = b * bucket_size + torch.arange(bucket_size, device=device)
+    if striped:
+        return local * ring_size + rank
+    return rank * shard_len + local
 
 
 def bucket_mode(
-    bi: int, bj: int, rq: int, rk: int, nb: int,
-    causal: bool, striped: bool, lookback_buckets: int | None,
-) -> int:
-    """Mask mode for (q bucket bi on ring rank rq) x (kv bucket bj from ring rank rk)."""
-    if not causal:
+    bi: int, bj: int, rq: int, rk: int, *,
+    bucket_size: int, shard_len: int, ring_size: int,
+    causal: bool, striped: bool, lookback: int | None, device,
+):
+    """Mask decision for (q bucket bi on ring rank rq) x (kv bucket bj of rank rk).
+
+    Returns SKIP, FULL, or a boolean (bk, bk) tensor (True = masked).
+    Causality and lookback (exact token-level sliding window) are evaluated
+    on GLOBAL positions, so the result is identical for any bucket size and
+    for contiguous vs striped layouts.
+    """
+    if not causal and lookback is None:
         return FULL
-    if striped:
-        if bj > bi:
-            return SKIP
-        if bj == bi:
-            return TRI_INCL if rk <= rq else TRI_STRICT
+
+    def lo_hi(b, r):
+        if striped:
+            lo = b * bucket_size * ring_size + r
+            hi = (b * bucket_size + bucket_size - 1) * ring_size + r
+        else:
+            lo = r * shard_len + b * bucket_size
+            hi = lo + bucket_size - 1
+        return lo, hi
+
+    row_lo, row_hi = lo_hi(bi, rq)
+    col_lo, col_hi = lo_hi(bj, rk)
+
+    if causal and col_lo > row_hi:
+        return SKIP
+    if lookback is not None and (row_lo - col_hi) > lookback:
+        return SKIP
+    full = (not causal or col_hi <= row_lo) and \
+        (lookback is None or (row_hi - col_lo) <= lookback)
+    if full:
         return FULL
-    # contiguous layout: global bucket indices
-    rgb = rq * nb + bi
-    cgb = rk * nb + bj
-    if cgb > rgb:
-        return SKIP
-    if lookback_buckets is not None and (rgb - cgb) > lookback_buckets:
-        return SKIP
-    if cgb == rgb:
-        return TRI_INCL
-    return FULL
+
+    rp = bucket_positions(bi, bucket_size, rq, shard_len, ring_size, striped, device)
+    cp = bucket_positions(bj, bucket_size, rk, shard_len, ring_size, striped, device)
+    masked = torch.zeros(bucket_size, bucket_size, dtype=torch.bool, device=device)
+    if causal:
+        masked |= cp[None, :] > rp[:, None]
+    if lookback is not None:
+        masked |= (rp[:, None] - cp[None, :]) > lookback
+    return masked
 
 
 def max_hops_for_lookback(
-    causal: bool, striped: bool, lookback_buckets: int | None, nb: int, ring_size: int
+    causal: bool, striped: bool, lookback: int | None, shard_len: int, ring_size: int
 ) -> int:
     """Uniform number of ring passes actually needed (lookback truncates the walk)."""
-    if striped or not causal or lookback_buckets is None:
+    if striped or not causal or lookback is None:
         return ring_size
-    # hop t reaches shards (t-1)*nb + 1 buckets back at minimum distance
-    return min(ring_size, (max(lookback_buckets - 1, 0)) // nb + 2)
+    # hop t's nearest (row, col) pair is (t-1)*shard_len + 1 tokens apart
+    return min(ring_size, max(lookback - 1, 0) // shard_len + 2)
 
 
-def _apply_bucket_masks(sim: Tensor, mode: int, key_mask_bucket: Tensor | None) -> Tensor:
+def _apply_bucket_masks(sim: Tensor, mode, key_mask_bucket: Tensor | None) -> Tensor:
     if key_mask_bucket is not None:
         sim = sim.masked_fill(~key_mask_bucket[:, None, None, :], MASK_VALUE)
-    if mode in (TRI_INCL, TRI_STRICT):
-        i = sim.shape[-2]
-        j = sim.shape[-1]
-        tri = torch.ones(i, j, device=sim.device, dtype=torch.bool)
-        tri = tri.triu(1 if mode == TRI_INCL else 0)
-        sim = sim.masked_fill(tri[None, None, :, :], MASK_VALUE)
+    if isinstance(mode, Tensor):
+        sim = sim.masked_fill(mode[None, None, :, :], MASK_VALUE)
     return sim
 
 
@@ -113,6 +137,9 @@ class RingFlashAttentionFunction(Function):
         assert h % hk == 0, "query heads must be a multiple of kv heads"
         groups = h // hk
         bucket_size = min(bucket_size, n)
+        if n % bucket_size != 0:
+            # non-ring local use with awkward lengths: fall back to a divisor
+            bucket_size = math.gcd(bucket_size, n)
         assert n % bucket_size == 0, f"seq {n} not divisible by bucket size {bucket_size}"
         nb = n // bucket_size
         scale = d ** -0.5
@@ -122,11 +149,10 @@ class RingFlashAttentionFunction(Function):
                             rank=None if use_ring else 0,
                             world_size=None if use_ring else 1)
 
-        lookback_buckets = None
-        if max_lookback_seq_len is not None:
-            assert causal and not striped, "lookback requires causal contiguous layout"
-            lookback_buckets = max_lookback_seq_len // bucket_size
-        hops = max_hops_for_lookback(causal, striped, lookback_buckets, nb, topo.ring_size)
+        lookback = max_lookback_seq_len
+        if lookback is not None:
+            assert causal, "lookback (sliding window) requires causal"
+        hops = max_hops_for_lookback(causal, striped, lookback, n, topo.ring_size)
 
         qf = q.float()
         o = torch.zeros((b, n, h, d), device=q.device, dtype=torch.float32)
@@ -151,8 +177,11 @@ class RingFlashAttentionFunction(Function):
                 vj = v_t[:, bj * bucket_size:(bj + 1) * bucket_size]
                 mj = mask_t[:, bj * bucket_size:(bj + 1) * bucket_size] if mask_t is not None else None
                 for bi in range(nb):
-                    mode = bucket_mode(bi, bj, rq, rk, nb, causal, striped, lookback_buckets)
-                    if mode == SKIP:
+                    mode = bucket_mode(
+                        bi, bj, rq, rk, bucket_size=bucket_size, shard_len=n,
+                        ring_size=topo.ring_size, causal=causal, striped=striped,
+                        lookback=lookback, device=q.device)
+                    if mode is SKIP:
                         continue
                     sl = slice(bi * bucket_size, (bi + 1) * bucket_size)
                     qi = qf[:, sl]
@@ -176,14 +205,14 @@ class RingFlashAttentionFunction(Function):
 
         ctx.save_for_backward(q, k, v, o, lse,
                               mask.to(torch.uint8) if mask is not None else torch.empty(0))
-        ctx.params = (causal, bucket_size, striped, lookback_buckets, hops,
+        ctx.params = (causal, bucket_size, striped, lookback, hops,
                       softclamp_qk_sim, softclamp_value, use_ring, topo.ring_size, groups)
         return o.to(q.dtype), lse
 
     @staticmethod
     def backward(ctx, do: Tensor, _dlse):
         q, k, v, o, lse, mask_u8 = ctx.saved_tensors
-        (causal, bucket_size, striped, lookback_buckets, hops,
+        (causal, bucket_size, striped, lookback, hops,
          softclamp_qk_sim, softclamp_value, use_ring, ring_size, groups) = ctx.params
         mask = mask_u8.bool() if mask_u8.numel() else None
 
@@ -223,8 +252,11 @@ class RingFlashAttentionFunction(Function):
                 kj, vj = k_t[:, slj], v_t[:, slj]
                 mj = mask_t[:, slj] if mask_t is not None else None
                 for bi in range(nb):
-                    mode = bucket_mode(bi, bj, rq, rk, nb, causal, striped, lookback_buckets)
-                    if mode == SKIP:
+                    mode = bucket_mode(
+                        bi, bj, rq, rk, bucket_size=bucket_size, shard_len=n,
+                        ring_size=topo.ring_size, causal=causal, striped=striped,
+                        lookback=lookback, device=q.device)
+                    if mode is SKIP:
                         continue
                     sli = slice(bi * bucket_size, (bi + 1) * bucket_size)
                     qi, doi = qf[:, sli], dof[:, sli]

@@ -69,7 +69,15 @@ def _pad_spec(ndim: int, dim: int, amount: int) -> list[int]:
 
 
 class AllGatherFunction(Function):
-    """Autograd all-gather along ``dim``; backward takes this rank's grad slice."""
+    """Autograd all-gather along ``dim``.
+
+    Backward is the mathematically correct distributed adjoint: a
+    reduce-scatter (sum every rank's gradient for the gathered tensor, keep
+    this rank's slice).  The reference instead took the LOCAL grad's own
+    slice with no reduction (distributed.py:103-107), dropping every other
+    rank's contribution to this rank's input — only tolerable under DDP
+    parameter averaging, wrong for input gradients.
+    """
 
     @staticmethod
     def forward(ctx, x: Tensor, dim: int, sizes: Tensor | None):
@@ -87,6 +95,9 @@ class AllGatherFunction(Function):
     def backward(ctx, grad: Tensor, _grad_sizes):
         rank = get_rank()
         splits = ctx.sizes.tolist()
+        if is_distributed():
+            grad = grad.contiguous()
+            dist.all_reduce(grad)   # reduce-scatter expressed as all-reduce + slice
         grads = grad.split(splits, dim=ctx.dim)
         return grads[rank].contiguous(), None, None
 

@@ -1,0 +1,116 @@
+"""Zig-zag context parallelism (the Llama-3 scheme, arXiv:2407.21783).
+
+Capability parity with the reference's zig_zag_attention
+(/root/reference/ring_attention_pytorch/zig_zag_attention.py:35-140): pad the
+sequence to a multiple of 2*world; rank r owns chunks r and 2W-1-r (causal
+load balance); K/V are all-gathered over the sequence (one RCCL all-gather
+that stripes across all 7 xGMI links — bandwidth-optimal on one MI355X node,
+unlike a ring which is single-link-bound) and attention runs with an explicit
+mask derived from the exported positions.
+
+Layout parity with the reference: (batch, heads, seq, dim) — "b h n d".
+"""
+
+from __future__ import annotations
+
+import math
+from collections import namedtuple
+
+import torch
+import torch.nn.functional as F
+from torch import Tensor
+
+from .ops.reference import MASK_VALUE
+from .parallel import AllGather, get_rank, get_world_size
+
+ShardOutput = namedtuple("ShardOutput", [
+    "local_sequence",
+    "query_positions",
+    "key_value_positions",
+])
+
+
+def zig_zag_pad_seq(t: Tensor):
+    """Pad dim -2 to a multiple of 2*world; returns (padded, inverse_fn)."""
+    seq_len = t.shape[-2]
+    chunks = 2 * get_world_size()
+    padded = math.ceil(seq_len / chunks) * chunks
+    t = F.pad(t, (0, 0, 0, padded - seq_len), value=0.0)
+
+    def inverse(out: Tensor) -> Tensor:
+        return out[..., :seq_len, :]
+
+    return t, inverse
+
+
+def zig_zag_shard(t: Tensor, all_gather_batch: bool = False):
+    """Rank r keeps chunks (r, 2W-1-r); exports global positions for rotary/masking.
+
+    Returns (ShardOutput(local_sequence, query_positions, key_value_positions),
+    inverse_fn)."""
+    device, seq_len = t.device, t.shape[-2]
+    rank, world = get_rank(), get_world_size()
+
+    gather_sizes = None
+    if all_gather_batch:
+        t, gather_sizes = AllGather(dim=0)(t)
+
+    chunks = 2 * world
+    chunk_size = seq_len // chunks
+    pieces = t.chunk(chunks, dim=-2)
+    local = torch.cat((pieces[rank], pieces[chunks - 1 - rank]), dim=-2).contiguous()
+
+    pos = torch.arange(seq_len, device=device).view(chunks, chunk_size)
+    first, second = pos.chunk(2, dim=0)
+    paired = torch.stack((first, second.flip(dims=(0,))), dim=1)  # (world, 2, chunk)
+
+    q_indices = paired[rank].reshape(-1)
+    kv_indices = paired.reshape(-1)
+
+    def inverse(two_chunks: Tensor) -> Tensor:
+        tc = two_chunks.reshape(*two_chunks.shape[:-2], 2,
+                                two_chunks.shape[-2] // 2, two_chunks.shape[-1])
+        all_chunks, _ = AllGather(dim=-3)(tc)               # (b, 2W, chunk, d) pair-ordered
+        shape = all_chunks.shape
+        all_chunks = all_chunks.reshape(*shape[:-3], shape[-3] // 2, 2, *shape[-2:])
+        first_half = all_chunks[..., 0, :, :]
+        second_half = all_chunks[..., 1, :, :].flip(dims=(-3,))
+        out = torch.cat((first_half, second_half), dim=-3)
+        out = out.reshape(*out.shape[:-3], -1, out.shape[-1])
+        if all_gather_batch:
+            out = out.split(gather_sizes.tolist(), dim=0)
+            out = out[rank]
+        return out
+
+    return ShardOutput(local, q_indices, kv_indices), inverse
+
+
+def zig_zag_attn(
+    q: Tensor,                      # (b, h, i, d)
+    k: Tensor,                      # (b, hk, j, d)  local shard
+    v: Tensor,                      # (b, hk, j, dv)
+    dropout: float = 0.0,
+    attn_mask: Tensor | None = None,  # bool, True = attend
+) -> Tensor:
+    heads, kv_heads = q.shape[1], k.shape[1]
+    assert heads % kv_heads == 0
+    groups = heads // kv_heads
+
+    gather_seq = AllGather(dim=-2)
+    k, _ = gather_seq(k)
+    v, _ = gather_seq(v)
+    if groups > 1:
+        # repeat pattern parity with the reference: 'b h n d -> b (g h) n d'
+        k = k.repeat(1, groups, 1, 1)
+        v = v.repeat(1, groups, 1, 1)
+
+    dtype = q.dtype
+    scale = q.shape[-1] ** -0.5
+    sim = torch.einsum("bhid,bhjd->bhij", q.float(), k.float()) * scale
+    if attn_mask is not None:
+        sim = sim.masked_fill(~attn_mask, MASK_VALUE)
+    attn = sim.softmax(dim=-1)
+    if dropout > 0.0:
+        attn = F.dropout(attn, p=dropout)
+    out = torch.einsum("bhij,bhjd->bhid", attn, v.float())
+    return out.to(dtype)

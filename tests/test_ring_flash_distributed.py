@@ -58,10 +58,9 @@ def _ring_case(rank, world, causal, striped, groups, mask_on, bucket_size, lookb
     v2 = v.detach().requires_grad_(True)
     lookback_mask = None
     if lookback is not None:
-        # bucket-quantized lookback: global bucket distance > lookback_buckets masked
-        lb_b = lookback // bucket_size
-        gb = torch.arange(n_total) // bucket_size
-        lookback_mask = (gb[:, None] - gb[None, :]) > lb_b
+        # exact token-level sliding window: distance > lookback masked
+        pos_ = torch.arange(n_total)
+        lookback_mask = (pos_[:, None] - pos_[None, :]) > lookback
     ref = default_attention(q2, k2, v2, mask=mask, causal=causal)
     if lookback is not None:
         # redo with lookback folded into sim via positions trick: do it manually
