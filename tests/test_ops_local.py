@@ -53,8 +53,8 @@ def test_flash_gqa(groups):
 def test_flash_softclamp():
     torch.manual_seed(2)
     b, n, h, d = 1, 32, 2, 16
-    q = torch.randn(b, n, h, d, requires_grad=True) * 3
-    k = torch.randn(b, n, h, d, requires_grad=True) * 3
+    q = (torch.randn(b, n, h, d) * 3).requires_grad_(True)
+    k = (torch.randn(b, n, h, d) * 3).requires_grad_(True)
     v = torch.randn(b, n, h, d, requires_grad=True)
     out = ring_flash_attn(q, k, v, causal=True, bucket_size=8,
                           softclamp_qk_sim=True, softclamp_value=5.0)
