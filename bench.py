@@ -1,0 +1,139 @@
+"""Flagship benchmark: ring flash attention fwd+bwd on MI355X.
+
+Measures BASELINE.json's headline metric — attention TFLOP/s, d_head=64,
+fwd+bwd — on BASELINE config #2 (non-causal ring flash attention, 8 heads,
+d_head 64, 8k tokens per GPU => 64k total at ring_size 8).  Weak scaling:
+per-GPU sequence shard is fixed as N grows.
+
+Single process:   python bench.py --gpus 1 --steps 20 --warmup 5
+Multi-GPU (driver): torchrun --nproc-per-node N bench.py --gpus N ...
+  (one rank per GPU over RCCL; reads RANK/LOCAL_RANK/WORLD_SIZE/MASTER_*)
+
+FLOP convention: fwd = 4 * n_shard * n_total * d * h * b per
+GPU (halved for causal); fwd+bwd = 2.5 x fwd.  Values reported are the
+WHOLE-JOB aggregate over all N GPUs.
+"""
+
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import sys
+import time
+
+import torch
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=20)
+    ap.add_argument("--warmup", type=int, default=5)
+    ap.add_argument("--seq-per-gpu", type=int, default=8192)
+    ap.add_argument("--heads", type=int, default=8)
+    ap.add_argument("--d-head", type=int, default=64)
+    ap.add_argument("--batch", type=int, default=1)
+    ap.add_argument("--causal", action="store_true")
+    ap.add_argument("--striped", action="store_true")
+    args = ap.parse_args()
+
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+    local_rank = int(os.environ.get("LOCAL_RANK", "0"))
+    distributed = world > 1
+
+    on_gpu = torch.cuda.is_available()
+    if distributed:
+        os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+        torch.distributed.init_process_group("nccl" if on_gpu else "gloo")
+    if on_gpu:
+        torch.cuda.set_device(local_rank)
+        device = torch.device("cuda", local_rank)
+        dtype = torch.bfloat16
+        from ring_attention_amd.ops.ring_flash_hip import ring_flash_attn_hip_ as attn
+    else:
+        # no-GPU fallback so the contract is testable anywhere: tiny oracle run
+        device = torch.device("cpu")
+        dtype = torch.float32
+        args.seq_per_gpu = min(args.seq_per_gpu, 512)
+        from ring_attention_amd.ops.ring_flash import ring_flash_attn_ as attn
+
+    b, n, h, d = args.batch, args.seq_per_gpu, args.heads, args.d_head
+    torch.manual_seed(1234 + rank)
+    q = torch.randn(b, n, h, d, device=device, dtype=dtype, requires_grad=True)
+    k = torch.randn(b, n, h, d, device=device, dtype=dtype, requires_grad=True)
+    v = torch.randn(b, n, h, d, device=device, dtype=dtype, requires_grad=True)
+
+    def step():
+        out, _ = attn(q, k, v, causal=args.causal, ring_reduce_col=True,
+                      striped_ring_attn=args.striped, ring_size=world,
+                      bucket_size=min(n, 1024))
+        out.backward(out.detach())  # fwd + full bwd (dq, dk, dv incl. ring)
+        q.grad = None; k.grad = None; v.grad = None
+
+    for _ in range(args.warmup):
+        step()
+
+    if distributed:
+        torch.distributed.barrier()
+    if on_gpu:
+        torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        step()
+    if distributed:
+        torch.distributed.barrier()
+    if on_gpu:
+        torch.cuda.synchronize()
+    t1 = time.perf_counter()
+
+    elapsed = torch.tensor([t1 - t0], dtype=torch.float64)
+    if distributed:
+        # max over ranks
+        elapsed_d = elapsed.to(device if on_gpu else "cpu")
+        torch.distributed.all_reduce(elapsed_d, torch.distributed.ReduceOp.MAX)
+        elapsed = elapsed_d.cpu()
+    secs = float(elapsed.item())
+    ms_per_step = secs / args.steps * 1e3
+
+    n_total = n * world
+    fwd_flops_per_gpu = 4.0 * b * n * n_total * d * h
+    if args.causal:
+        fwd_flops_per_gpu /= 2
+    total_flops = 2.5 * fwd_flops_per_gpu * world * args.steps
+    tflops_aggregate = total_flops / secs / 1e12
+
+    if rank == 0:
+        print(json.dumps({
+            "metric": "attn_tflops",
+            "value": round(tflops_aggregate, 2),
+            "unit": "TFLOP/s",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(ms_per_step, 3),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "bf16" if on_gpu else "fp32-cpu-fallback",
+            "data": "synthetic",
+            "config": {
+                "model": "ring_flash_attn (non-causal, d_head 64)" if not args.causal
+                         else "ring_flash_attn (causal)",
+                "global_batch": b,
+                "seq_len": n_total,
+                "seq_per_gpu": n,
+                "heads": h,
+                "d_head": d,
+                "parallelism": f"ring{world}",
+                "flop_convention": "fwd=4*b*n_shard*n_total*d*h (/2 causal); fwd+bwd=2.5x",
+            },
+        }))
+
+    if distributed:
+        torch.distributed.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

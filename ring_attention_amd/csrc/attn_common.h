@@ -1,0 +1,87 @@
+// Shared types for the ring_attention_amd CDNA4 kernels.
+#pragma once
+
+#include <hip/hip_runtime.h>
+#include <cstdint>
+
+#ifdef __HIPCC__
+#include <hip/hip_bf16.h>
+#endif
+
+namespace ring_attn {
+
+#ifdef __HIPCC__
+typedef __bf16 bf16_t;
+typedef bf16_t bf16x8 __attribute__((ext_vector_type(8)));
+typedef float f32x16 __attribute__((ext_vector_type(16)));
+typedef float f32x4 __attribute__((ext_vector_type(4)));
+typedef unsigned int u32x2 __attribute__((ext_vector_type(2)));
+#endif
+
+// Must match ring_attention_amd.ops.reference.MASK_VALUE (torch.finfo(f32).min)
+static constexpr float MASK_VALUE_F = -3.4028234663852886e38f;
+
+struct FwdParams {
+    const void* q;          // bf16 (B, Nq, H, D)
+    const void* k;          // bf16 (B, Nk, HK, D)
+    const void* v;          // bf16 (B, Nk, HK, D)
+    const void* kmask;      // uint8 (B, Nk) or nullptr; 1 = attend
+    float* o_acc;           // fp32 (B, H, D, Nq) scratch (nullptr for single-pass)
+    float* m;               // fp32 (B, H, Nq)
+    float* l;               // fp32 (B, H, Nq)
+    void* out;              // bf16 (B, Nq, H, D) (written when is_last)
+    float* lse;             // fp32 (B, H, Nq)    (written when is_last)
+    int b, h, hk, group;    // group = H / HK
+    long nq, nk;
+    float scale;
+    float softclamp_value;
+    int softclamp;          // bool
+    int causal;             // bool
+    long diag;              // attend iff j <= i + diag (when causal)
+    long win;               // attend iff i - j <= win (only when has_win)
+    int has_win;            // lookback window enabled
+    int is_first;           // initialize m/l/o instead of loading
+    int is_last;            // normalize + write out/lse instead of o_acc/m/l
+};
+
+void launch_attn_fwd(const FwdParams& p, int head_dim, hipStream_t stream);
+
+struct BwdParams {
+    const void* q;          // bf16 (B, Nq, H, D)
+    const void* k;          // bf16 (B, Nk, HK, D)
+    const void* v;          // bf16 (B, Nk, HK, D)
+    const void* dout;       // bf16 (B, Nq, H, D)
+    const void* kmask;      // uint8 (B, Nk) or nullptr
+    const float* lse;       // fp32 (B, H, Nq)
+    const float* delta;     // fp32 (B, H, Nq)  rowsum(do*o)
+    float* dq;              // fp32 (B, Nq, H, D)  accumulated via atomics
+    float* dk;              // fp32 (B, Nk, HK, D) accumulated (plain adds per WG)
+    float* dv;              // fp32 (B, Nk, HK, D)
+    int b, h, hk, group;
+    long nq, nk;
+    float scale;
+    float softclamp_value;
+    int softclamp;
+    int causal;
+    long diag;
+    long win;
+    int has_win;
+    int accumulate;         // dk/dv: 0 = overwrite, 1 = add to existing
+};
+
+void launch_attn_bwd(const BwdParams& p, int head_dim, hipStream_t stream);
+
+struct DecodeParams {
+    const void* q;          // bf16 (B, H, 1, D)
+    const void* k;          // bf16 (B, H, N, D)
+    const void* v;          // bf16 (B, H, N, D)
+    float* out;             // fp32 (B, H, 1, D)
+    float* lse;             // fp32 (B, H, 1, 1)
+    int b, h;
+    long n;
+    float scale;
+};
+
+void launch_decode_partial(const DecodeParams& p, int head_dim, hipStream_t stream);
+
+}  // namespace ring_attn

@@ -1,0 +1,321 @@
+// CDNA4 (gfx950) flash-attention FORWARD kernel with online-softmax
+// resume across ring passes.
+//
+// Brand-new MI355X design (capability counterpart of the reference's Triton
+// _fwd_kernel, /root/reference/ring_attention_pytorch/triton_flash_attn.py:52-430,
+// re-thought for wave64 + MFMA; no code ported):
+//
+//   * 8 waves / 512 threads per workgroup; each wave owns 32 Q rows
+//     (256-row Q tile per workgroup), KV tile = 64.
+//   * swapped QK^T: S^T[kv][q] = mfma(A=K, B=Q^T) so each lane holds a full
+//     slice of ONE q row's scores -> softmax is almost entirely in-register
+//     (31 VALU max/sum + one cross-half __shfl_xor), no LDS round trip.
+//   * P -> bf16 via v_cvt_pk_bf16_f32 pairs + v_permlane32_swap to build the
+//     PV B-operand fragments in-register (T12 pattern).
+//   * K and V^T staged in LDS with an XOR-16B swizzle (conflict-free
+//     ds_read_b128 column slices).
+//   * causality, striping and lookback all reduce to two integers:
+//     attend(i,j) <=> j <= i + diag  AND  i - j <= win
+//     (host folds ring-rank offsets / stride-R striping into diag/win).
+//   * resume contract: fp32 o_acc (B,H,D,Nq transposed scratch), m, l
+//     (B,H,Nq) persist between ring hops; IS_FIRST initializes instead of
+//     loading, IS_LAST normalizes and emits bf16 out (B,Nq,H,D) + lse.
+//     Single-hop launches (IS_FIRST && IS_LAST) never touch o_acc at all.
+//
+// Verified fragment maps (csrc/tools/mfma_probe.hip, run on MI355X):
+//   mfma_f32_32x32x16_bf16: A(i=l&31, k=8*(l>>5)+r)  B(k=8*(l>>5)+r, j=l&31)
+//                           C(i=(r&3)+8*(r>>2)+4*(l>>5), j=l&31)
+//   permlane32_swap: r0 = {lo: vdst.lo, hi: src.lo}, r1 = {lo: vdst.hi, hi: src.hi}
+
+#include <hip/hip_runtime.h>
+#include <hip/hip_bf16.h>
+#include <cstdint>
+
+#include "attn_common.h"
+
+namespace ring_attn {
+
+// ---------------------------------------------------------------------------
+// geometry
+// ---------------------------------------------------------------------------
+// QROWS_WG = 256 (8 waves x 32 rows), KVBLK = 64
+static constexpr int WAVES = 8;
+static constexpr int QROWS_WAVE = 32;
+static constexpr int QROWS_WG = WAVES * QROWS_WAVE;
+static constexpr int KVBLK = 64;
+
+// XOR swizzle of a 16-byte chunk index within a row (row stride D*2 bytes):
+// chunk' = chunk ^ (row & 7).  Applied identically on the staging write and
+// the fragment read, so the LDS image is consistent (both-sides rule).
+__device__ __forceinline__ int swz(int row, int chunk) { return chunk ^ (row & 7); }
+
+template <int D>
+struct FwdLds {
+    // K tile: [KVBLK][D] bf16, rows swizzled in 16B chunks
+    // V^T tile: [D][KVBLK] bf16, rows swizzled in 16B chunks (row = d)
+    __align__(16) __bf16 k[KVBLK * D];
+    __align__(16) __bf16 vt[D * KVBLK];
+    unsigned char kmask[KVBLK];
+};
+
+// ---------------------------------------------------------------------------
+// forward kernel
+// ---------------------------------------------------------------------------
+template <int D>
+__global__ __launch_bounds__(512, 1) void attn_fwd_kernel(FwdParams p) {
+    static_assert(D % 32 == 0);
+    constexpr int DBLK = D / 32;     // 32-d output blocks
+    constexpr int KSTEPS = D / 16;   // QK^T k-steps
+
+    __shared__ FwdLds<D> lds;
+
+    const int tid = threadIdx.x;
+    const int lane = tid & 63;
+    const int wid = tid >> 6;
+    const int l31 = lane & 31;
+    const int lhi = lane >> 5;        // 0 or 1
+
+    const int qtile = blockIdx.x;
+    const int bh = blockIdx.y;
+    const int b = bh / p.h;
+    const int h = bh % p.h;
+    const int hk = h / p.group;       // kv head
+
+    const long irow0 = (long)qtile * QROWS_WG + wid * QROWS_WAVE;  // this wave's first q row
+    const long i = irow0 + l31;                                     // this lane's q row
+    const bool row_valid = i < p.nq;
+    const long i_clamped = row_valid ? i : 0;
+
+    // ---- load Q fragments (bf16x8 per k-step): q[b, i, h, ks*16 + lhi*8 .. +8]
+    const __bf16* qbase = (const __bf16*)p.q + ((long)b * p.nq + i_clamped) * p.h * D + (long)h * D;
+    bf16x8 qf[KSTEPS];
+    #pragma unroll
+    for (int ks = 0; ks < KSTEPS; ++ks)
+        qf[ks] = *(const bf16x8*)(qbase + ks * 16 + lhi * 8);
+
+    // ---- accumulators
+    float m_run = MASK_VALUE_F;   // running row max of scaled scores (lane's q row)
+    float l_run = 0.f;            // running row sum
+    f32x16 o_acc[DBLK];           // O^T[d][q]: j = l31 = q, rows = d pattern
+    #pragma unroll
+    for (int db = 0; db < DBLK; ++db) o_acc[db] = f32x16{};
+
+    if (!p.is_first) {            // resume from a previous ring hop
+        const float* mrow = p.m + ((long)b * p.h + h) * p.nq;
+        const float* lrow = p.l + ((long)b * p.h + h) * p.nq;
+        m_run = mrow[i_clamped];
+        l_run = lrow[i_clamped];
+        const float* oa = p.o_acc + (((long)b * p.h + h) * D) * p.nq;
+        #pragma unroll
+        for (int db = 0; db < DBLK; ++db)
+            #pragma unroll
+            for (int r = 0; r < 16; ++r) {
+                int d = db * 32 + (r & 3) + 8 * (r >> 2) + 4 * lhi;
+                o_acc[db][r] = oa[(long)d * p.nq + i_clamped];
+            }
+    }
+
+    // ---- causal / lookback tile bounds for this workgroup
+    const long wg_i_min = (long)qtile * QROWS_WG;
+    const long wg_i_max = min((long)(qtile + 1) * QROWS_WG, p.nq) - 1;
+
+    const int num_kv_tiles = (int)((p.nk + KVBLK - 1) / KVBLK);
+
+    for (int t = 0; t < num_kv_tiles; ++t) {
+        const long j0 = (long)t * KVBLK;
+        const long jmax = min(j0 + KVBLK, p.nk) - 1;
+
+        if (p.causal && j0 > wg_i_max + p.diag) break;              // all future
+        if (p.has_win && (wg_i_min - jmax) > p.win) continue;      // all beyond window
+        const bool full_tile =
+            (jmax - j0 == KVBLK - 1) &&
+            (!p.causal || jmax <= wg_i_min + p.diag) &&
+            (!p.has_win || (wg_i_max - j0) <= p.win) &&
+            !p.kmask;
+
+        // ---- stage K tile -> lds.k [kv][D] (swizzled 16B chunks)
+        {
+            constexpr int CH_PER_ROW = D * 2 / 16;                  // 16B chunks per row
+            constexpr int CHUNKS = KVBLK * CH_PER_ROW;              // total (<=1024)
+            const __bf16* kbase = (const __bf16*)p.k + ((long)b * p.nk) * p.hk * D + (long)hk * D;
+            for (int c = tid; c < CHUNKS; c += 512) {
+                int row = c / CH_PER_ROW, ch = c % CH_PER_ROW;
+                long j = j0 + row;
+                uint4 val;
+                if (j <= jmax) val = *(const uint4*)(kbase + j * p.hk * D + ch * 8);
+                else val = uint4{0, 0, 0, 0};
+                *(uint4*)(lds.k + row * D + swz(row, ch) * 8) = val;
+            }
+            // ---- stage V^T tile -> lds.vt [d][kv] (swizzled)
+            // thread handles 2 kv rows x 8 d elements -> 8 ds_write_b32 pairs
+            const __bf16* vbase = (const __bf16*)p.v + ((long)b * p.nk) * p.hk * D + (long)hk * D;
+            constexpr int PAIRS = (KVBLK / 2) * (D / 8);            // threads needed
+            for (int c = tid; c < PAIRS; c += 512) {
+                int jp = c % (KVBLK / 2);                            // kv pair index
+                int d0 = (c / (KVBLK / 2)) * 8;
+                long ja = j0 + jp * 2, jb_ = ja + 1;
+                bf16x8 va = (ja <= jmax) ? *(const bf16x8*)(vbase + ja * p.hk * D + d0) : bf16x8{};
+                bf16x8 vb = (jb_ <= jmax) ? *(const bf16x8*)(vbase + jb_ * p.hk * D + d0) : bf16x8{};
+                #pragma unroll
+                for (int e = 0; e < 8; ++e) {
+                    int d = d0 + e;
+                    // vt row = d (KVBLK*2 bytes per row, 8 chunks of 16B)
+                    int byte_off = d * KVBLK * 2 + ((jp * 2 * 2) ^ ((d & 7) << 4));
+                    __bf16 pair[2] = {va[e], vb[e]};
+                    *(uint32_t*)((char*)lds.vt + byte_off) = *(uint32_t*)pair;
+                }
+            }
+            if (p.kmask) {
+                const unsigned char* mb = (const unsigned char*)p.kmask + (long)b * p.nk;
+                for (int c = tid; c < KVBLK; c += 512)
+                    lds.kmask[c] = (j0 + c <= jmax) ? mb[j0 + c] : 0;
+            }
+        }
+        __syncthreads();
+
+        // ---- QK^T: S^T[kv][q] for kv blocks {0,1} (32 rows each)
+        f32x16 s[2];
+        s[0] = f32x16{}; s[1] = f32x16{};
+        #pragma unroll
+        for (int kb = 0; kb < 2; ++kb) {
+            int krow = kb * 32 + l31;
+            #pragma unroll
+            for (int ks = 0; ks < KSTEPS; ++ks) {
+                int chunk = ks * 2 + lhi;                            // 16B chunk within K row
+                bf16x8 kf = *(const bf16x8*)(lds.k + krow * D + swz(krow, chunk) * 8);
+                s[kb] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf, qf[ks], s[kb], 0, 0, 0);
+            }
+        }
+
+        // ---- scale, clamp, mask in place; each lane: 32 scores of q row i
+        float smax = MASK_VALUE_F;
+        #pragma unroll
+        for (int kb = 0; kb < 2; ++kb)
+            #pragma unroll
+            for (int r = 0; r < 16; ++r) {
+                long j = j0 + kb * 32 + (r & 3) + 8 * (r >> 2) + 4 * lhi;
+                float x = s[kb][r] * p.scale;
+                if (p.softclamp) x = p.softclamp_value * tanhf(x / p.softclamp_value);
+                if (!full_tile) {
+                    bool ok = j <= jmax;
+                    if (p.causal) ok = ok && (j <= i + p.diag);
+                    if (p.has_win) ok = ok && (i - j <= p.win);
+                    if (p.kmask) ok = ok && lds.kmask[j - j0];
+                    if (!ok) x = MASK_VALUE_F;
+                }
+                s[kb][r] = x;
+                smax = fmaxf(smax, x);
+            }
+        smax = fmaxf(smax, __shfl_xor(smax, 32));
+
+        // ---- online softmax update
+        float m_new = fmaxf(m_run, smax);
+        float alpha = __expf(m_run - m_new);
+        float rowsum = 0.f;
+        uint32_t pk[16];                                            // packed bf16 pairs
+        #pragma unroll
+        for (int x2 = 0; x2 < 16; ++x2) {
+            float e0 = __expf(s[x2 >> 3][(2 * x2) & 15] - m_new);
+            float e1 = __expf(s[x2 >> 3][(2 * x2 + 1) & 15] - m_new);
+            rowsum += e0 + e1;
+            union { __hip_bfloat162 h2; uint32_t u; } cvt;
+            cvt.h2 = __float22bfloat162_rn(float2{e0, e1});
+            pk[x2] = cvt.u;
+        }
+        rowsum += __shfl_xor(rowsum, 32);
+        l_run = l_run * alpha + rowsum;
+        m_run = m_new;
+
+        // ---- build PV B-operand fragments via permlane32_swap
+        // pk[pb + x] holds the exp'd pair for kv rows (pattern):
+        //   x=0:(0,1)+4lhi  x=1:(2,3)+4lhi  x=2:(8,9)+4lhi   x=3:(10,11)+4lhi
+        //   x=4:(16,17)+4lhi x=5:(18,19)+4lhi x=6:(24,25)+4lhi x=7:(26,27)+4lhi
+        // B fragment for k-step needs u32 slot c = kv (8*lhi + 2c, +1), so:
+        //   swap(pk[pb+h*4+c], pk[pb+h*4+c+2]) -> r0 = slot c, r1 = slot c+2
+        uint32_t frag[4][4];
+        #pragma unroll
+        for (int kb = 0; kb < 2; ++kb) {
+            #pragma unroll
+            for (int half = 0; half < 2; ++half) {                   // k-step within block
+                #pragma unroll
+                for (int c = 0; c < 2; ++c) {
+                    u32x2 r = __builtin_amdgcn_permlane32_swap(
+                        pk[kb * 8 + half * 4 + c], pk[kb * 8 + half * 4 + c + 2], false, false);
+                    frag[kb * 2 + half][c] = r[0];
+                    frag[kb * 2 + half][c + 2] = r[1];
+                }
+            }
+        }
+
+        // ---- rescale O accumulator
+        #pragma unroll
+        for (int db = 0; db < DBLK; ++db)
+            #pragma unroll
+            for (int r = 0; r < 16; ++r) o_acc[db][r] *= alpha;
+
+        // ---- PV: O^T[d][q] += V^T[d][kv] P^T[kv][q]
+        #pragma unroll
+        for (int db = 0; db < DBLK; ++db) {
+            int drow = db * 32 + l31;                                // V^T row for A operand
+            #pragma unroll
+            for (int ks = 0; ks < 4; ++ks) {
+                int chunk = ks * 2 + lhi;                            // 16B chunk in vt row
+                bf16x8 vf = *(const bf16x8*)(lds.vt + drow * KVBLK + (swz(drow, chunk) & 7) * 8);
+                o_acc[db] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+                    vf, *(const bf16x8*)frag[ks], o_acc[db], 0, 0, 0);
+            }
+        }
+        __syncthreads();
+    }
+
+    // ---- epilogue
+    if (!row_valid) return;
+
+    if (p.is_last) {
+        float l_safe = fmaxf(l_run, 1e-38f);
+        float inv_l = 1.f / l_safe;
+        __bf16* ob = (__bf16*)p.out + ((long)b * p.nq + i) * p.h * D + (long)h * D;
+        #pragma unroll
+        for (int db = 0; db < DBLK; ++db)
+            #pragma unroll
+            for (int g = 0; g < 4; ++g) {                            // groups of 4 contiguous d
+                __bf16 four[4];
+                #pragma unroll
+                for (int e = 0; e < 4; ++e)
+                    four[e] = (__bf16)(o_acc[db][g * 4 + e] * inv_l);
+                int d = db * 32 + 8 * g + 4 * lhi;
+                *(uint2*)(ob + d) = *(uint2*)four;
+            }
+        if (lhi == 0) {
+            float* lsep = p.lse + ((long)b * p.h + h) * p.nq;
+            lsep[i] = __logf(l_safe) + m_run;
+        }
+    } else {
+        float* mrow = p.m + ((long)b * p.h + h) * p.nq;
+        float* lrow = p.l + ((long)b * p.h + h) * p.nq;
+        if (lhi == 0) { mrow[i] = m_run; lrow[i] = l_run; }
+        float* oa = p.o_acc + (((long)b * p.h + h) * D) * p.nq;
+        #pragma unroll
+        for (int db = 0; db < DBLK; ++db)
+            #pragma unroll
+            for (int r = 0; r < 16; ++r) {
+                int d = db * 32 + (r & 3) + 8 * (r >> 2) + 4 * lhi;
+                oa[(long)d * p.nq + i] = o_acc[db][r];
+            }
+    }
+}
+
+void launch_attn_fwd(const FwdParams& p, int head_dim, hipStream_t stream) {
+    dim3 grid((p.nq + QROWS_WG - 1) / QROWS_WG, p.b * p.h);
+    dim3 block(512);
+    if (head_dim == 64) {
+        hipLaunchKernelGGL(attn_fwd_kernel<64>, grid, block, 0, stream, p);
+    } else if (head_dim == 128) {
+        hipLaunchKernelGGL(attn_fwd_kernel<128>, grid, block, 0, stream, p);
+    } else {
+        // unsupported head dim is a host-side error (checked in bindings)
+        __builtin_trap();
+    }
+}
+
+}  // namespace ring_attn

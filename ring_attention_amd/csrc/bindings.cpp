@@ -1,0 +1,132 @@
+// PyTorch bindings for the ring_attention_amd CDNA4 kernels.
+//
+// Thin layer: shape/dtype/contiguity checks + parameter marshalling.  All
+// allocation happens in Python; kernels are launched on the current HIP
+// stream so they compose with RCCL comm streams and hipGraph capture.
+
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+#include <c10/hip/HIPStream.h>
+
+#include "attn_common.h"
+
+namespace ring_attn {
+
+#define CHECK_BF16_CONTIG(t) \
+    TORCH_CHECK(t.is_contiguous() && t.scalar_type() == at::kBFloat16, #t " must be contiguous bf16")
+#define CHECK_F32_CONTIG(t) \
+    TORCH_CHECK(t.is_contiguous() && t.scalar_type() == at::kFloat, #t " must be contiguous fp32")
+
+void attn_fwd(
+    at::Tensor q, at::Tensor k, at::Tensor v,
+    std::optional<at::Tensor> kmask,
+    std::optional<at::Tensor> o_acc,
+    std::optional<at::Tensor> m,
+    std::optional<at::Tensor> l,
+    std::optional<at::Tensor> out,
+    std::optional<at::Tensor> lse,
+    double scale, bool causal, int64_t diag, int64_t win, bool has_win,
+    bool softclamp, double softclamp_value,
+    bool is_first, bool is_last) {
+    CHECK_BF16_CONTIG(q); CHECK_BF16_CONTIG(k); CHECK_BF16_CONTIG(v);
+    TORCH_CHECK(q.dim() == 4 && k.dim() == 4 && v.dim() == 4, "q/k/v must be (B,N,H,D)");
+    const int64_t B = q.size(0), Nq = q.size(1), H = q.size(2), D = q.size(3);
+    const int64_t Nk = k.size(1), HK = k.size(2);
+    TORCH_CHECK(D == 64 || D == 128, "head dim must be 64 or 128 (got ", D, ")");
+    TORCH_CHECK(H % HK == 0, "q heads must be a multiple of kv heads");
+    TORCH_CHECK(k.size(0) == B && v.size(0) == B && v.size(1) == Nk && v.size(2) == HK && v.size(3) == D);
+
+    FwdParams p{};
+    p.q = q.data_ptr(); p.k = k.data_ptr(); p.v = v.data_ptr();
+    p.kmask = nullptr;
+    if (kmask.has_value()) {
+        TORCH_CHECK(kmask->is_contiguous() && kmask->scalar_type() == at::kByte, "kmask must be contiguous uint8");
+        TORCH_CHECK(kmask->size(0) == B && kmask->size(1) == Nk);
+        p.kmask = kmask->data_ptr();
+    }
+    if (!(is_first && is_last)) {
+        TORCH_CHECK(o_acc && m && l, "multi-pass launches need o_acc/m/l scratch");
+        CHECK_F32_CONTIG((*o_acc)); CHECK_F32_CONTIG((*m)); CHECK_F32_CONTIG((*l));
+        TORCH_CHECK(o_acc->numel() == B * H * D * Nq && m->numel() == B * H * Nq);
+        p.o_acc = o_acc->data_ptr<float>();
+        p.m = m->data_ptr<float>();
+        p.l = l->data_ptr<float>();
+    }
+    if (is_last) {
+        TORCH_CHECK(out && lse, "last pass needs out/lse");
+        CHECK_BF16_CONTIG((*out)); CHECK_F32_CONTIG((*lse));
+        TORCH_CHECK(out->sizes() == q.sizes() && lse->numel() == B * H * Nq);
+        p.out = out->data_ptr();
+        p.lse = lse->data_ptr<float>();
+    }
+    p.b = (int)B; p.h = (int)H; p.hk = (int)HK; p.group = (int)(H / HK);
+    p.nq = Nq; p.nk = Nk;
+    p.scale = (float)scale;
+    p.softclamp = softclamp; p.softclamp_value = (float)softclamp_value;
+    p.causal = causal; p.diag = diag; p.win = win; p.has_win = has_win;
+    p.is_first = is_first; p.is_last = is_last;
+
+    launch_attn_fwd(p, (int)D, at::hip::getCurrentHIPStream());
+    TORCH_CHECK(hipGetLastError() == hipSuccess, "attn_fwd launch failed");
+}
+
+void attn_bwd(
+    at::Tensor q, at::Tensor k, at::Tensor v, at::Tensor dout,
+    std::optional<at::Tensor> kmask,
+    at::Tensor lse, at::Tensor delta,
+    at::Tensor dq, at::Tensor dk, at::Tensor dv,
+    double scale, bool causal, int64_t diag, int64_t win, bool has_win,
+    bool softclamp, double softclamp_value, bool accumulate) {
+    CHECK_BF16_CONTIG(q); CHECK_BF16_CONTIG(k); CHECK_BF16_CONTIG(v); CHECK_BF16_CONTIG(dout);
+    CHECK_F32_CONTIG(lse); CHECK_F32_CONTIG(delta);
+    CHECK_F32_CONTIG(dq); CHECK_F32_CONTIG(dk); CHECK_F32_CONTIG(dv);
+    const int64_t B = q.size(0), Nq = q.size(1), H = q.size(2), D = q.size(3);
+    const int64_t Nk = k.size(1), HK = k.size(2);
+    TORCH_CHECK(D == 64 || D == 128, "head dim must be 64 or 128");
+    TORCH_CHECK(dq.numel() == q.numel() && dk.numel() == k.numel() && dv.numel() == v.numel());
+    TORCH_CHECK(lse.numel() == B * H * Nq && delta.numel() == B * H * Nq);
+
+    BwdParams p{};
+    p.q = q.data_ptr(); p.k = k.data_ptr(); p.v = v.data_ptr(); p.dout = dout.data_ptr();
+    p.kmask = nullptr;
+    if (kmask.has_value()) {
+        TORCH_CHECK(kmask->is_contiguous() && kmask->scalar_type() == at::kByte);
+        p.kmask = kmask->data_ptr();
+    }
+    p.lse = lse.data_ptr<float>(); p.delta = delta.data_ptr<float>();
+    p.dq = dq.data_ptr<float>(); p.dk = dk.data_ptr<float>(); p.dv = dv.data_ptr<float>();
+    p.b = (int)B; p.h = (int)H; p.hk = (int)HK; p.group = (int)(H / HK);
+    p.nq = Nq; p.nk = Nk;
+    p.scale = (float)scale;
+    p.softclamp = softclamp; p.softclamp_value = (float)softclamp_value;
+    p.causal = causal; p.diag = diag; p.win = win; p.has_win = has_win;
+    p.accumulate = accumulate;
+
+    launch_attn_bwd(p, (int)D, at::hip::getCurrentHIPStream());
+    TORCH_CHECK(hipGetLastError() == hipSuccess, "attn_bwd launch failed");
+}
+
+std::vector<at::Tensor> decode_partial(at::Tensor q, at::Tensor k, at::Tensor v) {
+    // q (B,H,1,D); k,v (B,H,N,D) bf16 -> (out fp32 (B,H,1,D), lse fp32 (B,H,1,1))
+    CHECK_BF16_CONTIG(q); CHECK_BF16_CONTIG(k); CHECK_BF16_CONTIG(v);
+    const int64_t B = q.size(0), H = q.size(1), D = q.size(3), N = k.size(2);
+    TORCH_CHECK(D == 64 || D == 128, "head dim must be 64 or 128");
+    auto out = at::empty({B, H, 1, D}, q.options().dtype(at::kFloat));
+    auto lse = at::empty({B, H, 1, 1}, q.options().dtype(at::kFloat));
+    DecodeParams p{};
+    p.q = q.data_ptr(); p.k = k.data_ptr(); p.v = v.data_ptr();
+    p.out = out.data_ptr<float>(); p.lse = lse.data_ptr<float>();
+    p.b = (int)B; p.h = (int)H; p.n = N;
+    p.scale = (float)(1.0 / std::sqrt((double)D));
+    launch_decode_partial(p, (int)D, at::hip::getCurrentHIPStream());
+    TORCH_CHECK(hipGetLastError() == hipSuccess, "decode launch failed");
+    return {out, lse};
+}
+
+}  // namespace ring_attn
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
+    mod.def("attn_fwd", &ring_attn::attn_fwd, "CDNA4 flash attention forward (resumable)");
+    mod.def("attn_bwd", &ring_attn::attn_bwd, "CDNA4 flash attention backward");
+    mod.def("decode_partial", &ring_attn::decode_partial, "CDNA4 single-query decode partial");
+}

@@ -1,0 +1,103 @@
+// Single-query (decode) attention partial for tree-attention decoding.
+//
+// Capability counterpart of the reference's tree-decode local partial
+// (/root/reference/ring_attention_pytorch/tree_attn_decoding.py:54-79).
+// Memory-bound: one wave per (b, h); KV streamed with 16-byte loads; softmax
+// via in-wave shuffle reduction; fp32 out + lse emitted for the cross-rank
+// RCCL merge (done in Python with a packed all-reduce).
+
+#include <hip/hip_runtime.h>
+#include <hip/hip_bf16.h>
+
+#include "attn_common.h"
+
+namespace ring_attn {
+
+template <int D>
+__global__ __launch_bounds__(256) void decode_partial_kernel(DecodeParams p) {
+    // one wave per (b, h); 4 waves per block
+    const int wave_global = (blockIdx.x * 4) + (threadIdx.x >> 6);
+    if (wave_global >= p.b * p.h) return;
+    const int b = wave_global / p.h;
+    const int h = wave_global % p.h;
+    const int lane = threadIdx.x & 63;
+
+    const __bf16* qp = (const __bf16*)p.q + ((long)b * p.h + h) * D;
+    const __bf16* kp = (const __bf16*)p.k + ((long)b * p.h + h) * p.n * D;
+    const __bf16* vp = (const __bf16*)p.v + ((long)b * p.h + h) * p.n * D;
+
+    // q in registers (fp32), replicated per lane as needed
+    float qreg[D];
+    #pragma unroll
+    for (int d = 0; d < D; ++d) qreg[d] = (float)qp[d];
+
+    // pass 1+2 fused online: each lane handles keys lane, lane+64, ...
+    float m = MASK_VALUE_F, l = 0.f;
+    float acc[D];
+    #pragma unroll
+    for (int d = 0; d < D; ++d) acc[d] = 0.f;
+
+    for (long j = lane; j < p.n; j += 64) {
+        const __bf16* krow = kp + j * D;
+        float s = 0.f;
+        #pragma unroll
+        for (int d = 0; d < D; d += 8) {
+            bf16x8 kv8 = *(const bf16x8*)(krow + d);
+            #pragma unroll
+            for (int e = 0; e < 8; ++e) s += qreg[d + e] * (float)kv8[e];
+        }
+        s *= p.scale;
+        float m_new = fmaxf(m, s);
+        float alpha = __expf(m - m_new);
+        float w = __expf(s - m_new);
+        l = l * alpha + w;
+        const __bf16* vrow = vp + j * D;
+        #pragma unroll
+        for (int d = 0; d < D; d += 8) {
+            bf16x8 vv8 = *(const bf16x8*)(vrow + d);
+            #pragma unroll
+            for (int e = 0; e < 8; ++e) acc[d + e] = acc[d + e] * alpha + w * (float)vv8[e];
+        }
+        m = m_new;
+    }
+
+    // cross-lane merge (all 64 lanes -> lane 0's running stats)
+    #pragma unroll
+    for (int off = 32; off > 0; off >>= 1) {
+        float m2 = __shfl_xor(m, off);
+        float l2 = __shfl_xor(l, off);
+        float m_new = fmaxf(m, m2);
+        float a1 = __expf(m - m_new), a2 = __expf(m2 - m_new);
+        l = l * a1 + l2 * a2;
+        #pragma unroll
+        for (int d = 0; d < D; ++d) {
+            float o2 = __shfl_xor(acc[d], off);
+            acc[d] = acc[d] * a1 + o2 * a2;
+        }
+        m = m_new;
+    }
+
+    float l_safe = fmaxf(l, 1e-38f);
+    if (lane == 0) {
+        float inv = 1.f / l_safe;
+        float* op = p.out + ((long)b * p.h + h) * D;
+        #pragma unroll
+        for (int d = 0; d < D; ++d) op[d] = acc[d] * inv;
+        p.lse[(long)b * p.h + h] = __logf(l_safe) + m;
+    }
+}
+
+void launch_decode_partial(const DecodeParams& p, int head_dim, hipStream_t stream) {
+    int waves = p.b * p.h;
+    dim3 grid((waves + 3) / 4);
+    dim3 block(256);
+    if (head_dim == 64) {
+        hipLaunchKernelGGL(decode_partial_kernel<64>, grid, block, 0, stream, p);
+    } else if (head_dim == 128) {
+        hipLaunchKernelGGL(decode_partial_kernel<128>, grid, block, 0, stream, p);
+    } else {
+        __builtin_trap();
+    }
+}
+
+}  // namespace ring_attn

@@ -46,7 +46,7 @@ class RingRotaryEmbedding(nn.Module):
                 pos = topo.ring_rank * seq_len + torch.arange(seq_len, device=device)
         else:
             pos = torch.arange(seq_len, device=device)
-        freqs = torch.einsum("i,j->ij", pos.float(), self.inv_freq)
+        freqs = torch.einsum("i,j->ij", pos.float(), self.inv_freq.float())
         return torch.cat((freqs, freqs), dim=-1)  # (n, dim)
 
 

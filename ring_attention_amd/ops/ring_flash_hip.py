@@ -1,0 +1,201 @@
+"""Ring flash attention backed by the CDNA4 HIP kernels (the GPU compute path).
+
+Same semantics as ops/ring_flash.py (the oracle), same ring transport
+(double-buffered overlapped P2P), but each hop's compute is ONE resumable
+HIP kernel launch:
+
+- forward: attn_fwd with (o_acc fp32, m, l) persisting across hops; causality,
+  striping and lookback folded into two integers (diag, win) per hop (host
+  computes them from ring ranks — see csrc/attn_fwd.hip header).  Hops whose
+  kv shard is entirely masked are skipped host-side (kv still circulates).
+- backward: delta precomputed once; attn_bwd per hop; dq accumulates in ONE
+  fp32 buffer via kernel atomics across hops (no per-hop add); dk/dv per hop
+  go through the pipelined RingAccumulator, in the kernel's transposed
+  scratch layouts (dk (B,HK,Nk,D), dv (B,HK,D,Nk)); final transpose once.
+
+Capability parity with the reference's ring_flash_attn_cuda
+(/root/reference/ring_attention_pytorch/ring_flash_attention_cuda.py:40-371)
+with its §2.5 bugs fixed and full comm/compute overlap.
+"""
+
+from __future__ import annotations
+
+import torch
+from torch import Tensor
+from torch.autograd import Function
+
+from ..parallel import RingAccumulator, RingTopology, all_ring_pass, is_distributed
+from . import hip_ext
+from .ring_flash import max_hops_for_lookback
+
+
+def _hop_geometry(rq: int, rk: int, n: int, ring_size: int, striped: bool,
+                  causal: bool, lookback: int | None) -> tuple[bool, int, int]:
+    """Returns (skip_hop, diag, win) for a (q-rank, kv-source-rank) pair.
+
+    attend(i, j) <=> (not causal or j <= i + diag) and (win < 0 or i - j <= win)
+    """
+    if striped:
+        diag = 0 if rk <= rq else -1
+        if lookback is not None:
+            # (i - j) * R + (rq - rk) <= L  <=>  i - j <= floor((L - rq + rk) / R)
+            win = (lookback - rq + rk) // ring_size
+        else:
+            win = -1
+    else:
+        diag = (rq - rk) * n
+        # (rq - rk) * n + i - j <= L  <=>  i - j <= L - diag
+        win = (lookback - diag) if lookback is not None else -1
+    skip = False
+    if causal and diag < 1 - n:
+        skip = True        # whole shard is in the future
+    if lookback is not None:
+        if win < 1 - n:
+            skip = True    # whole shard beyond the window
+        if causal and win < -diag:
+            skip = True    # window and causal triangle do not intersect
+    return skip, diag, win
+
+
+class RingFlashAttentionHIPFunction(Function):
+    @staticmethod
+    def forward(ctx, q, k, v, mask, causal, bucket_size, ring_reduce, striped,
+                max_lookback_seq_len, ring_size, softclamp_qk_sim, softclamp_value):
+        assert q.is_cuda, "HIP path requires GPU tensors"
+        b, n, h, d = q.shape
+        hk = k.shape[2]
+        in_dtype = q.dtype
+        scale = d ** -0.5
+        lookback = max_lookback_seq_len
+        if lookback is not None:
+            assert causal, "lookback (sliding window) requires causal"
+
+        use_ring = ring_reduce and is_distributed()
+        topo = RingTopology(ring_size if use_ring else 1,
+                            rank=None if use_ring else 0,
+                            world_size=None if use_ring else 1)
+        hops = max_hops_for_lookback(causal, striped, lookback, n, topo.ring_size)
+
+        qb = q.to(torch.bfloat16).contiguous()
+        kb = k.to(torch.bfloat16).contiguous()
+        vb = v.to(torch.bfloat16).contiguous()
+        mask_u8 = mask.to(torch.uint8).contiguous() if mask is not None else None
+
+        ext = hip_ext.require()
+        out = torch.empty_like(qb)
+        lse = torch.empty(b, h, n, device=q.device, dtype=torch.float32)
+
+        multi = hops > 1
+        o_acc = m = l = None
+        if multi:
+            o_acc = torch.empty(b, h, d, n, device=q.device, dtype=torch.float32)
+            m = torch.empty(b, h, n, device=q.device, dtype=torch.float32)
+            l = torch.empty(b, h, n, device=q.device, dtype=torch.float32)
+
+        # which hops actually compute (host-side skip of fully-masked shards)
+        rq = topo.ring_rank
+        plan = []
+        for hop in range(hops):
+            rk = topo.source_of_hop(hop)
+            skip, diag, win = _hop_geometry(rq, rk, n, topo.ring_size, striped,
+                                            causal, lookback)
+            plan.append((skip, diag, win))
+        active = [i for i, (s, _, _) in enumerate(plan) if not s]
+        assert active, "every hop masked — degenerate configuration"
+        first_active, last_active = active[0], active[-1]
+
+        kv = torch.stack((kb, vb))
+        ring_tensors = (kv,) if mask_u8 is None else (kv, mask_u8)
+
+        for info, tensors in all_ring_pass(topo, *ring_tensors, max_hops=hops):
+            skip, diag, win = plan[info.hop]
+            if skip:
+                continue
+            kv_t = tensors[0]
+            mk = tensors[1] if mask_u8 is not None else None
+            ext.attn_fwd(qb, kv_t[0], kv_t[1], mk,
+                         o_acc, m, l, out, lse,
+                         scale, causal, diag, win, lookback is not None,
+                         softclamp_qk_sim, softclamp_value,
+                         info.hop == first_active, info.hop == last_active)
+
+        ctx.save_for_backward(qb, kb, vb, out, lse,
+                              mask_u8 if mask_u8 is not None else torch.empty(0))
+        ctx.params = (causal, striped, lookback, hops, softclamp_qk_sim,
+                      softclamp_value, use_ring, topo.ring_size, in_dtype)
+        return out.to(in_dtype), lse
+
+    @staticmethod
+    def backward(ctx, do, _dlse):
+        qb, kb, vb, out, lse, mask_u8 = ctx.saved_tensors
+        mask_u8 = mask_u8 if mask_u8.numel() else None
+        (causal, striped, lookback, hops, softclamp_qk_sim, softclamp_value,
+         use_ring, ring_size, in_dtype) = ctx.params
+        b, n, h, d = qb.shape
+        hk = kb.shape[2]
+        scale = d ** -0.5
+
+        topo = RingTopology(ring_size if use_ring else 1,
+                            rank=None if use_ring else 0,
+                            world_size=None if use_ring else 1)
+        ext = hip_ext.require()
+
+        dob = do.to(torch.bfloat16).contiguous()
+        # delta = rowsum(do * o) in fp32: (b, h, n)
+        delta = (dob.float() * out.float()).sum(dim=-1).permute(0, 2, 1).contiguous()
+
+        dq = torch.zeros(b, n, h, d, device=qb.device, dtype=torch.float32)
+        rq = topo.ring_rank
+
+        kv = torch.stack((kb, vb))
+        ring_tensors = (kv,) if mask_u8 is None else (kv, mask_u8)
+        acc = RingAccumulator(topo)
+
+        for info, tensors in all_ring_pass(topo, *ring_tensors, max_hops=hops):
+            rk = info.source_ring_rank
+            skip, diag, win = _hop_geometry(rq, rk, n, topo.ring_size, striped,
+                                            causal, lookback)
+            kv_t = tensors[0]
+            mk = tensors[1] if mask_u8 is not None else None
+            # circulate dk/dv flat in the kernel's native scratch layouts
+            # (dk (b,hk,n,d), dv (b,hk,d,n)); elementwise accumulation is
+            # layout-agnostic, so only ONE final permute happens at home
+            contrib = torch.zeros(2, b * hk * n * d, device=qb.device,
+                                  dtype=torch.float32)
+            if not skip:
+                dk_n = contrib[0].view(b, hk, n, d)
+                dv_n = contrib[1].view(b, hk, d, n)
+                ext.attn_bwd(qb, kv_t[0], kv_t[1], dob, mk, lse, delta,
+                             dq, dk_n, dv_n,
+                             scale, causal, diag, win, lookback is not None,
+                             softclamp_qk_sim, softclamp_value, False)
+            acc.step(contrib, info.is_last)
+
+        dkv = acc.finish(hops)
+        dk_home = dkv[0].view(b, hk, n, d).permute(0, 2, 1, 3).contiguous()
+        dv_home = dkv[1].view(b, hk, d, n).permute(0, 3, 1, 2).contiguous()
+
+        return (dq.to(in_dtype), dk_home.to(in_dtype), dv_home.to(in_dtype),
+                None, None, None, None, None, None, None, None, None)
+
+
+def ring_flash_attn_hip_(
+    q: Tensor, k: Tensor, v: Tensor,
+    mask: Tensor | None = None,
+    causal: bool = False,
+    bucket_size: int = 1024,
+    ring_reduce_col: bool = False,
+    striped_ring_attn: bool = False,
+    max_lookback_seq_len: int | None = None,
+    ring_size: int | None = None,
+    softclamp_qk_sim: bool = False,
+    softclamp_value: float = 50.0,
+) -> tuple[Tensor, Tensor]:
+    return RingFlashAttentionHIPFunction.apply(
+        q, k, v, mask, causal, bucket_size, ring_reduce_col, striped_ring_attn,
+        max_lookback_seq_len, ring_size, softclamp_qk_sim, softclamp_value)
+
+
+def ring_flash_attn_hip(q, k, v, **kwargs) -> Tensor:
+    out, _ = ring_flash_attn_hip_(q, k, v, **kwargs)
+    return out
