@@ -56,12 +56,17 @@ struct BwdLds {
     float delta[QT];
 };
 
+// XOR swizzle valid for rows of CH chunks (CH a power of two, <= 8 kept)
+template <int CH>
+__device__ __forceinline__ int tmask() { return (CH - 1) < 7 ? (CH - 1) : 7; }
+
 // stage a [rows][D] bf16 tile row-major (swizzled) AND transposed (swizzled)
 template <int D, int QT>
 __device__ void stage_rowmajor_and_t(
     const __bf16* gbase, long row0, long rowmax, long row_stride,
     __bf16* lds_rm, __bf16* lds_t, int tid) {
     constexpr int CH_PER_ROW = D * 2 / 16;
+    constexpr int TM = (QT / 8 - 1) < 7 ? (QT / 8 - 1) : 7;   // transposed-row chunk mask
     for (int c = tid; c < QT * CH_PER_ROW; c += 512) {
         int row = c / CH_PER_ROW, ch = c % CH_PER_ROW;
         long gr = row0 + row;
@@ -73,7 +78,7 @@ __device__ void stage_rowmajor_and_t(
         #pragma unroll
         for (int e = 0; e < 8; ++e) {
             int d = ch * 8 + e;
-            int byte_off = d * QT * 2 + ((row * 2) ^ ((d & 7) << 4));
+            int byte_off = d * QT * 2 + ((row * 2) ^ ((d & TM) << 4));
             *((__bf16*)((char*)lds_t + byte_off)) = vals[e];
         }
     }
@@ -256,14 +261,15 @@ __global__ __launch_bounds__(512, 1) void attn_bwd_kernel(BwdParams p) {
                         int qk = qb * 2 + half;                  // 16-q k-step index
                         int drow = db * 32 + l31;
                         // dO^T row = d, contiguous q at qk*16 + lhi*8
+                        constexpr int TM = (QT / 8 - 1) < 7 ? (QT / 8 - 1) : 7;
                         int ch_dot = qk * 2 + lhi;
                         bf16x8 doa = *(const bf16x8*)(lds.dot + drow * QT +
-                                                      (bswz(drow, ch_dot)) * 8);
+                                                      ((ch_dot ^ (drow & TM))) * 8);
                         dv_acc[db] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
                             doa, *(const bf16x8*)p_frag[half], dv_acc[db], 0, 0, 0);
                         // Q^T as B: lane j = d column; read Q^T row (d) contiguous q
                         bf16x8 qta = *(const bf16x8*)(lds.qt + drow * QT +
-                                                      (bswz(drow, ch_dot)) * 8);
+                                                      ((ch_dot ^ (drow & TM))) * 8);
                         dk_acc[db] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
                             *(const bf16x8*)ds_frag[half], qta, dk_acc[db], 0, 0, 0);
                     }
