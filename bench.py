@@ -36,6 +36,8 @@ def main():
     ap.add_argument("--batch", type=int, default=1)
     ap.add_argument("--causal", action="store_true")
     ap.add_argument("--striped", action="store_true")
+    ap.add_argument("--fwd-only", action="store_true",
+                    help="measure forward only (diagnostics; not the headline metric)")
     args = ap.parse_args()
 
     world = int(os.environ.get("WORLD_SIZE", "1"))
@@ -69,8 +71,9 @@ def main():
         out, _ = attn(q, k, v, causal=args.causal, ring_reduce_col=True,
                       striped_ring_attn=args.striped, ring_size=world,
                       bucket_size=min(n, 1024))
-        out.backward(out.detach())  # fwd + full bwd (dq, dk, dv incl. ring)
-        q.grad = None; k.grad = None; v.grad = None
+        if not args.fwd_only:
+            out.backward(out.detach())  # fwd + full bwd (dq, dk, dv incl. ring)
+            q.grad = None; k.grad = None; v.grad = None
 
     for _ in range(args.warmup):
         step()
@@ -101,7 +104,7 @@ def main():
     fwd_flops_per_gpu = 4.0 * b * n * n_total * d * h
     if args.causal:
         fwd_flops_per_gpu /= 2
-    total_flops = 2.5 * fwd_flops_per_gpu * world * args.steps
+    total_flops = (1.0 if args.fwd_only else 2.5) * fwd_flops_per_gpu * world * args.steps
     tflops_aggregate = total_flops / secs / 1e12
 
     if rank == 0:
