@@ -141,8 +141,14 @@ __global__ __launch_bounds__(512, 1) void attn_bwd_dq_kernel(BwdParams p) {
     const long wg_i_min = (long)qtile * DQ_QROWS_WG;
     const long wg_i_max = min((long)(qtile + 1) * DQ_QROWS_WG, p.nq) - 1;
     const int num_kv_tiles = (int)((p.nk + DQ_KVBLK - 1) / DQ_KVBLK);
+    int zt_lo = 0, zt_hi = num_kv_tiles;
+    if (p.split > 1) {                     // grid.z splits the kv walk
+        int per = (num_kv_tiles + p.split - 1) / p.split;
+        zt_lo = blockIdx.z * per;
+        zt_hi = min(num_kv_tiles, zt_lo + per);
+    }
 
-    for (int t = 0; t < num_kv_tiles; ++t) {
+    for (int t = zt_lo; t < zt_hi; ++t) {
         const long j0 = (long)t * DQ_KVBLK;
         const long jmax = min(j0 + DQ_KVBLK, p.nk) - 1;
         if (p.causal && j0 > wg_i_max + p.diag) break;
@@ -240,14 +246,16 @@ __global__ __launch_bounds__(512, 1) void attn_bwd_dq_kernel(BwdParams p) {
     }
 
     if (!row_valid) return;
-    // epilogue: dq (B, Nq, H, D) fp32 plain accumulate (unique writer per row)
+    // epilogue: dq (B, Nq, H, D) fp32; unique writer per row unless the kv
+    // walk is split across grid.z (then fp32 atomics, contention = split)
     float* dqp = p.dq + ((long)b * p.nq + i) * p.h * D + (long)h * D;
     #pragma unroll
     for (int db = 0; db < DBLK; ++db)
         #pragma unroll
         for (int r = 0; r < 16; ++r) {
             int d = db * 32 + (r & 3) + 8 * (r >> 2) + 4 * lhi;
-            dqp[d] += dq_acc[db][r];
+            if (p.split > 1) atomicAdd(dqp + d, dq_acc[db][r]);
+            else dqp[d] += dq_acc[db][r];
         }
 }
 
@@ -328,6 +336,11 @@ __global__ __launch_bounds__(512, 1) void attn_bwd_dkv_kernel(BwdParams p) {
             long i_max_needed = jmax + p.win;
             if (i_max_needed < (long)num_q_tiles * QT)
                 t1 = (int)min((long)num_q_tiles, i_max_needed / QT + 1);
+        }
+        if (p.split > 1) {                 // grid.z splits the q walk
+            int per = (num_q_tiles + p.split - 1) / p.split;
+            t0 = max(t0, (int)(blockIdx.z * per));
+            t1 = min(t1, (int)((blockIdx.z + 1) * per));
         }
 
         for (int t = t0; t < t1; ++t) {
@@ -445,7 +458,9 @@ __global__ __launch_bounds__(512, 1) void attn_bwd_dkv_kernel(BwdParams p) {
                 int d = db * 32 + l31;
                 if (kvrow <= jmax) {
                     float* dst = dkb + kvrow * D + d;
-                    if (p.accumulate) *dst += dk_acc[db][r]; else *dst = dk_acc[db][r];
+                    if (p.split > 1) atomicAdd(dst, dk_acc[db][r]);
+                    else if (p.accumulate) *dst += dk_acc[db][r];
+                    else *dst = dk_acc[db][r];
                 }
             }
         float* dvb = p.dv + (((long)b * p.hk + hkh) * D) * p.nk;
@@ -455,15 +470,18 @@ __global__ __launch_bounds__(512, 1) void attn_bwd_dkv_kernel(BwdParams p) {
             for (int r = 0; r < 16; ++r) {
                 int d = db * 32 + (r & 3) + 8 * (r >> 2) + 4 * lhi;
                 float* dst = dvb + (long)d * p.nk + j;
-                if (p.accumulate) *dst += dv_acc[db][r]; else *dst = dv_acc[db][r];
+                if (p.split > 1) atomicAdd(dst, dv_acc[db][r]);
+                else if (p.accumulate) *dst += dv_acc[db][r];
+                else *dst = dv_acc[db][r];
             }
     }
 }
 
 void launch_attn_bwd(const BwdParams& p, int head_dim, hipStream_t stream) {
     dim3 block(512);
-    dim3 grid_dq((p.nq + DQ_QROWS_WG - 1) / DQ_QROWS_WG, p.b * p.h);
-    dim3 grid_dkv((p.nk + KVROWS_WG - 1) / KVROWS_WG, p.b * p.hk);
+    int z = p.split > 1 ? p.split : 1;
+    dim3 grid_dq((p.nq + DQ_QROWS_WG - 1) / DQ_QROWS_WG, p.b * p.h, z);
+    dim3 grid_dkv((p.nk + KVROWS_WG - 1) / KVROWS_WG, p.b * p.hk, z);
     if (head_dim == 64) {
         hipLaunchKernelGGL((attn_bwd_dq_kernel<64>), grid_dq, block, 0, stream, p);
         hipLaunchKernelGGL((attn_bwd_dkv_kernel<64, 64>), grid_dkv, block, 0, stream, p);

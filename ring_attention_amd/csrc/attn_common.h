@@ -42,7 +42,27 @@ struct FwdParams {
     int has_win;            // lookback window enabled
     int is_first;           // initialize m/l/o instead of loading
     int is_last;            // normalize + write out/lse instead of o_acc/m/l
+    int kv_split;           // >1: grid.z splits the kv range; o_acc/m/l hold
+                            // kv_split partials (merged by attn_fwd_merge)
 };
+
+struct FwdMergeParams {
+    const float* o_part;    // fp32 (S, B, H, D, Nq) unnormalized partials
+    const float* m_part;    // fp32 (S, B, H, Nq)
+    const float* l_part;    // fp32 (S, B, H, Nq)
+    float* o_acc;           // fp32 (B, H, D, Nq) running accumulator (ring)
+    float* m;               // fp32 (B, H, Nq)
+    float* l;               // fp32 (B, H, Nq)
+    void* out;              // bf16 (B, Nq, H, D) written when is_last
+    float* lse;             // fp32 (B, H, Nq)
+    int splits;
+    int b, h;
+    long nq;
+    int is_first;           // no running state to fold in
+    int is_last;            // normalize + emit out/lse
+};
+
+void launch_attn_fwd_merge(const FwdMergeParams& p, int head_dim, hipStream_t stream);
 
 void launch_attn_fwd(const FwdParams& p, int head_dim, hipStream_t stream);
 
@@ -66,7 +86,9 @@ struct BwdParams {
     long diag;
     long win;
     int has_win;
-    int accumulate;         // dk/dv: 0 = overwrite, 1 = add to existing
+    int accumulate;
+    int split;              // >1: grid.z splits the contraction range; dq/dk/dv
+                            // accumulated with fp32 atomics instead of plain ops         // dk/dv: 0 = overwrite, 1 = add to existing
 };
 
 void launch_attn_bwd(const BwdParams& p, int head_dim, hipStream_t stream);

@@ -39,10 +39,11 @@ namespace ring_attn {
 // geometry
 // ---------------------------------------------------------------------------
 // QROWS_WG = 256 (8 waves x 32 rows), KVBLK = 64
-static constexpr int WAVES = 8;
-static constexpr int QROWS_WAVE = 32;
+static constexpr int WAVES = 8;          // 8-wave WGs: the 2-waves/SIMD paired
+static constexpr int QROWS_WAVE = 32;    // regime (4-wave variant measured slower)
 static constexpr int QROWS_WG = WAVES * QROWS_WAVE;
 static constexpr int KVBLK = 64;
+static constexpr int NTHREADS = WAVES * 64;
 
 // XOR swizzle of a 16-byte chunk index within a row (row stride D*2 bytes):
 // chunk' = chunk ^ (row & 7).  Applied identically on the staging write and
@@ -62,7 +63,7 @@ struct FwdLds {
 // forward kernel
 // ---------------------------------------------------------------------------
 template <int D>
-__global__ __launch_bounds__(512, 1) void attn_fwd_kernel(FwdParams p) {
+__global__ __launch_bounds__(NTHREADS, 1) void attn_fwd_kernel(FwdParams p) {
     static_assert(D % 32 == 0);
     constexpr int DBLK = D / 32;     // 32-d output blocks
     constexpr int KSTEPS = D / 16;   // QK^T k-steps
@@ -100,7 +101,9 @@ __global__ __launch_bounds__(512, 1) void attn_fwd_kernel(FwdParams p) {
     #pragma unroll
     for (int db = 0; db < DBLK; ++db) o_acc[db] = f32x16{};
 
-    if (!p.is_first) {            // resume from a previous ring hop
+    const bool split_mode = p.kv_split > 1;
+    const int zsplit = blockIdx.z;
+    if (!p.is_first && !split_mode) {   // resume from a previous ring hop
         const float* mrow = p.m + ((long)b * p.h + h) * p.nq;
         const float* lrow = p.l + ((long)b * p.h + h) * p.nq;
         m_run = mrow[i_clamped];
@@ -115,67 +118,119 @@ __global__ __launch_bounds__(512, 1) void attn_fwd_kernel(FwdParams p) {
             }
     }
 
-    // ---- causal / lookback tile bounds for this workgroup
+    // ---- causal / lookback tile bounds for this workgroup: the masked
+    // region is contiguous, so the tile range is computed once (keeps the
+    // staging pipeline branch-free)
     const long wg_i_min = (long)qtile * QROWS_WG;
     const long wg_i_max = min((long)(qtile + 1) * QROWS_WG, p.nq) - 1;
-
     const int num_kv_tiles = (int)((p.nk + KVBLK - 1) / KVBLK);
 
-    for (int t = 0; t < num_kv_tiles; ++t) {
+    int t_lo = 0, t_hi = num_kv_tiles;
+    if (p.causal) {
+        long last = wg_i_max + p.diag;               // largest attendable j
+        t_hi = last < 0 ? 0 : min((long)num_kv_tiles, last / KVBLK + 1);
+    }
+    if (p.has_win) {
+        // tile t attends iff j0 + KVBLK - 1 >= wg_i_min - win
+        long x = wg_i_min - p.win - KVBLK + 1;
+        t_lo = x <= 0 ? 0 : (int)((x + KVBLK - 1) / KVBLK);
+        if (t_lo > t_hi) t_lo = t_hi;
+    }
+    if (split_mode) {                     // this z's tile-aligned kv share
+        int per_split = (num_kv_tiles + p.kv_split - 1) / p.kv_split;
+        t_lo = max(t_lo, zsplit * per_split);
+        t_hi = min(t_hi, (zsplit + 1) * per_split);
+        if (t_lo > t_hi) t_lo = t_hi;
+    }
+
+    // ---- T14 async-stage split: per-thread staging registers
+    constexpr int CH_PER_ROW = D * 2 / 16;
+    constexpr int KCHUNKS = KVBLK * CH_PER_ROW;          // K-tile 16B chunks
+    constexpr int KREGS = (KCHUNKS + NTHREADS - 1) / NTHREADS;
+    constexpr int VPAIRS = (KVBLK / 2) * (D / 8);
+    constexpr int VREGS = (VPAIRS + NTHREADS - 1) / NTHREADS;
+    const __bf16* kbase = (const __bf16*)p.k + ((long)b * p.nk) * p.hk * D + (long)hk * D;
+    const __bf16* vbase = (const __bf16*)p.v + ((long)b * p.nk) * p.hk * D + (long)hk * D;
+    const unsigned char* mbase = p.kmask ? (const unsigned char*)p.kmask + (long)b * p.nk : nullptr;
+
+    uint4 kst[KREGS];
+    bf16x8 vsta[VREGS], vstb[VREGS];
+    unsigned char mst = 1;
+
+    auto load_tile = [&](int t) {
         const long j0 = (long)t * KVBLK;
         const long jmax = min(j0 + KVBLK, p.nk) - 1;
+        #pragma unroll
+        for (int r = 0; r < KREGS; ++r) {
+            int c = tid + r * NTHREADS;
+            if (c < KCHUNKS) {
+                long j = j0 + c / CH_PER_ROW;
+                int ch = c % CH_PER_ROW;
+                kst[r] = (j <= jmax) ? *(const uint4*)(kbase + j * p.hk * D + ch * 8)
+                                     : uint4{0, 0, 0, 0};
+            }
+        }
+        #pragma unroll
+        for (int r = 0; r < VREGS; ++r) {
+            int c = tid + r * NTHREADS;
+            if (c < VPAIRS) {
+                int jp = c % (KVBLK / 2);
+                int d0 = (c / (KVBLK / 2)) * 8;
+                long ja = j0 + jp * 2, jb_ = ja + 1;
+                vsta[r] = (ja <= jmax) ? *(const bf16x8*)(vbase + ja * p.hk * D + d0) : bf16x8{};
+                vstb[r] = (jb_ <= jmax) ? *(const bf16x8*)(vbase + jb_ * p.hk * D + d0) : bf16x8{};
+            }
+        }
+        if (mbase && tid < KVBLK)
+            mst = (j0 + tid <= jmax) ? mbase[j0 + tid] : 0;
+    };
 
-        if (p.causal && j0 > wg_i_max + p.diag) break;              // all future
-        if (p.has_win && (wg_i_min - jmax) > p.win) continue;      // all beyond window
+    auto write_tile = [&]() {
+        #pragma unroll
+        for (int r = 0; r < KREGS; ++r) {
+            int c = tid + r * NTHREADS;
+            if (c < KCHUNKS) {
+                int row = c / CH_PER_ROW, ch = c % CH_PER_ROW;
+                *(uint4*)(lds.k + row * D + swz(row, ch) * 8) = kst[r];
+            }
+        }
+        #pragma unroll
+        for (int r = 0; r < VREGS; ++r) {
+            int c = tid + r * NTHREADS;
+            if (c < VPAIRS) {
+                int jp = c % (KVBLK / 2);
+                int d0 = (c / (KVBLK / 2)) * 8;
+                #pragma unroll
+                for (int e = 0; e < 8; ++e) {
+                    int d = d0 + e;
+                    int byte_off = d * KVBLK * 2 + ((jp * 4) ^ ((d & 7) << 4));
+                    __bf16 pair[2] = {vsta[r][e], vstb[r][e]};
+                    *(uint32_t*)((char*)lds.vt + byte_off) = *(uint32_t*)pair;
+                }
+            }
+        }
+        if (mbase && tid < KVBLK) lds.kmask[tid] = mst;
+    };
+
+    if (t_lo < t_hi) load_tile(t_lo);
+
+    for (int t = t_lo; t < t_hi; ++t) {
+        const long j0 = (long)t * KVBLK;
+        const long jmax = min(j0 + KVBLK, p.nk) - 1;
         const bool full_tile =
             (jmax - j0 == KVBLK - 1) &&
             (!p.causal || jmax <= wg_i_min + p.diag) &&
             (!p.has_win || (wg_i_max - j0) <= p.win) &&
             !p.kmask;
 
-        // ---- stage K tile -> lds.k [kv][D] (swizzled 16B chunks)
-        {
-            constexpr int CH_PER_ROW = D * 2 / 16;                  // 16B chunks per row
-            constexpr int CHUNKS = KVBLK * CH_PER_ROW;              // total (<=1024)
-            const __bf16* kbase = (const __bf16*)p.k + ((long)b * p.nk) * p.hk * D + (long)hk * D;
-            for (int c = tid; c < CHUNKS; c += 512) {
-                int row = c / CH_PER_ROW, ch = c % CH_PER_ROW;
-                long j = j0 + row;
-                uint4 val;
-                if (j <= jmax) val = *(const uint4*)(kbase + j * p.hk * D + ch * 8);
-                else val = uint4{0, 0, 0, 0};
-                *(uint4*)(lds.k + row * D + swz(row, ch) * 8) = val;
-            }
-            // ---- stage V^T tile -> lds.vt [d][kv] (swizzled)
-            // thread handles 2 kv rows x 8 d elements -> 8 ds_write_b32 pairs
-            const __bf16* vbase = (const __bf16*)p.v + ((long)b * p.nk) * p.hk * D + (long)hk * D;
-            constexpr int PAIRS = (KVBLK / 2) * (D / 8);            // threads needed
-            for (int c = tid; c < PAIRS; c += 512) {
-                int jp = c % (KVBLK / 2);                            // kv pair index
-                int d0 = (c / (KVBLK / 2)) * 8;
-                long ja = j0 + jp * 2, jb_ = ja + 1;
-                bf16x8 va = (ja <= jmax) ? *(const bf16x8*)(vbase + ja * p.hk * D + d0) : bf16x8{};
-                bf16x8 vb = (jb_ <= jmax) ? *(const bf16x8*)(vbase + jb_ * p.hk * D + d0) : bf16x8{};
-                #pragma unroll
-                for (int e = 0; e < 8; ++e) {
-                    int d = d0 + e;
-                    // vt row = d (KVBLK*2 bytes per row, 8 chunks of 16B)
-                    int byte_off = d * KVBLK * 2 + ((jp * 2 * 2) ^ ((d & 7) << 4));
-                    __bf16 pair[2] = {va[e], vb[e]};
-                    *(uint32_t*)((char*)lds.vt + byte_off) = *(uint32_t*)pair;
-                }
-            }
-            if (p.kmask) {
-                const unsigned char* mb = (const unsigned char*)p.kmask + (long)b * p.nk;
-                for (int c = tid; c < KVBLK; c += 512)
-                    lds.kmask[c] = (j0 + c <= jmax) ? mb[j0 + c] : 0;
-            }
-        }
+        write_tile();
         __syncthreads();
+        if (t + 1 < t_hi) load_tile(t + 1);   // HBM latency hides under compute
 
         // ---- QK^T: S^T[kv][q] for kv blocks {0,1} (32 rows each)
         f32x16 s[2];
         s[0] = f32x16{}; s[1] = f32x16{};
+        __builtin_amdgcn_s_setprio(1);
         #pragma unroll
         for (int kb = 0; kb < 2; ++kb) {
             int krow = kb * 32 + l31;
@@ -186,6 +241,7 @@ __global__ __launch_bounds__(512, 1) void attn_fwd_kernel(FwdParams p) {
                 s[kb] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf, qf[ks], s[kb], 0, 0, 0);
             }
         }
+        __builtin_amdgcn_s_setprio(0);
 
         // ---- scale, clamp, mask in place; each lane: 32 scores of q row i
         float smax = MASK_VALUE_F;
@@ -254,6 +310,7 @@ __global__ __launch_bounds__(512, 1) void attn_fwd_kernel(FwdParams p) {
             for (int r = 0; r < 16; ++r) o_acc[db][r] *= alpha;
 
         // ---- PV: O^T[d][q] += V^T[d][kv] P^T[kv][q]
+        __builtin_amdgcn_s_setprio(1);
         #pragma unroll
         for (int db = 0; db < DBLK; ++db) {
             int drow = db * 32 + l31;                                // V^T row for A operand
@@ -265,11 +322,29 @@ __global__ __launch_bounds__(512, 1) void attn_fwd_kernel(FwdParams p) {
                     vf, *(const bf16x8*)frag[ks], o_acc[db], 0, 0, 0);
             }
         }
+        __builtin_amdgcn_s_setprio(0);
         __syncthreads();
     }
 
     // ---- epilogue
     if (!row_valid) return;
+
+    if (split_mode) {
+        // write this split's unnormalized partial (merged by attn_fwd_merge)
+        const long part = (long)zsplit * p.b * p.h;
+        float* mrow = p.m + (part + (long)b * p.h + h) * p.nq;
+        float* lrow = p.l + (part + (long)b * p.h + h) * p.nq;
+        if (lhi == 0) { mrow[i] = m_run; lrow[i] = l_run; }
+        float* oa = p.o_acc + (part + (long)b * p.h + h) * D * p.nq;
+        #pragma unroll
+        for (int db = 0; db < DBLK; ++db)
+            #pragma unroll
+            for (int r = 0; r < 16; ++r) {
+                int d = db * 32 + (r & 3) + 8 * (r >> 2) + 4 * lhi;
+                oa[(long)d * p.nq + i] = o_acc[db][r];
+            }
+        return;
+    }
 
     if (p.is_last) {
         float l_safe = fmaxf(l_run, 1e-38f);
@@ -305,9 +380,85 @@ __global__ __launch_bounds__(512, 1) void attn_fwd_kernel(FwdParams p) {
     }
 }
 
+// ---------------------------------------------------------------------------
+// split-merge: combine kv_split partials (and optionally the running ring
+// accumulator) with the standard online-softmax merge; memory-bound.
+// One thread per q row; d-loop vectorized along n across the wave.
+// ---------------------------------------------------------------------------
+template <int D>
+__global__ __launch_bounds__(256) void attn_fwd_merge_kernel(FwdMergeParams p) {
+    const long row = (long)blockIdx.x * 256 + threadIdx.x;   // global (b*h*n) row
+    const long total = (long)p.b * p.h * p.nq;
+    if (row >= total) return;
+    const long bh = row / p.nq;
+    const long n = row % p.nq;
+    const long b = bh / p.h, h = bh % p.h;
+
+    const long bhn = bh * p.nq + n;
+    const long stride_bh = (long)p.b * p.h;
+
+    float m_tot = MASK_VALUE_F;
+    if (!p.is_first) m_tot = p.m[bhn];
+    for (int s = 0; s < p.splits; ++s)
+        m_tot = fmaxf(m_tot, p.m_part[s * stride_bh * p.nq + bhn]);
+
+    float l_tot = 0.f;
+    float alpha_prev = 0.f;
+    if (!p.is_first) {
+        alpha_prev = __expf(p.m[bhn] - m_tot);
+        l_tot = p.l[bhn] * alpha_prev;
+    }
+    float alpha_s[16];                      // splits <= 16
+    for (int s = 0; s < p.splits; ++s) {
+        alpha_s[s] = __expf(p.m_part[s * stride_bh * p.nq + bhn] - m_tot);
+        l_tot += p.l_part[s * stride_bh * p.nq + bhn] * alpha_s[s];
+    }
+
+    const long od_base = (bh * D) * p.nq + n;     // (b,h,d,n) index at d=0
+    if (p.is_last) {
+        float l_safe = fmaxf(l_tot, 1e-38f);
+        float inv = 1.f / l_safe;
+        __bf16* ob = (__bf16*)p.out + ((long)b * p.nq + n) * p.h * D + h * D;
+        for (int d = 0; d < D; d += 4) {
+            float acc[4];
+            #pragma unroll
+            for (int e = 0; e < 4; ++e) {
+                float v = p.is_first ? 0.f : p.o_acc[od_base + (long)(d + e) * p.nq] * alpha_prev;
+                for (int s = 0; s < p.splits; ++s)
+                    v += p.o_part[s * stride_bh * D * p.nq + od_base + (long)(d + e) * p.nq]
+                         * alpha_s[s];
+                acc[e] = v * inv;
+            }
+            __bf16 four[4] = {(__bf16)acc[0], (__bf16)acc[1], (__bf16)acc[2], (__bf16)acc[3]};
+            *(uint2*)(ob + d) = *(uint2*)four;
+        }
+        p.lse[bhn] = __logf(l_safe) + m_tot;
+    } else {
+        for (int d = 0; d < D; ++d) {
+            float v = p.is_first ? 0.f : p.o_acc[od_base + (long)d * p.nq] * alpha_prev;
+            for (int s = 0; s < p.splits; ++s)
+                v += p.o_part[s * stride_bh * D * p.nq + od_base + (long)d * p.nq] * alpha_s[s];
+            p.o_acc[od_base + (long)d * p.nq] = v;
+        }
+        p.m[bhn] = m_tot;
+        p.l[bhn] = l_tot;
+    }
+}
+
+void launch_attn_fwd_merge(const FwdMergeParams& p, int head_dim, hipStream_t stream) {
+    long total = (long)p.b * p.h * p.nq;
+    dim3 grid((total + 255) / 256);
+    dim3 block(256);
+    if (head_dim == 64) {
+        hipLaunchKernelGGL(attn_fwd_merge_kernel<64>, grid, block, 0, stream, p);
+    } else {
+        hipLaunchKernelGGL(attn_fwd_merge_kernel<128>, grid, block, 0, stream, p);
+    }
+}
+
 void launch_attn_fwd(const FwdParams& p, int head_dim, hipStream_t stream) {
-    dim3 grid((p.nq + QROWS_WG - 1) / QROWS_WG, p.b * p.h);
-    dim3 block(512);
+    dim3 grid((p.nq + QROWS_WG - 1) / QROWS_WG, p.b * p.h, p.kv_split > 1 ? p.kv_split : 1);
+    dim3 block(NTHREADS);
     if (head_dim == 64) {
         hipLaunchKernelGGL(attn_fwd_kernel<64>, grid, block, 0, stream, p);
     } else if (head_dim == 128) {

@@ -27,7 +27,7 @@ void attn_fwd(
     std::optional<at::Tensor> lse,
     double scale, bool causal, int64_t diag, int64_t win, bool has_win,
     bool softclamp, double softclamp_value,
-    bool is_first, bool is_last) {
+    bool is_first, bool is_last, int64_t kv_split) {
     CHECK_BF16_CONTIG(q); CHECK_BF16_CONTIG(k); CHECK_BF16_CONTIG(v);
     TORCH_CHECK(q.dim() == 4 && k.dim() == 4 && v.dim() == 4, "q/k/v must be (B,N,H,D)");
     const int64_t B = q.size(0), Nq = q.size(1), H = q.size(2), D = q.size(3);
@@ -44,7 +44,16 @@ void attn_fwd(
         TORCH_CHECK(kmask->size(0) == B && kmask->size(1) == Nk);
         p.kmask = kmask->data_ptr();
     }
-    if (!(is_first && is_last)) {
+    if (kv_split > 1) {
+        TORCH_CHECK(o_acc && m && l, "split launches need partial o/m/l buffers");
+        CHECK_F32_CONTIG((*o_acc)); CHECK_F32_CONTIG((*m)); CHECK_F32_CONTIG((*l));
+        TORCH_CHECK(kv_split <= 16, "kv_split capped at 16");
+        TORCH_CHECK(o_acc->numel() == kv_split * B * H * D * Nq
+                    && m->numel() == kv_split * B * H * Nq);
+        p.o_acc = o_acc->data_ptr<float>();
+        p.m = m->data_ptr<float>();
+        p.l = l->data_ptr<float>();
+    } else if (!(is_first && is_last)) {
         TORCH_CHECK(o_acc && m && l, "multi-pass launches need o_acc/m/l scratch");
         CHECK_F32_CONTIG((*o_acc)); CHECK_F32_CONTIG((*m)); CHECK_F32_CONTIG((*l));
         TORCH_CHECK(o_acc->numel() == B * H * D * Nq && m->numel() == B * H * Nq);
@@ -52,7 +61,7 @@ void attn_fwd(
         p.m = m->data_ptr<float>();
         p.l = l->data_ptr<float>();
     }
-    if (is_last) {
+    if (is_last && kv_split <= 1) {
         TORCH_CHECK(out && lse, "last pass needs out/lse");
         CHECK_BF16_CONTIG((*out)); CHECK_F32_CONTIG((*lse));
         TORCH_CHECK(out->sizes() == q.sizes() && lse->numel() == B * H * Nq);
@@ -65,9 +74,42 @@ void attn_fwd(
     p.softclamp = softclamp; p.softclamp_value = (float)softclamp_value;
     p.causal = causal; p.diag = diag; p.win = win; p.has_win = has_win;
     p.is_first = is_first; p.is_last = is_last;
+    p.kv_split = (int)kv_split;
 
     launch_attn_fwd(p, (int)D, at::hip::getCurrentHIPStream());
     TORCH_CHECK(hipGetLastError() == hipSuccess, "attn_fwd launch failed");
+}
+
+void attn_fwd_merge(
+    at::Tensor o_part, at::Tensor m_part, at::Tensor l_part,
+    std::optional<at::Tensor> o_acc, std::optional<at::Tensor> m,
+    std::optional<at::Tensor> l,
+    std::optional<at::Tensor> out, std::optional<at::Tensor> lse,
+    int64_t splits, int64_t B, int64_t H, int64_t D, int64_t Nq,
+    bool is_first, bool is_last) {
+    CHECK_F32_CONTIG(o_part); CHECK_F32_CONTIG(m_part); CHECK_F32_CONTIG(l_part);
+    TORCH_CHECK(splits >= 1 && splits <= 16);
+    TORCH_CHECK(o_part.numel() == splits * B * H * D * Nq);
+    FwdMergeParams p{};
+    p.o_part = o_part.data_ptr<float>();
+    p.m_part = m_part.data_ptr<float>();
+    p.l_part = l_part.data_ptr<float>();
+    if (!(is_first && is_last)) {
+        TORCH_CHECK(o_acc && m && l, "ring merge needs running o_acc/m/l");
+        p.o_acc = o_acc->data_ptr<float>();
+        p.m = m->data_ptr<float>();
+        p.l = l->data_ptr<float>();
+    }
+    if (is_last) {
+        TORCH_CHECK(out && lse);
+        CHECK_BF16_CONTIG((*out)); CHECK_F32_CONTIG((*lse));
+        p.out = out->data_ptr();
+        p.lse = lse->data_ptr<float>();
+    }
+    p.splits = (int)splits; p.b = (int)B; p.h = (int)H; p.nq = Nq;
+    p.is_first = is_first; p.is_last = is_last;
+    launch_attn_fwd_merge(p, (int)D, at::hip::getCurrentHIPStream());
+    TORCH_CHECK(hipGetLastError() == hipSuccess, "attn_fwd_merge launch failed");
 }
 
 void attn_bwd(
@@ -76,7 +118,7 @@ void attn_bwd(
     at::Tensor lse, at::Tensor delta,
     at::Tensor dq, at::Tensor dk, at::Tensor dv,
     double scale, bool causal, int64_t diag, int64_t win, bool has_win,
-    bool softclamp, double softclamp_value, bool accumulate) {
+    bool softclamp, double softclamp_value, bool accumulate, int64_t split) {
     CHECK_BF16_CONTIG(q); CHECK_BF16_CONTIG(k); CHECK_BF16_CONTIG(v); CHECK_BF16_CONTIG(dout);
     CHECK_F32_CONTIG(lse); CHECK_F32_CONTIG(delta);
     CHECK_F32_CONTIG(dq); CHECK_F32_CONTIG(dk); CHECK_F32_CONTIG(dv);
@@ -101,6 +143,7 @@ void attn_bwd(
     p.softclamp = softclamp; p.softclamp_value = (float)softclamp_value;
     p.causal = causal; p.diag = diag; p.win = win; p.has_win = has_win;
     p.accumulate = accumulate;
+    p.split = (int)split;
 
     launch_attn_bwd(p, (int)D, at::hip::getCurrentHIPStream());
     TORCH_CHECK(hipGetLastError() == hipSuccess, "attn_bwd launch failed");
@@ -127,6 +170,7 @@ std::vector<at::Tensor> decode_partial(at::Tensor q, at::Tensor k, at::Tensor v)
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
     mod.def("attn_fwd", &ring_attn::attn_fwd, "CDNA4 flash attention forward (resumable)");
+    mod.def("attn_fwd_merge", &ring_attn::attn_fwd_merge, "merge kv-split partials");
     mod.def("attn_bwd", &ring_attn::attn_bwd, "CDNA4 flash attention backward");
     mod.def("decode_partial", &ring_attn::decode_partial, "CDNA4 single-query decode partial");
 }

@@ -85,12 +85,23 @@ class RingFlashAttentionHIPFunction(Function):
         out = torch.empty_like(qb)
         lse = torch.empty(b, h, n, device=q.device, dtype=torch.float32)
 
+        # kv-split: fill the 256 CUs when the natural grid (q-tiles x b*h) is
+        # small (flash-decoding-style partials + a softmax-correct merge)
+        qtiles = (n + 255) // 256
+        kv_tiles = (n + 63) // 64
+        kv_split = min(16, kv_tiles, max(1, 192 // max(1, qtiles * b * h)))
+
         multi = hops > 1
         o_acc = m = l = None
         if multi:
             o_acc = torch.empty(b, h, d, n, device=q.device, dtype=torch.float32)
             m = torch.empty(b, h, n, device=q.device, dtype=torch.float32)
             l = torch.empty(b, h, n, device=q.device, dtype=torch.float32)
+        o_part = m_part = l_part = None
+        if kv_split > 1:
+            o_part = torch.empty(kv_split, b, h, d, n, device=q.device, dtype=torch.float32)
+            m_part = torch.empty(kv_split, b, h, n, device=q.device, dtype=torch.float32)
+            l_part = torch.empty(kv_split, b, h, n, device=q.device, dtype=torch.float32)
 
         # which hops actually compute (host-side skip of fully-masked shards)
         rq = topo.ring_rank
@@ -113,11 +124,23 @@ class RingFlashAttentionHIPFunction(Function):
                 continue
             kv_t = tensors[0]
             mk = tensors[1] if mask_u8 is not None else None
-            ext.attn_fwd(qb, kv_t[0], kv_t[1], mk,
-                         o_acc, m, l, out, lse,
-                         scale, causal, diag, win, lookback is not None,
-                         softclamp_qk_sim, softclamp_value,
-                         info.hop == first_active, info.hop == last_active)
+            is_f = info.hop == first_active
+            is_l = info.hop == last_active
+            if kv_split > 1:
+                ext.attn_fwd(qb, kv_t[0], kv_t[1], mk,
+                             o_part, m_part, l_part, None, None,
+                             scale, causal, diag, win, lookback is not None,
+                             softclamp_qk_sim, softclamp_value,
+                             is_f, is_l, kv_split)
+                ext.attn_fwd_merge(o_part, m_part, l_part, o_acc, m, l,
+                                   out if is_l else None, lse if is_l else None,
+                                   kv_split, b, h, d, n, is_f, is_l)
+            else:
+                ext.attn_fwd(qb, kv_t[0], kv_t[1], mk,
+                             o_acc, m, l, out, lse,
+                             scale, causal, diag, win, lookback is not None,
+                             softclamp_qk_sim, softclamp_value,
+                             is_f, is_l, 1)
 
         ctx.save_for_backward(qb, kb, vb, out, lse,
                               mask_u8 if mask_u8 is not None else torch.empty(0))
@@ -146,6 +169,8 @@ class RingFlashAttentionHIPFunction(Function):
 
         dq = torch.zeros(b, n, h, d, device=qb.device, dtype=torch.float32)
         rq = topo.ring_rank
+        qtiles = (n + 255) // 256
+        bwd_split = min(8, max(1, 192 // max(1, qtiles * b * h)))
 
         kv = torch.stack((kb, vb))
         ring_tensors = (kv,) if mask_u8 is None else (kv, mask_u8)
@@ -168,7 +193,7 @@ class RingFlashAttentionHIPFunction(Function):
                 ext.attn_bwd(qb, kv_t[0], kv_t[1], dob, mk, lse, delta,
                              dq, dk_n, dv_n,
                              scale, causal, diag, win, lookback is not None,
-                             softclamp_qk_sim, softclamp_value, False)
+                             softclamp_qk_sim, softclamp_value, False, bwd_split)
             acc.step(contrib, info.is_last)
 
         dkv = acc.finish(hops)
