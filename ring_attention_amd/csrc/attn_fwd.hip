@@ -52,12 +52,14 @@ __device__ __forceinline__ int swz(int row, int chunk) { return chunk ^ (row & 7
 
 template <int D>
 struct FwdLds {
-    // K tile: [KVBLK][D] bf16, rows swizzled in 16B chunks
-    // V^T tile: [D][KVBLK] bf16, rows swizzled in 16B chunks (row = d)
-    __align__(16) __bf16 k[KVBLK * D];
-    __align__(16) __bf16 vt[D * KVBLK];
-    unsigned char kmask[KVBLK];
+    // double-buffered: K tile [kv][D] + V^T tile [d][kv], 16B-chunk swizzled
+    __align__(16) __bf16 k[2][KVBLK * D];
+    __align__(16) __bf16 vt[2][D * KVBLK];
+    unsigned char kmask[2][KVBLK];
 };
+
+static constexpr float LOG2E = 1.4426950408889634f;
+static constexpr float LN2 = 0.6931471805599453f;
 
 // ---------------------------------------------------------------------------
 // forward kernel
@@ -106,7 +108,7 @@ __global__ __launch_bounds__(NTHREADS, 1) void attn_fwd_kernel(FwdParams p) {
     if (!p.is_first && !split_mode) {   // resume from a previous ring hop
         const float* mrow = p.m + ((long)b * p.h + h) * p.nq;
         const float* lrow = p.l + ((long)b * p.h + h) * p.nq;
-        m_run = mrow[i_clamped];
+        m_run = mrow[i_clamped] * LOG2E;   // external contract is natural log
         l_run = lrow[i_clamped];
         const float* oa = p.o_acc + (((long)b * p.h + h) * D) * p.nq;
         #pragma unroll
@@ -185,13 +187,13 @@ __global__ __launch_bounds__(NTHREADS, 1) void attn_fwd_kernel(FwdParams p) {
             mst = (j0 + tid <= jmax) ? mbase[j0 + tid] : 0;
     };
 
-    auto write_tile = [&]() {
+    auto write_tile = [&](int par) {
         #pragma unroll
         for (int r = 0; r < KREGS; ++r) {
             int c = tid + r * NTHREADS;
             if (c < KCHUNKS) {
                 int row = c / CH_PER_ROW, ch = c % CH_PER_ROW;
-                *(uint4*)(lds.k + row * D + swz(row, ch) * 8) = kst[r];
+                *(uint4*)(lds.k[par] + row * D + swz(row, ch) * 8) = kst[r];
             }
         }
         #pragma unroll
@@ -205,16 +207,24 @@ __global__ __launch_bounds__(NTHREADS, 1) void attn_fwd_kernel(FwdParams p) {
                     int d = d0 + e;
                     int byte_off = d * KVBLK * 2 + ((jp * 4) ^ ((d & 7) << 4));
                     __bf16 pair[2] = {vsta[r][e], vstb[r][e]};
-                    *(uint32_t*)((char*)lds.vt + byte_off) = *(uint32_t*)pair;
+                    *(uint32_t*)((char*)lds.vt[par] + byte_off) = *(uint32_t*)pair;
                 }
             }
         }
-        if (mbase && tid < KVBLK) lds.kmask[tid] = mst;
+        if (mbase && tid < KVBLK) lds.kmask[par][tid] = mst;
     };
 
-    if (t_lo < t_hi) load_tile(t_lo);
+    // 3-deep pipeline over DOUBLE-buffered LDS: one barrier per tile; LDS
+    // writes and the next-next tile's HBM loads fully overlap the MFMAs
+    const float scale2 = p.scale * LOG2E;    // softmax runs in the exp2 domain
+    if (t_lo < t_hi) {
+        load_tile(t_lo);
+        write_tile(t_lo & 1);
+        if (t_lo + 1 < t_hi) load_tile(t_lo + 1);
+    }
 
     for (int t = t_lo; t < t_hi; ++t) {
+        const int par = t & 1;
         const long j0 = (long)t * KVBLK;
         const long jmax = min(j0 + KVBLK, p.nk) - 1;
         const bool full_tile =
@@ -223,9 +233,7 @@ __global__ __launch_bounds__(NTHREADS, 1) void attn_fwd_kernel(FwdParams p) {
             (!p.has_win || (wg_i_max - j0) <= p.win) &&
             !p.kmask;
 
-        write_tile();
         __syncthreads();
-        if (t + 1 < t_hi) load_tile(t + 1);   // HBM latency hides under compute
 
         // ---- QK^T: S^T[kv][q] for kv blocks {0,1} (32 rows each)
         f32x16 s[2];
@@ -236,27 +244,37 @@ __global__ __launch_bounds__(NTHREADS, 1) void attn_fwd_kernel(FwdParams p) {
             int krow = kb * 32 + l31;
             #pragma unroll
             for (int ks = 0; ks < KSTEPS; ++ks) {
-                int chunk = ks * 2 + lhi;                            // 16B chunk within K row
-                bf16x8 kf = *(const bf16x8*)(lds.k + krow * D + swz(krow, chunk) * 8);
+                int chunk = ks * 2 + lhi;
+                bf16x8 kf = *(const bf16x8*)(lds.k[par] + krow * D + swz(krow, chunk) * 8);
                 s[kb] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf, qf[ks], s[kb], 0, 0, 0);
             }
         }
         __builtin_amdgcn_s_setprio(0);
 
-        // ---- scale, clamp, mask in place; each lane: 32 scores of q row i
+        // stage tile t+1 into the other buffer while the MFMAs above retire
+        if (t + 1 < t_hi) write_tile(par ^ 1);
+        if (t + 2 < t_hi) load_tile(t + 2);
+
+        // ---- scale (exp2 domain), clamp, mask in place
         float smax = MASK_VALUE_F;
         #pragma unroll
         for (int kb = 0; kb < 2; ++kb)
             #pragma unroll
             for (int r = 0; r < 16; ++r) {
-                long j = j0 + kb * 32 + (r & 3) + 8 * (r >> 2) + 4 * lhi;
-                float x = s[kb][r] * p.scale;
-                if (p.softclamp) x = p.softclamp_value * tanhf(x / p.softclamp_value);
+                float x;
+                if (p.softclamp) {
+                    float xs = s[kb][r] * p.scale;
+                    xs = p.softclamp_value * tanhf(xs / p.softclamp_value);
+                    x = xs * LOG2E;
+                } else {
+                    x = s[kb][r] * scale2;
+                }
                 if (!full_tile) {
+                    long j = j0 + kb * 32 + (r & 3) + 8 * (r >> 2) + 4 * lhi;
                     bool ok = j <= jmax;
                     if (p.causal) ok = ok && (j <= i + p.diag);
                     if (p.has_win) ok = ok && (i - j <= p.win);
-                    if (p.kmask) ok = ok && lds.kmask[j - j0];
+                    if (p.kmask) ok = ok && lds.kmask[par][j - j0];
                     if (!ok) x = MASK_VALUE_F;
                 }
                 s[kb][r] = x;
@@ -264,23 +282,33 @@ __global__ __launch_bounds__(NTHREADS, 1) void attn_fwd_kernel(FwdParams p) {
             }
         smax = fmaxf(smax, __shfl_xor(smax, 32));
 
-        // ---- online softmax update
+        // ---- online softmax update (defer-max THR=0: exact — skip the O
+        // rescale whenever the running max did not grow on any lane)
         float m_new = fmaxf(m_run, smax);
-        float alpha = __expf(m_run - m_new);
+        const bool any_growth = !__all(smax <= m_run);
         float rowsum = 0.f;
         uint32_t pk[16];                                            // packed bf16 pairs
         #pragma unroll
         for (int x2 = 0; x2 < 16; ++x2) {
-            float e0 = __expf(s[x2 >> 3][(2 * x2) & 15] - m_new);
-            float e1 = __expf(s[x2 >> 3][(2 * x2 + 1) & 15] - m_new);
+            float e0 = __builtin_amdgcn_exp2f(s[x2 >> 3][(2 * x2) & 15] - m_new);
+            float e1 = __builtin_amdgcn_exp2f(s[x2 >> 3][(2 * x2 + 1) & 15] - m_new);
             rowsum += e0 + e1;
             union { __hip_bfloat162 h2; uint32_t u; } cvt;
             cvt.h2 = __float22bfloat162_rn(float2{e0, e1});
             pk[x2] = cvt.u;
         }
         rowsum += __shfl_xor(rowsum, 32);
-        l_run = l_run * alpha + rowsum;
-        m_run = m_new;
+        if (any_growth) {
+            float alpha = __builtin_amdgcn_exp2f(m_run - m_new);
+            l_run = l_run * alpha + rowsum;
+            #pragma unroll
+            for (int db = 0; db < DBLK; ++db)
+                #pragma unroll
+                for (int r = 0; r < 16; ++r) o_acc[db][r] *= alpha;
+            m_run = m_new;
+        } else {
+            l_run += rowsum;
+        }
 
         // ---- build PV B-operand fragments via permlane32_swap
         // pk[pb + x] holds the exp'd pair for kv rows (pattern):
@@ -292,7 +320,7 @@ __global__ __launch_bounds__(NTHREADS, 1) void attn_fwd_kernel(FwdParams p) {
         #pragma unroll
         for (int kb = 0; kb < 2; ++kb) {
             #pragma unroll
-            for (int half = 0; half < 2; ++half) {                   // k-step within block
+            for (int half = 0; half < 2; ++half) {
                 #pragma unroll
                 for (int c = 0; c < 2; ++c) {
                     u32x2 r = __builtin_amdgcn_permlane32_swap(
@@ -303,27 +331,20 @@ __global__ __launch_bounds__(NTHREADS, 1) void attn_fwd_kernel(FwdParams p) {
             }
         }
 
-        // ---- rescale O accumulator
-        #pragma unroll
-        for (int db = 0; db < DBLK; ++db)
-            #pragma unroll
-            for (int r = 0; r < 16; ++r) o_acc[db][r] *= alpha;
-
         // ---- PV: O^T[d][q] += V^T[d][kv] P^T[kv][q]
         __builtin_amdgcn_s_setprio(1);
         #pragma unroll
         for (int db = 0; db < DBLK; ++db) {
-            int drow = db * 32 + l31;                                // V^T row for A operand
+            int drow = db * 32 + l31;
             #pragma unroll
             for (int ks = 0; ks < 4; ++ks) {
-                int chunk = ks * 2 + lhi;                            // 16B chunk in vt row
-                bf16x8 vf = *(const bf16x8*)(lds.vt + drow * KVBLK + (swz(drow, chunk) & 7) * 8);
+                int chunk = ks * 2 + lhi;
+                bf16x8 vf = *(const bf16x8*)(lds.vt[par] + drow * KVBLK + (swz(drow, chunk) & 7) * 8);
                 o_acc[db] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
                     vf, *(const bf16x8*)frag[ks], o_acc[db], 0, 0, 0);
             }
         }
         __builtin_amdgcn_s_setprio(0);
-        __syncthreads();
     }
 
     // ---- epilogue
@@ -334,7 +355,7 @@ __global__ __launch_bounds__(NTHREADS, 1) void attn_fwd_kernel(FwdParams p) {
         const long part = (long)zsplit * p.b * p.h;
         float* mrow = p.m + (part + (long)b * p.h + h) * p.nq;
         float* lrow = p.l + (part + (long)b * p.h + h) * p.nq;
-        if (lhi == 0) { mrow[i] = m_run; lrow[i] = l_run; }
+        if (lhi == 0) { mrow[i] = m_run * LN2; lrow[i] = l_run; }
         float* oa = p.o_acc + (part + (long)b * p.h + h) * D * p.nq;
         #pragma unroll
         for (int db = 0; db < DBLK; ++db)
@@ -363,12 +384,12 @@ __global__ __launch_bounds__(NTHREADS, 1) void attn_fwd_kernel(FwdParams p) {
             }
         if (lhi == 0) {
             float* lsep = p.lse + ((long)b * p.h + h) * p.nq;
-            lsep[i] = __logf(l_safe) + m_run;
+            lsep[i] = __logf(l_safe) + m_run * LN2;
         }
     } else {
         float* mrow = p.m + ((long)b * p.h + h) * p.nq;
         float* lrow = p.l + ((long)b * p.h + h) * p.nq;
-        if (lhi == 0) { mrow[i] = m_run; lrow[i] = l_run; }
+        if (lhi == 0) { mrow[i] = m_run * LN2; lrow[i] = l_run; }
         float* oa = p.o_acc + (((long)b * p.h + h) * D) * p.nq;
         #pragma unroll
         for (int db = 0; db < DBLK; ++db)
