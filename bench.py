@@ -46,11 +46,11 @@ def main():
     distributed = world > 1
 
     on_gpu = torch.cuda.is_available()
+    if on_gpu:
+        torch.cuda.set_device(local_rank)   # before init_process_group (RCCL)
     if distributed:
         os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
         torch.distributed.init_process_group("nccl" if on_gpu else "gloo")
-    if on_gpu:
-        torch.cuda.set_device(local_rank)
         device = torch.device("cuda", local_rank)
         dtype = torch.bfloat16
         from ring_attention_amd.ops.ring_flash_hip import ring_flash_attn_hip_ as attn

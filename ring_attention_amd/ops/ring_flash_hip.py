@@ -224,3 +224,55 @@ def ring_flash_attn_hip_(
 def ring_flash_attn_hip(q, k, v, **kwargs) -> Tensor:
     out, _ = ring_flash_attn_hip_(q, k, v, **kwargs)
     return out
+
+
+class FlashAttnOffsetFunction(Function):
+    """Local (non-ring) flash attention with a global q-position offset:
+    attend(i, j) <=> j <= i + q_offset.  Used by the zig-zag CP scheme, where
+    each rank's q chunks are contiguous spans of global positions attending
+    the full (all-gathered) KV — one kernel call per span, no score-matrix
+    mask materialization (the reference built an explicit O(n*N) bool mask,
+    zig_zag_attention.py:123-139)."""
+
+    @staticmethod
+    def forward(ctx, q, k, v, q_offset, causal):
+        assert q.is_cuda
+        b, n, h, d = q.shape
+        scale = d ** -0.5
+        ext = hip_ext.require()
+        qb = q.to(torch.bfloat16).contiguous()
+        kb = k.to(torch.bfloat16).contiguous()
+        vb = v.to(torch.bfloat16).contiguous()
+        out = torch.empty_like(qb)
+        lse = torch.empty(b, h, n, device=q.device, dtype=torch.float32)
+        ext.attn_fwd(qb, kb, vb, None, None, None, None, out, lse,
+                     scale, causal, q_offset, 0, False, False, 50.0,
+                     True, True, 1)
+        ctx.save_for_backward(qb, kb, vb, out, lse)
+        ctx.meta = (q_offset, causal, q.dtype)
+        return out.to(q.dtype)
+
+    @staticmethod
+    def backward(ctx, do):
+        qb, kb, vb, out, lse = ctx.saved_tensors
+        q_offset, causal, in_dtype = ctx.meta
+        b, n, h, d = qb.shape
+        hk = kb.shape[2]
+        nk = kb.shape[1]
+        scale = d ** -0.5
+        ext = hip_ext.require()
+        dob = do.to(torch.bfloat16).contiguous()
+        delta = (dob.float() * out.float()).sum(dim=-1).permute(0, 2, 1).contiguous()
+        dq = torch.zeros(b, n, h, d, device=qb.device, dtype=torch.float32)
+        dk_n = torch.zeros(b, hk, nk, d, device=qb.device, dtype=torch.float32)
+        dv_n = torch.zeros(b, hk, d, nk, device=qb.device, dtype=torch.float32)
+        ext.attn_bwd(qb, kb, vb, dob, None, lse, delta, dq, dk_n, dv_n,
+                     scale, causal, q_offset, 0, False, False, 50.0, False, 1)
+        dk = dk_n.permute(0, 2, 1, 3).contiguous()
+        dv = dv_n.permute(0, 3, 1, 2).contiguous()
+        return (dq.to(in_dtype), dk.to(in_dtype), dv.to(in_dtype), None, None)
+
+
+def flash_attn_offset(q, k, v, q_offset=0, causal=True):
+    """q (b, n, h, d) attends k/v (b, nk, hk, d) with attend(i,j) <=> j <= i + q_offset."""
+    return FlashAttnOffsetFunction.apply(q, k, v, q_offset, causal)

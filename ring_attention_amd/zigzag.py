@@ -91,7 +91,19 @@ def zig_zag_attn(
     v: Tensor,                      # (b, hk, j, dv)
     dropout: float = 0.0,
     attn_mask: Tensor | None = None,  # bool, True = attend
+    causal: bool = False,
+    q_chunk_starts: tuple[int, int] | None = None,
+    kv_valid_len: int | None = None,   # unpadded global length (pad keys masked)
 ) -> Tensor:
+    """Zig-zag attention over all-gathered KV.
+
+    Fast path (GPU): when ``causal`` and ``q_chunk_starts`` (the global start
+    positions of this rank's two chunks, derivable from zig_zag_shard's
+    query_positions) are given, the CDNA4 flash kernel runs once per chunk
+    with an offset-causal mask — no O(n*N) mask tensor at all.  Otherwise the
+    portable masked path runs (parity with the reference API, which takes an
+    explicit attn_mask).
+    """
     heads, kv_heads = q.shape[1], k.shape[1]
     assert heads % kv_heads == 0
     groups = heads // kv_heads
@@ -99,6 +111,26 @@ def zig_zag_attn(
     gather_seq = AllGather(dim=-2)
     k, _ = gather_seq(k)
     v, _ = gather_seq(v)
+
+    if (causal and q_chunk_starts is not None and q.is_cuda
+            and dropout == 0.0 and attn_mask is None):
+        from .ops.ring_flash_hip import flash_attn_offset
+        n = q.shape[-2]
+        half = n // 2
+        # to (b, n, h, d) layout for the kernel; pad keys are a global-order
+        # suffix, so masking them is a slice (grads zero-pad automatically)
+        q_ = q.permute(0, 2, 1, 3)
+        k_ = k.permute(0, 2, 1, 3)
+        v_ = v.permute(0, 2, 1, 3)
+        if kv_valid_len is not None:
+            k_ = k_[:, :kv_valid_len]
+            v_ = v_[:, :kv_valid_len]
+        outs = []
+        for c, start in enumerate(q_chunk_starts):
+            qc = q_[:, c * half:(c + 1) * half]
+            outs.append(flash_attn_offset(qc, k_, v_, q_offset=start, causal=True))
+        out = torch.cat(outs, dim=1)
+        return out.permute(0, 2, 1, 3)
     if groups > 1:
         # repeat pattern parity with the reference: 'b h n d -> b (g h) n d'
         k = k.repeat(1, groups, 1, 1)
