@@ -51,6 +51,7 @@ def main():
     if distributed:
         os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
         torch.distributed.init_process_group("nccl" if on_gpu else "gloo")
+    if on_gpu:
         device = torch.device("cuda", local_rank)
         dtype = torch.bfloat16
         from ring_attention_amd.ops.ring_flash_hip import ring_flash_attn_hip_ as attn

@@ -41,6 +41,11 @@ namespace ring_attn {
 
 __device__ __forceinline__ int bswz(int row, int chunk) { return chunk ^ (row & 7); }
 
+__device__ __forceinline__ float bfast_tanhf(float x) {
+    float e = __builtin_amdgcn_exp2f(x * 2.885390081777927f);
+    return 1.f - 2.f * __builtin_amdgcn_rcpf(e + 1.f);
+}
+
 // ---------------------------------------------------------------------------
 // shared staging helpers (512-thread workgroups)
 // ---------------------------------------------------------------------------
@@ -99,7 +104,7 @@ struct DqLds {
     unsigned char kmask[DQ_KVBLK];
 };
 
-template <int D>
+template <int D, bool SOFTCLAMP>
 __global__ __launch_bounds__(512, 1) void attn_bwd_dq_kernel(BwdParams p) {
     constexpr int DBLK = D / 32;
     constexpr int KSTEPS = D / 16;
@@ -198,9 +203,11 @@ __global__ __launch_bounds__(512, 1) void attn_bwd_dq_kernel(BwdParams p) {
                 long j = j0 + kb * 32 + (r & 3) + 8 * (r >> 2) + 4 * lhi;
                 float x = s[kb][r] * p.scale;
                 float dtanh = 1.f;
-                if (p.softclamp) {
-                    x = p.softclamp_value * tanhf(x / p.softclamp_value);
-                    dtanh = 1.f - (x / p.softclamp_value) * (x / p.softclamp_value);
+                if constexpr (SOFTCLAMP) {
+                    float inv_v = __builtin_amdgcn_rcpf(p.softclamp_value);
+                    float th = bfast_tanhf(x * inv_v);
+                    x = p.softclamp_value * th;
+                    dtanh = 1.f - th * th;
                 }
                 bool ok = row_valid && j <= jmax;
                 if (!full_tile) {
@@ -276,7 +283,7 @@ struct DkvLds {
     float delta[QT];
 };
 
-template <int D, int QT>
+template <int D, int QT, bool SOFTCLAMP>
 __global__ __launch_bounds__(512, 1) void attn_bwd_dkv_kernel(BwdParams p) {
     static_assert(D % 32 == 0 && QT % 32 == 0);
     constexpr int DBLK = D / 32;
@@ -389,9 +396,11 @@ __global__ __launch_bounds__(512, 1) void attn_bwd_dkv_kernel(BwdParams p) {
                         long i = i0 + qloc;
                         float x = s2[r] * p.scale;
                         float dtanh = 1.f;
-                        if (p.softclamp) {
-                            x = p.softclamp_value * tanhf(x / p.softclamp_value);
-                            dtanh = 1.f - (x / p.softclamp_value) * (x / p.softclamp_value);
+                        if constexpr (SOFTCLAMP) {
+                            float inv_v = __builtin_amdgcn_rcpf(p.softclamp_value);
+                            float th = bfast_tanhf(x * inv_v);
+                            x = p.softclamp_value * th;
+                            dtanh = 1.f - th * th;
                         }
                         bool ok = col_valid && i <= imax;
                         if (!full_tile) {
@@ -483,11 +492,21 @@ void launch_attn_bwd(const BwdParams& p, int head_dim, hipStream_t stream) {
     dim3 grid_dq((p.nq + DQ_QROWS_WG - 1) / DQ_QROWS_WG, p.b * p.h, z);
     dim3 grid_dkv((p.nk + KVROWS_WG - 1) / KVROWS_WG, p.b * p.hk, z);
     if (head_dim == 64) {
-        hipLaunchKernelGGL((attn_bwd_dq_kernel<64>), grid_dq, block, 0, stream, p);
-        hipLaunchKernelGGL((attn_bwd_dkv_kernel<64, 64>), grid_dkv, block, 0, stream, p);
+        if (p.softclamp) {
+            hipLaunchKernelGGL((attn_bwd_dq_kernel<64, true>), grid_dq, block, 0, stream, p);
+            hipLaunchKernelGGL((attn_bwd_dkv_kernel<64, 64, true>), grid_dkv, block, 0, stream, p);
+        } else {
+            hipLaunchKernelGGL((attn_bwd_dq_kernel<64, false>), grid_dq, block, 0, stream, p);
+            hipLaunchKernelGGL((attn_bwd_dkv_kernel<64, 64, false>), grid_dkv, block, 0, stream, p);
+        }
     } else if (head_dim == 128) {
-        hipLaunchKernelGGL((attn_bwd_dq_kernel<128>), grid_dq, block, 0, stream, p);
-        hipLaunchKernelGGL((attn_bwd_dkv_kernel<128, 32>), grid_dkv, block, 0, stream, p);
+        if (p.softclamp) {
+            hipLaunchKernelGGL((attn_bwd_dq_kernel<128, true>), grid_dq, block, 0, stream, p);
+            hipLaunchKernelGGL((attn_bwd_dkv_kernel<128, 32, true>), grid_dkv, block, 0, stream, p);
+        } else {
+            hipLaunchKernelGGL((attn_bwd_dq_kernel<128, false>), grid_dq, block, 0, stream, p);
+            hipLaunchKernelGGL((attn_bwd_dkv_kernel<128, 32, false>), grid_dkv, block, 0, stream, p);
+        }
     } else {
         __builtin_trap();
     }

@@ -44,6 +44,7 @@ struct FwdParams {
     int is_last;            // normalize + write out/lse instead of o_acc/m/l
     int kv_split;           // >1: grid.z splits the kv range; o_acc/m/l hold
                             // kv_split partials (merged by attn_fwd_merge)
+    int ablate;             // diagnostics: 1 = stage first tile only
 };
 
 struct FwdMergeParams {

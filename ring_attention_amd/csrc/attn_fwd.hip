@@ -50,6 +50,13 @@ static constexpr int NTHREADS = WAVES * 64;
 // the fragment read, so the LDS image is consistent (both-sides rule).
 __device__ __forceinline__ int swz(int row, int chunk) { return chunk ^ (row & 7); }
 
+// fast tanh from builtins: tanh(x) = 1 - 2/(exp2(2x*log2e) + 1); avoids the
+// libm tanhf whose inlined body injects v_div_* sequences into the hot loop
+__device__ __forceinline__ float fast_tanhf(float x) {
+    float e = __builtin_amdgcn_exp2f(x * 2.885390081777927f);   // 2*log2(e)
+    return 1.f - 2.f * __builtin_amdgcn_rcpf(e + 1.f);
+}
+
 template <int D>
 struct FwdLds {
     // double-buffered: K tile [kv][D] + V^T tile [d][kv], 16B-chunk swizzled
@@ -64,7 +71,7 @@ static constexpr float LN2 = 0.6931471805599453f;
 // ---------------------------------------------------------------------------
 // forward kernel
 // ---------------------------------------------------------------------------
-template <int D>
+template <int D, bool SOFTCLAMP>
 __global__ __launch_bounds__(NTHREADS, 1) void attn_fwd_kernel(FwdParams p) {
     static_assert(D % 32 == 0);
     constexpr int DBLK = D / 32;     // 32-d output blocks
@@ -220,7 +227,8 @@ __global__ __launch_bounds__(NTHREADS, 1) void attn_fwd_kernel(FwdParams p) {
     if (t_lo < t_hi) {
         load_tile(t_lo);
         write_tile(t_lo & 1);
-        if (t_lo + 1 < t_hi) load_tile(t_lo + 1);
+        if (p.ablate == 1) write_tile((t_lo & 1) ^ 1);   // both buffers valid
+        if (t_lo + 1 < t_hi && p.ablate != 1) load_tile(t_lo + 1);
     }
 
     for (int t = t_lo; t < t_hi; ++t) {
@@ -252,8 +260,10 @@ __global__ __launch_bounds__(NTHREADS, 1) void attn_fwd_kernel(FwdParams p) {
         __builtin_amdgcn_s_setprio(0);
 
         // stage tile t+1 into the other buffer while the MFMAs above retire
-        if (t + 1 < t_hi) write_tile(par ^ 1);
-        if (t + 2 < t_hi) load_tile(t + 2);
+        if (p.ablate != 1) {
+            if (t + 1 < t_hi) write_tile(par ^ 1);
+            if (t + 2 < t_hi) load_tile(t + 2);
+        }
 
         // ---- scale (exp2 domain), clamp, mask in place
         float smax = MASK_VALUE_F;
@@ -262,9 +272,9 @@ __global__ __launch_bounds__(NTHREADS, 1) void attn_fwd_kernel(FwdParams p) {
             #pragma unroll
             for (int r = 0; r < 16; ++r) {
                 float x;
-                if (p.softclamp) {
-                    float xs = s[kb][r] * p.scale;
-                    xs = p.softclamp_value * tanhf(xs / p.softclamp_value);
+                if constexpr (SOFTCLAMP) {
+                    float xs = s[kb][r] * (p.scale * __builtin_amdgcn_rcpf(p.softclamp_value));
+                    xs = p.softclamp_value * fast_tanhf(xs);
                     x = xs * LOG2E;
                 } else {
                     x = s[kb][r] * scale2;
@@ -481,9 +491,11 @@ void launch_attn_fwd(const FwdParams& p, int head_dim, hipStream_t stream) {
     dim3 grid((p.nq + QROWS_WG - 1) / QROWS_WG, p.b * p.h, p.kv_split > 1 ? p.kv_split : 1);
     dim3 block(NTHREADS);
     if (head_dim == 64) {
-        hipLaunchKernelGGL(attn_fwd_kernel<64>, grid, block, 0, stream, p);
+        if (p.softclamp) hipLaunchKernelGGL((attn_fwd_kernel<64, true>), grid, block, 0, stream, p);
+        else hipLaunchKernelGGL((attn_fwd_kernel<64, false>), grid, block, 0, stream, p);
     } else if (head_dim == 128) {
-        hipLaunchKernelGGL(attn_fwd_kernel<128>, grid, block, 0, stream, p);
+        if (p.softclamp) hipLaunchKernelGGL((attn_fwd_kernel<128, true>), grid, block, 0, stream, p);
+        else hipLaunchKernelGGL((attn_fwd_kernel<128, false>), grid, block, 0, stream, p);
     } else {
         // unsupported head dim is a host-side error (checked in bindings)
         __builtin_trap();

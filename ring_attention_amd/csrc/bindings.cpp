@@ -27,7 +27,7 @@ void attn_fwd(
     std::optional<at::Tensor> lse,
     double scale, bool causal, int64_t diag, int64_t win, bool has_win,
     bool softclamp, double softclamp_value,
-    bool is_first, bool is_last, int64_t kv_split) {
+    bool is_first, bool is_last, int64_t kv_split, int64_t ablate) {
     CHECK_BF16_CONTIG(q); CHECK_BF16_CONTIG(k); CHECK_BF16_CONTIG(v);
     TORCH_CHECK(q.dim() == 4 && k.dim() == 4 && v.dim() == 4, "q/k/v must be (B,N,H,D)");
     const int64_t B = q.size(0), Nq = q.size(1), H = q.size(2), D = q.size(3);
@@ -75,6 +75,7 @@ void attn_fwd(
     p.causal = causal; p.diag = diag; p.win = win; p.has_win = has_win;
     p.is_first = is_first; p.is_last = is_last;
     p.kv_split = (int)kv_split;
+    p.ablate = (int)ablate;
 
     launch_attn_fwd(p, (int)D, at::hip::getCurrentHIPStream());
     TORCH_CHECK(hipGetLastError() == hipSuccess, "attn_fwd launch failed");

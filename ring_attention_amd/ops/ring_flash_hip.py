@@ -131,7 +131,7 @@ class RingFlashAttentionHIPFunction(Function):
                              o_part, m_part, l_part, None, None,
                              scale, causal, diag, win, lookback is not None,
                              softclamp_qk_sim, softclamp_value,
-                             is_f, is_l, kv_split)
+                             is_f, is_l, kv_split, 0)
                 ext.attn_fwd_merge(o_part, m_part, l_part, o_acc, m, l,
                                    out if is_l else None, lse if is_l else None,
                                    kv_split, b, h, d, n, is_f, is_l)
@@ -140,7 +140,7 @@ class RingFlashAttentionHIPFunction(Function):
                              o_acc, m, l, out, lse,
                              scale, causal, diag, win, lookback is not None,
                              softclamp_qk_sim, softclamp_value,
-                             is_f, is_l, 1)
+                             is_f, is_l, 1, 0)
 
         ctx.save_for_backward(qb, kb, vb, out, lse,
                               mask_u8 if mask_u8 is not None else torch.empty(0))
@@ -247,7 +247,7 @@ class FlashAttnOffsetFunction(Function):
         lse = torch.empty(b, h, n, device=q.device, dtype=torch.float32)
         ext.attn_fwd(qb, kb, vb, None, None, None, None, out, lse,
                      scale, causal, q_offset, 0, False, False, 50.0,
-                     True, True, 1)
+                     True, True, 1, 0)
         ctx.save_for_backward(qb, kb, vb, out, lse)
         ctx.meta = (q_offset, causal, q.dtype)
         return out.to(q.dtype)

@@ -155,9 +155,9 @@ def test_resume_contract_two_passes():
     # q positions are 0..n-1, shard0 cols 0..half-1 (diag = 0), shard1 cols
     # half.. (j_local <= i - half  => diag = -half)
     ext.attn_fwd(q, k0, v0, None, o_acc, m, l, out, lse,
-                 scale, True, 0, 0, False, False, 50.0, True, False, 1)
+                 scale, True, 0, 0, False, False, 50.0, True, False, 1, 0)
     ext.attn_fwd(q, k1, v1, None, o_acc, m, l, out, lse,
-                 scale, True, -half, 0, False, False, 50.0, False, True, 1)
+                 scale, True, -half, 0, False, False, 50.0, False, True, 1, 0)
 
     _, _, _, ref, ref_lse = _oracle(q, k, v, causal=True)
     err = (out.float().cpu() - ref).abs().max().item()
