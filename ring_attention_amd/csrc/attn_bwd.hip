@@ -486,27 +486,31 @@ __global__ __launch_bounds__(512, 1) void attn_bwd_dkv_kernel(BwdParams p) {
     }
 }
 
-void launch_attn_bwd(const BwdParams& p, int head_dim, hipStream_t stream) {
+void launch_attn_bwd_dq(const BwdParams& p, int head_dim, hipStream_t stream) {
     dim3 block(512);
     int z = p.split > 1 ? p.split : 1;
     dim3 grid_dq((p.nq + DQ_QROWS_WG - 1) / DQ_QROWS_WG, p.b * p.h, z);
+    if (head_dim == 64) {
+        if (p.softclamp) hipLaunchKernelGGL((attn_bwd_dq_kernel<64, true>), grid_dq, block, 0, stream, p);
+        else hipLaunchKernelGGL((attn_bwd_dq_kernel<64, false>), grid_dq, block, 0, stream, p);
+    } else if (head_dim == 128) {
+        if (p.softclamp) hipLaunchKernelGGL((attn_bwd_dq_kernel<128, true>), grid_dq, block, 0, stream, p);
+        else hipLaunchKernelGGL((attn_bwd_dq_kernel<128, false>), grid_dq, block, 0, stream, p);
+    } else {
+        __builtin_trap();
+    }
+}
+
+void launch_attn_bwd_dkv(const BwdParams& p, int head_dim, hipStream_t stream) {
+    dim3 block(512);
+    int z = p.split > 1 ? p.split : 1;
     dim3 grid_dkv((p.nk + KVROWS_WG - 1) / KVROWS_WG, p.b * p.hk, z);
     if (head_dim == 64) {
-        if (p.softclamp) {
-            hipLaunchKernelGGL((attn_bwd_dq_kernel<64, true>), grid_dq, block, 0, stream, p);
-            hipLaunchKernelGGL((attn_bwd_dkv_kernel<64, 64, true>), grid_dkv, block, 0, stream, p);
-        } else {
-            hipLaunchKernelGGL((attn_bwd_dq_kernel<64, false>), grid_dq, block, 0, stream, p);
-            hipLaunchKernelGGL((attn_bwd_dkv_kernel<64, 64, false>), grid_dkv, block, 0, stream, p);
-        }
+        if (p.softclamp) hipLaunchKernelGGL((attn_bwd_dkv_kernel<64, 64, true>), grid_dkv, block, 0, stream, p);
+        else hipLaunchKernelGGL((attn_bwd_dkv_kernel<64, 64, false>), grid_dkv, block, 0, stream, p);
     } else if (head_dim == 128) {
-        if (p.softclamp) {
-            hipLaunchKernelGGL((attn_bwd_dq_kernel<128, true>), grid_dq, block, 0, stream, p);
-            hipLaunchKernelGGL((attn_bwd_dkv_kernel<128, 32, true>), grid_dkv, block, 0, stream, p);
-        } else {
-            hipLaunchKernelGGL((attn_bwd_dq_kernel<128, false>), grid_dq, block, 0, stream, p);
-            hipLaunchKernelGGL((attn_bwd_dkv_kernel<128, 32, false>), grid_dkv, block, 0, stream, p);
-        }
+        if (p.softclamp) hipLaunchKernelGGL((attn_bwd_dkv_kernel<128, 32, true>), grid_dkv, block, 0, stream, p);
+        else hipLaunchKernelGGL((attn_bwd_dkv_kernel<128, 32, false>), grid_dkv, block, 0, stream, p);
     } else {
         __builtin_trap();
     }

@@ -92,7 +92,8 @@ struct BwdParams {
                             // accumulated with fp32 atomics instead of plain ops         // dk/dv: 0 = overwrite, 1 = add to existing
 };
 
-void launch_attn_bwd(const BwdParams& p, int head_dim, hipStream_t stream);
+void launch_attn_bwd_dq(const BwdParams& p, int head_dim, hipStream_t stream);
+void launch_attn_bwd_dkv(const BwdParams& p, int head_dim, hipStream_t stream);
 
 struct DecodeParams {
     const void* q;          // bf16 (B, H, 1, D)

@@ -42,7 +42,8 @@ namespace ring_attn {
 static constexpr int WAVES = 8;          // 8-wave WGs: the 2-waves/SIMD paired
 static constexpr int QROWS_WAVE = 32;    // regime (4-wave variant measured slower)
 static constexpr int QROWS_WG = WAVES * QROWS_WAVE;
-static constexpr int KVBLK = 64;
+static constexpr int KVBLK = 128;        // kv tile; NBLK 32-row MFMA blocks
+static constexpr int NBLK = KVBLK / 32;
 static constexpr int NTHREADS = WAVES * 64;
 
 // XOR swizzle of a 16-byte chunk index within a row (row stride D*2 bytes):
@@ -166,32 +167,49 @@ __global__ __launch_bounds__(NTHREADS, 1) void attn_fwd_kernel(FwdParams p) {
     bf16x8 vsta[VREGS], vstb[VREGS];
     unsigned char mst = 1;
 
-    auto load_tile = [&](int t) {
-        const long j0 = (long)t * KVBLK;
+    // running per-thread source pointers: load_tile is always called for
+    // consecutive tiles, so addresses advance by a constant — no per-tile
+    // 64-bit multiplies in the hot loop
+    const long kv_row_stride = (long)p.hk * D;
+    const long tile_stride = KVBLK * kv_row_stride;
+    const __bf16* kptr = kbase + (long)t_lo * tile_stride
+        + (tid / CH_PER_ROW) * kv_row_stride + (tid % CH_PER_ROW) * 8;
+    const __bf16* vptra = vbase + (long)t_lo * tile_stride
+        + ((tid % (KVBLK / 2)) * 2) * kv_row_stride + (tid / (KVBLK / 2)) * 8;
+    const __bf16* vptrb = vptra + kv_row_stride;
+    long j0_next = (long)t_lo * KVBLK;
+
+    auto load_tile = [&]() {
+        const long j0 = j0_next;
         const long jmax = min(j0 + KVBLK, p.nk) - 1;
+        const bool full = jmax - j0 == KVBLK - 1;
         #pragma unroll
         for (int r = 0; r < KREGS; ++r) {
             int c = tid + r * NTHREADS;
             if (c < KCHUNKS) {
-                long j = j0 + c / CH_PER_ROW;
-                int ch = c % CH_PER_ROW;
-                kst[r] = (j <= jmax) ? *(const uint4*)(kbase + j * p.hk * D + ch * 8)
-                                     : uint4{0, 0, 0, 0};
+                const __bf16* src = kptr + (long)(r * (NTHREADS / CH_PER_ROW)) * kv_row_stride;
+                kst[r] = (full || (j0 + c / CH_PER_ROW) <= jmax)
+                         ? *(const uint4*)src : uint4{0, 0, 0, 0};
             }
         }
         #pragma unroll
         for (int r = 0; r < VREGS; ++r) {
             int c = tid + r * NTHREADS;
             if (c < VPAIRS) {
-                int jp = c % (KVBLK / 2);
-                int d0 = (c / (KVBLK / 2)) * 8;
-                long ja = j0 + jp * 2, jb_ = ja + 1;
-                vsta[r] = (ja <= jmax) ? *(const bf16x8*)(vbase + ja * p.hk * D + d0) : bf16x8{};
-                vstb[r] = (jb_ <= jmax) ? *(const bf16x8*)(vbase + jb_ * p.hk * D + d0) : bf16x8{};
+                // extra r steps advance along d (same kv pair)
+                const __bf16* sa = vptra + r * (NTHREADS / (KVBLK / 2)) * 8;
+                const __bf16* sb = vptrb + r * (NTHREADS / (KVBLK / 2)) * 8;
+                long ja = j0 + (c % (KVBLK / 2)) * 2;
+                vsta[r] = (full || ja <= jmax) ? *(const bf16x8*)sa : bf16x8{};
+                vstb[r] = (full || ja + 1 <= jmax) ? *(const bf16x8*)sb : bf16x8{};
             }
         }
         if (mbase && tid < KVBLK)
             mst = (j0 + tid <= jmax) ? mbase[j0 + tid] : 0;
+        kptr += tile_stride;
+        vptra += tile_stride;
+        vptrb += tile_stride;
+        j0_next += KVBLK;
     };
 
     auto write_tile = [&](int par) {
@@ -225,10 +243,10 @@ __global__ __launch_bounds__(NTHREADS, 1) void attn_fwd_kernel(FwdParams p) {
     // writes and the next-next tile's HBM loads fully overlap the MFMAs
     const float scale2 = p.scale * LOG2E;    // softmax runs in the exp2 domain
     if (t_lo < t_hi) {
-        load_tile(t_lo);
+        load_tile();
         write_tile(t_lo & 1);
         if (p.ablate == 1) write_tile((t_lo & 1) ^ 1);   // both buffers valid
-        if (t_lo + 1 < t_hi && p.ablate != 1) load_tile(t_lo + 1);
+        if (t_lo + 1 < t_hi && p.ablate != 1) load_tile();
     }
 
     for (int t = t_lo; t < t_hi; ++t) {
@@ -243,12 +261,13 @@ __global__ __launch_bounds__(NTHREADS, 1) void attn_fwd_kernel(FwdParams p) {
 
         __syncthreads();
 
-        // ---- QK^T: S^T[kv][q] for kv blocks {0,1} (32 rows each)
-        f32x16 s[2];
-        s[0] = f32x16{}; s[1] = f32x16{};
+        // ---- QK^T: S^T[kv][q] for the NBLK 32-row kv blocks
+        f32x16 s[NBLK];
+        #pragma unroll
+        for (int kb = 0; kb < NBLK; ++kb) s[kb] = f32x16{};
         __builtin_amdgcn_s_setprio(1);
         #pragma unroll
-        for (int kb = 0; kb < 2; ++kb) {
+        for (int kb = 0; kb < NBLK; ++kb) {
             int krow = kb * 32 + l31;
             #pragma unroll
             for (int ks = 0; ks < KSTEPS; ++ks) {
@@ -262,13 +281,13 @@ __global__ __launch_bounds__(NTHREADS, 1) void attn_fwd_kernel(FwdParams p) {
         // stage tile t+1 into the other buffer while the MFMAs above retire
         if (p.ablate != 1) {
             if (t + 1 < t_hi) write_tile(par ^ 1);
-            if (t + 2 < t_hi) load_tile(t + 2);
+            if (t + 2 < t_hi) load_tile();
         }
 
         // ---- scale (exp2 domain), clamp, mask in place
         float smax = MASK_VALUE_F;
         #pragma unroll
-        for (int kb = 0; kb < 2; ++kb)
+        for (int kb = 0; kb < NBLK; ++kb)
             #pragma unroll
             for (int r = 0; r < 16; ++r) {
                 float x;
@@ -297,9 +316,9 @@ __global__ __launch_bounds__(NTHREADS, 1) void attn_fwd_kernel(FwdParams p) {
         float m_new = fmaxf(m_run, smax);
         const bool any_growth = !__all(smax <= m_run);
         float rowsum = 0.f;
-        uint32_t pk[16];                                            // packed bf16 pairs
+        uint32_t pk[NBLK * 8];                                      // packed bf16 pairs
         #pragma unroll
-        for (int x2 = 0; x2 < 16; ++x2) {
+        for (int x2 = 0; x2 < NBLK * 8; ++x2) {
             float e0 = __builtin_amdgcn_exp2f(s[x2 >> 3][(2 * x2) & 15] - m_new);
             float e1 = __builtin_amdgcn_exp2f(s[x2 >> 3][(2 * x2 + 1) & 15] - m_new);
             rowsum += e0 + e1;
@@ -326,9 +345,9 @@ __global__ __launch_bounds__(NTHREADS, 1) void attn_fwd_kernel(FwdParams p) {
         //   x=4:(16,17)+4lhi x=5:(18,19)+4lhi x=6:(24,25)+4lhi x=7:(26,27)+4lhi
         // B fragment for k-step needs u32 slot c = kv (8*lhi + 2c, +1), so:
         //   swap(pk[pb+h*4+c], pk[pb+h*4+c+2]) -> r0 = slot c, r1 = slot c+2
-        uint32_t frag[4][4];
+        uint32_t frag[NBLK * 2][4];
         #pragma unroll
-        for (int kb = 0; kb < 2; ++kb) {
+        for (int kb = 0; kb < NBLK; ++kb) {
             #pragma unroll
             for (int half = 0; half < 2; ++half) {
                 #pragma unroll
@@ -347,9 +366,9 @@ __global__ __launch_bounds__(NTHREADS, 1) void attn_fwd_kernel(FwdParams p) {
         for (int db = 0; db < DBLK; ++db) {
             int drow = db * 32 + l31;
             #pragma unroll
-            for (int ks = 0; ks < 4; ++ks) {
+            for (int ks = 0; ks < NBLK * 2; ++ks) {
                 int chunk = ks * 2 + lhi;
-                bf16x8 vf = *(const bf16x8*)(lds.vt[par] + drow * KVBLK + (swz(drow, chunk) & 7) * 8);
+                bf16x8 vf = *(const bf16x8*)(lds.vt[par] + drow * KVBLK + swz(drow, chunk) * 8);
                 o_acc[db] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
                     vf, *(const bf16x8*)frag[ks], o_acc[db], 0, 0, 0);
             }

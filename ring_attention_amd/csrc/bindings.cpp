@@ -119,7 +119,8 @@ void attn_bwd(
     at::Tensor lse, at::Tensor delta,
     at::Tensor dq, at::Tensor dk, at::Tensor dv,
     double scale, bool causal, int64_t diag, int64_t win, bool has_win,
-    bool softclamp, double softclamp_value, bool accumulate, int64_t split) {
+    bool softclamp, double softclamp_value, bool accumulate, int64_t split,
+    int64_t which) {   // 0 = both, 1 = dq only, 2 = dk/dv only
     CHECK_BF16_CONTIG(q); CHECK_BF16_CONTIG(k); CHECK_BF16_CONTIG(v); CHECK_BF16_CONTIG(dout);
     CHECK_F32_CONTIG(lse); CHECK_F32_CONTIG(delta);
     CHECK_F32_CONTIG(dq); CHECK_F32_CONTIG(dk); CHECK_F32_CONTIG(dv);
@@ -146,7 +147,10 @@ void attn_bwd(
     p.accumulate = accumulate;
     p.split = (int)split;
 
-    launch_attn_bwd(p, (int)D, at::hip::getCurrentHIPStream());
+    if (which == 0 || which == 1)
+        launch_attn_bwd_dq(p, (int)D, at::hip::getCurrentHIPStream());
+    if (which == 0 || which == 2)
+        launch_attn_bwd_dkv(p, (int)D, at::hip::getCurrentHIPStream());
     TORCH_CHECK(hipGetLastError() == hipSuccess, "attn_bwd launch failed");
 }
 

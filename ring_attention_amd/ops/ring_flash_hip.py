@@ -28,6 +28,15 @@ from ..parallel import RingAccumulator, RingTopology, all_ring_pass, is_distribu
 from . import hip_ext
 from .ring_flash import max_hops_for_lookback
 
+_side_stream = None
+
+
+def _get_side_stream():
+    global _side_stream
+    if _side_stream is None:
+        _side_stream = torch.cuda.Stream()
+    return _side_stream
+
 
 def _hop_geometry(rq: int, rk: int, n: int, ring_size: int, striped: bool,
                   causal: bool, lookback: int | None) -> tuple[bool, int, int]:
@@ -190,10 +199,28 @@ class RingFlashAttentionHIPFunction(Function):
             if not skip:
                 dk_n = contrib[0].view(b, hk, n, d)
                 dv_n = contrib[1].view(b, hk, d, n)
+                # dq (row-parallel) and dk/dv (column-parallel) are independent:
+                # run them on separate streams so they co-occupy the CUs
+                s2 = _get_side_stream()
+                ev = torch.cuda.Event()
+                ev.record()                      # delta/do ready on current stream
+                with torch.cuda.stream(s2):
+                    s2.wait_event(ev)
+                    ext.attn_bwd(qb, kv_t[0], kv_t[1], dob, mk, lse, delta,
+                                 dq, dk_n, dv_n,
+                                 scale, causal, diag, win, lookback is not None,
+                                 softclamp_qk_sim, softclamp_value, False,
+                                 bwd_split, 1)   # dq only
                 ext.attn_bwd(qb, kv_t[0], kv_t[1], dob, mk, lse, delta,
                              dq, dk_n, dv_n,
                              scale, causal, diag, win, lookback is not None,
-                             softclamp_qk_sim, softclamp_value, False, bwd_split)
+                             softclamp_qk_sim, softclamp_value, False,
+                             bwd_split, 2)       # dk/dv only
+                ev2 = torch.cuda.Event()
+                ev2.record(s2)
+                # the default stream (and thus the next hop's RCCL exchange,
+                # which waits on it) must not overwrite kv buffers dq reads
+                torch.cuda.current_stream().wait_event(ev2)
             acc.step(contrib, info.is_last)
 
         dkv = acc.finish(hops)
@@ -267,7 +294,7 @@ class FlashAttnOffsetFunction(Function):
         dk_n = torch.zeros(b, hk, nk, d, device=qb.device, dtype=torch.float32)
         dv_n = torch.zeros(b, hk, d, nk, device=qb.device, dtype=torch.float32)
         ext.attn_bwd(qb, kb, vb, dob, None, lse, delta, dq, dk_n, dv_n,
-                     scale, causal, q_offset, 0, False, False, 50.0, False, 1)
+                     scale, causal, q_offset, 0, False, False, 50.0, False, 1, 0)
         dk = dk_n.permute(0, 2, 1, 3).contiguous()
         dv = dv_n.permute(0, 3, 1, 2).contiguous()
         return (dq.to(in_dtype), dk.to(in_dtype), dv.to(in_dtype), None, None)
