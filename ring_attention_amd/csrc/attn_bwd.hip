@@ -145,6 +145,9 @@ __global__ __launch_bounds__(512, 1) void attn_bwd_dq_kernel(BwdParams p) {
 
     const long wg_i_min = (long)qtile * DQ_QROWS_WG;
     const long wg_i_max = min((long)(qtile + 1) * DQ_QROWS_WG, p.nq) - 1;
+    const long wg_q_min = wg_i_min * p.q_stride + p.diag;
+    const long wg_q_max = wg_i_max * p.q_stride + p.diag;
+    const long qpos_i = i * p.q_stride + p.diag;
     const int num_kv_tiles = (int)((p.nk + DQ_KVBLK - 1) / DQ_KVBLK);
     int zt_lo = 0, zt_hi = num_kv_tiles;
     if (p.split > 1) {                     // grid.z splits the kv walk
@@ -156,12 +159,12 @@ __global__ __launch_bounds__(512, 1) void attn_bwd_dq_kernel(BwdParams p) {
     for (int t = zt_lo; t < zt_hi; ++t) {
         const long j0 = (long)t * DQ_KVBLK;
         const long jmax = min(j0 + DQ_KVBLK, p.nk) - 1;
-        if (p.causal && j0 > wg_i_max + p.diag) break;
-        if (p.has_win && (wg_i_min - jmax) > p.win) continue;
+        if (p.causal && j0 > wg_q_max) break;
+        if (p.has_win && (wg_q_min - jmax) > p.win) continue;
         const bool full_tile =
             (jmax - j0 == DQ_KVBLK - 1) &&
-            (!p.causal || jmax <= wg_i_min + p.diag) &&
-            (!p.has_win || (wg_i_max - j0) <= p.win) &&
+            (!p.causal || jmax <= wg_q_min) &&
+            (!p.has_win || (wg_q_max - j0) <= p.win) &&
             !p.kmask;
 
         const __bf16* kbase = (const __bf16*)p.k + ((long)b * p.nk) * p.hk * D + (long)hk * D;
@@ -211,8 +214,8 @@ __global__ __launch_bounds__(512, 1) void attn_bwd_dq_kernel(BwdParams p) {
                 }
                 bool ok = row_valid && j <= jmax;
                 if (!full_tile) {
-                    if (p.causal) ok = ok && (j <= i + p.diag);
-                    if (p.has_win) ok = ok && (i - j <= p.win);
+                    if (p.causal) ok = ok && (j <= qpos_i);
+                    if (p.has_win) ok = ok && (qpos_i - j <= p.win);
                     if (p.kmask) ok = ok && lds.kmask[j - j0];
                 }
                 float pv = ok ? __expf(x - lse_i) : 0.f;
@@ -336,13 +339,17 @@ __global__ __launch_bounds__(512, 1) void attn_bwd_dkv_kernel(BwdParams p) {
 
         int t0 = 0, t1 = num_q_tiles;
         if (p.causal) {
-            long i_min_needed = j0_wg - p.diag;
+            // need qpos(i) >= j0_wg  =>  i >= (j0_wg - diag) / q_stride
+            long i_min_needed = (j0_wg - p.diag + p.q_stride - 1) / p.q_stride;
             if (i_min_needed > 0) t0 = (int)(i_min_needed / QT);
         }
         if (p.has_win) {
-            long i_max_needed = jmax + p.win;
+            // need qpos(i) <= jmax + win  =>  i <= (jmax + win - diag) / q_stride
+            long num = jmax + p.win - p.diag;
+            long i_max_needed = num < 0 ? -1 : num / p.q_stride;
             if (i_max_needed < (long)num_q_tiles * QT)
-                t1 = (int)min((long)num_q_tiles, i_max_needed / QT + 1);
+                t1 = (int)min((long)num_q_tiles,
+                              i_max_needed < 0 ? 0 : i_max_needed / QT + 1);
         }
         if (p.split > 1) {                 // grid.z splits the q walk
             int per = (num_q_tiles + p.split - 1) / p.split;
@@ -353,10 +360,12 @@ __global__ __launch_bounds__(512, 1) void attn_bwd_dkv_kernel(BwdParams p) {
         for (int t = t0; t < t1; ++t) {
             const long i0 = (long)t * QT;
             const long imax = min(i0 + QT, p.nq) - 1;
+            const long q_lo = i0 * p.q_stride + p.diag;
+            const long q_hi = imax * p.q_stride + p.diag;
             const bool full_tile =
                 (imax - i0 == QT - 1) &&
-                (!p.causal || (i0 - (jmax - p.diag)) >= 0) &&
-                (!p.has_win || ((imax - j0_wg) <= p.win)) &&
+                (!p.causal || q_lo >= jmax) &&
+                (!p.has_win || ((q_hi - j0_wg) <= p.win)) &&
                 !p.kmask;
 
             __syncthreads();
@@ -404,8 +413,9 @@ __global__ __launch_bounds__(512, 1) void attn_bwd_dkv_kernel(BwdParams p) {
                         }
                         bool ok = col_valid && i <= imax;
                         if (!full_tile) {
-                            if (p.causal) ok = ok && (j <= i + p.diag);
-                            if (p.has_win) ok = ok && (i - j <= p.win);
+                            long qpos = i * p.q_stride + p.diag;
+                            if (p.causal) ok = ok && (j <= qpos);
+                            if (p.has_win) ok = ok && (qpos - j <= p.win);
                             if (p.kmask) ok = ok && kmask_own;
                         }
                         float pv = ok ? __expf(x - lds.lse[qloc]) : 0.f;

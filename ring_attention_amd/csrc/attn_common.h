@@ -37,8 +37,9 @@ struct FwdParams {
     float softclamp_value;
     int softclamp;          // bool
     int causal;             // bool
-    long diag;              // attend iff j <= i + diag (when causal)
-    long win;               // attend iff i - j <= win (only when has_win)
+    long diag;              // q position base: qpos(i) = i*q_stride + diag
+    long q_stride;          // attend iff j <= qpos(i) (causal)
+    long win;               // attend iff qpos(i) - j <= win (when has_win)
     int has_win;            // lookback window enabled
     int is_first;           // initialize m/l/o instead of loading
     int is_last;            // normalize + write out/lse instead of o_acc/m/l
@@ -85,6 +86,7 @@ struct BwdParams {
     int softclamp;
     int causal;
     long diag;
+    long q_stride;
     long win;
     int has_win;
     int accumulate;

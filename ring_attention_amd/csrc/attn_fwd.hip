@@ -131,18 +131,21 @@ __global__ __launch_bounds__(NTHREADS, 1) void attn_fwd_kernel(FwdParams p) {
     // ---- causal / lookback tile bounds for this workgroup: the masked
     // region is contiguous, so the tile range is computed once (keeps the
     // staging pipeline branch-free)
+    // q positions in kv-local coordinates: qpos(i) = i * q_stride + diag
     const long wg_i_min = (long)qtile * QROWS_WG;
     const long wg_i_max = min((long)(qtile + 1) * QROWS_WG, p.nq) - 1;
+    const long wg_q_min = wg_i_min * p.q_stride + p.diag;
+    const long wg_q_max = wg_i_max * p.q_stride + p.diag;
+    const long qpos_i = i * p.q_stride + p.diag;      // this lane's q position
     const int num_kv_tiles = (int)((p.nk + KVBLK - 1) / KVBLK);
 
     int t_lo = 0, t_hi = num_kv_tiles;
     if (p.causal) {
-        long last = wg_i_max + p.diag;               // largest attendable j
-        t_hi = last < 0 ? 0 : min((long)num_kv_tiles, last / KVBLK + 1);
+        t_hi = wg_q_max < 0 ? 0 : min((long)num_kv_tiles, wg_q_max / KVBLK + 1);
     }
     if (p.has_win) {
-        // tile t attends iff j0 + KVBLK - 1 >= wg_i_min - win
-        long x = wg_i_min - p.win - KVBLK + 1;
+        // tile t attends iff j0 + KVBLK - 1 >= wg_q_min - win
+        long x = wg_q_min - p.win - KVBLK + 1;
         t_lo = x <= 0 ? 0 : (int)((x + KVBLK - 1) / KVBLK);
         if (t_lo > t_hi) t_lo = t_hi;
     }
@@ -255,8 +258,8 @@ __global__ __launch_bounds__(NTHREADS, 1) void attn_fwd_kernel(FwdParams p) {
         const long jmax = min(j0 + KVBLK, p.nk) - 1;
         const bool full_tile =
             (jmax - j0 == KVBLK - 1) &&
-            (!p.causal || jmax <= wg_i_min + p.diag) &&
-            (!p.has_win || (wg_i_max - j0) <= p.win) &&
+            (!p.causal || jmax <= wg_q_min) &&
+            (!p.has_win || (wg_q_max - j0) <= p.win) &&
             !p.kmask;
 
         __syncthreads();
@@ -301,8 +304,8 @@ __global__ __launch_bounds__(NTHREADS, 1) void attn_fwd_kernel(FwdParams p) {
                 if (!full_tile) {
                     long j = j0 + kb * 32 + (r & 3) + 8 * (r >> 2) + 4 * lhi;
                     bool ok = j <= jmax;
-                    if (p.causal) ok = ok && (j <= i + p.diag);
-                    if (p.has_win) ok = ok && (i - j <= p.win);
+                    if (p.causal) ok = ok && (j <= qpos_i);
+                    if (p.has_win) ok = ok && (qpos_i - j <= p.win);
                     if (p.kmask) ok = ok && lds.kmask[par][j - j0];
                     if (!ok) x = MASK_VALUE_F;
                 }

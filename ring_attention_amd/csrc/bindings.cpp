@@ -25,7 +25,8 @@ void attn_fwd(
     std::optional<at::Tensor> l,
     std::optional<at::Tensor> out,
     std::optional<at::Tensor> lse,
-    double scale, bool causal, int64_t diag, int64_t win, bool has_win,
+    double scale, bool causal, int64_t diag, int64_t q_stride,
+    int64_t win, bool has_win,
     bool softclamp, double softclamp_value,
     bool is_first, bool is_last, int64_t kv_split, int64_t ablate) {
     CHECK_BF16_CONTIG(q); CHECK_BF16_CONTIG(k); CHECK_BF16_CONTIG(v);
@@ -72,7 +73,8 @@ void attn_fwd(
     p.nq = Nq; p.nk = Nk;
     p.scale = (float)scale;
     p.softclamp = softclamp; p.softclamp_value = (float)softclamp_value;
-    p.causal = causal; p.diag = diag; p.win = win; p.has_win = has_win;
+    p.causal = causal; p.diag = diag; p.q_stride = q_stride;
+    p.win = win; p.has_win = has_win;
     p.is_first = is_first; p.is_last = is_last;
     p.kv_split = (int)kv_split;
     p.ablate = (int)ablate;
@@ -118,7 +120,8 @@ void attn_bwd(
     std::optional<at::Tensor> kmask,
     at::Tensor lse, at::Tensor delta,
     at::Tensor dq, at::Tensor dk, at::Tensor dv,
-    double scale, bool causal, int64_t diag, int64_t win, bool has_win,
+    double scale, bool causal, int64_t diag, int64_t q_stride,
+    int64_t win, bool has_win,
     bool softclamp, double softclamp_value, bool accumulate, int64_t split,
     int64_t which) {   // 0 = both, 1 = dq only, 2 = dk/dv only
     CHECK_BF16_CONTIG(q); CHECK_BF16_CONTIG(k); CHECK_BF16_CONTIG(v); CHECK_BF16_CONTIG(dout);
@@ -143,7 +146,8 @@ void attn_bwd(
     p.nq = Nq; p.nk = Nk;
     p.scale = (float)scale;
     p.softclamp = softclamp; p.softclamp_value = (float)softclamp_value;
-    p.causal = causal; p.diag = diag; p.win = win; p.has_win = has_win;
+    p.causal = causal; p.diag = diag; p.q_stride = q_stride;
+    p.win = win; p.has_win = has_win;
     p.accumulate = accumulate;
     p.split = (int)split;
 

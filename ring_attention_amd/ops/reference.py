@@ -33,10 +33,12 @@ def default_attention(
     softclamp_value: float = 50.0,
     q_positions: Tensor | None = None,  # (n,) global positions of q rows (for permuted layouts)
     k_positions: Tensor | None = None,  # (n_kv,) global positions of k rows
+    max_lookback_seq_len: int | None = None,  # sliding window (token-exact)
 ) -> Tensor:
     """Exact attention in fp32.  ``*_positions`` generalize the causal mask to
     permuted (striped / zig-zag) sequence layouts: row i may attend col j iff
-    k_positions[j] <= q_positions[i] (default: identity positions)."""
+    k_positions[j] <= q_positions[i] (default: identity positions); the
+    optional window masks q_pos - k_pos > max_lookback_seq_len."""
     b, n, h, d = q.shape
     _, nk, hk, _ = k.shape
     assert h % hk == 0
@@ -57,11 +59,15 @@ def default_attention(
     if mask is not None:
         sim = sim.masked_fill(~mask[:, None, None, :], MASK_VALUE)
 
-    if causal:
+    if causal or max_lookback_seq_len is not None:
         qp = q_positions if q_positions is not None else torch.arange(n, device=q.device)
         kp = k_positions if k_positions is not None else torch.arange(nk, device=q.device)
-        causal_mask = kp[None, :] > qp[:, None]      # (n, nk): True = masked
-        sim = sim.masked_fill(causal_mask[None, None, :, :], MASK_VALUE)
+        if causal:
+            causal_mask = kp[None, :] > qp[:, None]      # (n, nk): True = masked
+            sim = sim.masked_fill(causal_mask[None, None, :, :], MASK_VALUE)
+        if max_lookback_seq_len is not None:
+            window_mask = (qp[:, None] - kp[None, :]) > max_lookback_seq_len
+            sim = sim.masked_fill(window_mask[None, None, :, :], MASK_VALUE)
 
     attn = sim.softmax(dim=-1)
     out = torch.einsum("bhij,bjhd->bihd", attn, v)

@@ -328,3 +328,80 @@ def ring_flash_attn(
         max_lookback_seq_len, ring_size, softclamp_qk_sim, softclamp_value,
     )
     return out
+
+
+# ---------------------------------------------------------------------------
+# all-gather-KV execution strategy (oracle form)
+# ---------------------------------------------------------------------------
+# On an MI355X node the 8 GPUs are FULLY connected by xGMI: a ring pass is
+# bound by one link (~153 GB/s) while an RCCL all-gather stripes across all
+# seven.  With 288 GB of HBM the gathered K/V fits for any practical config,
+# so gathering once and attending locally beats circulating shards.  This is
+# the same insight as the reference's zig-zag scheme (zig_zag_attention.py:
+# 123-127) promoted to a first-class strategy for every layout.  The ring
+# strategy remains for windows (lookback) and beyond-memory sequences.
+
+def layout_positions(n_shard: int, ring_size: int, ring_rank: int, striped: bool,
+                     device) -> tuple[Tensor, Tensor]:
+    """(q_positions of this rank's shard, k_positions of the rank-major gather)."""
+    local = torch.arange(n_shard, device=device)
+    if striped:
+        qp = local * ring_size + ring_rank
+        kp = torch.cat([local * ring_size + s for s in range(ring_size)])
+    else:
+        qp = ring_rank * n_shard + local
+        kp = torch.arange(n_shard * ring_size, device=device)
+    return qp, kp
+
+
+def _eager_with_lse(q, k, v, mask, causal, qp, kp, lookback, sc, scv):
+    b, n, h, d = q.shape
+    hk = k.shape[2]
+    groups = h // hk
+    kf, vf = k.float(), v.float()
+    if groups > 1:
+        kf = kf.repeat_interleave(groups, dim=2)
+        vf = vf.repeat_interleave(groups, dim=2)
+    sim = torch.einsum("bihd,bjhd->bhij", q.float(), kf) * d ** -0.5
+    if sc:
+        sim = softclamp(sim, scv)
+    if mask is not None:
+        sim = sim.masked_fill(~mask[:, None, None, :], MASK_VALUE)
+    if causal:
+        sim = sim.masked_fill((kp[None, :] > qp[:, None])[None, None], MASK_VALUE)
+    if lookback is not None:
+        sim = sim.masked_fill(((qp[:, None] - kp[None, :]) > lookback)[None, None],
+                              MASK_VALUE)
+    lse = sim.logsumexp(dim=-1)
+    out = torch.einsum("bhij,bjhd->bihd", sim.softmax(-1), vf)
+    return out.to(q.dtype), lse
+
+
+def ring_flash_attn_allgather_(
+    q: Tensor, k: Tensor, v: Tensor,
+    mask: Tensor | None = None,
+    causal: bool = False,
+    striped_ring_attn: bool = False,
+    max_lookback_seq_len: int | None = None,
+    ring_size: int | None = None,
+    softclamp_qk_sim: bool = False,
+    softclamp_value: float = 50.0,
+) -> tuple[Tensor, Tensor]:
+    """Oracle all-gather strategy: gather K/V once (autograd all-gather whose
+    backward is the reduce-scatter adjoint), attend locally with global
+    positions.  Semantically identical to the ring strategy."""
+    from ..parallel import RingTopology, all_gather
+    topo = RingTopology(ring_size)
+    # NOTE: gather runs over the WORLD; with sub-rings each ring only needs
+    # its own shards — the oracle keeps it simple and correct for full rings.
+    assert topo.ring_size == topo.world_size or topo.ring_size == 1, \
+        "oracle allgather strategy supports full-world rings"
+    kg, _ = all_gather(k, dim=1)
+    vg, _ = all_gather(v, dim=1)
+    mg = None
+    if mask is not None:
+        mg, _ = all_gather(mask, dim=1)
+    qp, kp = layout_positions(q.shape[1], topo.ring_size, topo.ring_rank,
+                              striped_ring_attn, q.device)
+    return _eager_with_lse(q, kg, vg, mg, causal, qp, kp,
+                           max_lookback_seq_len, softclamp_qk_sim, softclamp_value)

@@ -24,9 +24,61 @@ import torch
 from torch import Tensor
 from torch.autograd import Function
 
+import os
+
 from ..parallel import RingAccumulator, RingTopology, all_ring_pass, is_distributed
+from ..parallel.collectives import gather_cat, reduce_scatter_chunks
 from . import hip_ext
 from .ring_flash import max_hops_for_lookback
+
+# gathered-KV budget for the all-gather strategy (bytes of K+V per rank)
+_AG_BUDGET = int(os.environ.get("RING_ATTN_AG_BUDGET", 16 << 30))
+
+
+def _choose_strategy(use_ring: bool, topo, n, hk, d, lookback, dtype_bytes=2) -> str:
+    """'allgather' gathers K/V once over RCCL (stripes across all 7 xGMI
+    links; one local kernel; backward ends in ONE reduce-scatter) — the
+    MI355X-optimal plan whenever the gathered K/V fits.  'ring' circulates
+    shards (used for sliding windows, sub-rings, and beyond-memory seqs)."""
+    if not use_ring or topo.ring_size == 1:
+        return "local"
+    if topo.ring_size != topo.world_size:
+        return "ring"        # sub-rings: per-ring communicators (future work)
+    if lookback is not None:
+        return "ring"        # window truncates the ring walk — cheaper there
+    kv_bytes = 2 * topo.ring_size * n * hk * d * dtype_bytes
+    return "allgather" if kv_bytes <= _AG_BUDGET else "ring"
+
+
+def _gather_global_order(t, ring_size: int, striped: bool, dim: int = 1):
+    """All-gather seq shards and reorder so index == GLOBAL position
+    (striped shards interleave: global g = local * R + rank)."""
+    full = gather_cat(t, dim=dim)            # rank-major along dim
+    if not striped or ring_size == 1:
+        return full
+    # rank-major (s, local) -> position-major (local, s)
+    shape = t.shape
+    n = shape[dim]
+    view = full.reshape(*shape[:dim], ring_size, n, *shape[dim + 1:])
+    order = list(range(view.ndim))
+    order[dim], order[dim + 1] = order[dim + 1], order[dim]
+    return view.permute(order).reshape(*shape[:dim], ring_size * n, *shape[dim + 1:]).contiguous()
+
+
+def _scatter_chunks_of_global(t_full, ring_size: int, striped: bool, dim: int):
+    """Inverse mapping for gradients: reshape a global-position-ordered
+    tensor into (R, ...) chunks where chunk s is rank s's shard."""
+    shape = t_full.shape
+    N = shape[dim]
+    n = N // ring_size
+    if striped:
+        view = t_full.reshape(*shape[:dim], n, ring_size, *shape[dim + 1:])
+        sdim = dim + 1
+    else:
+        view = t_full.reshape(*shape[:dim], ring_size, n, *shape[dim + 1:])
+        sdim = dim
+    order = [sdim] + [i for i in range(view.ndim) if i != sdim]
+    return view.permute(order).contiguous()
 
 _side_stream = None
 
@@ -40,28 +92,30 @@ def _get_side_stream():
 
 def _hop_geometry(rq: int, rk: int, n: int, ring_size: int, striped: bool,
                   causal: bool, lookback: int | None) -> tuple[bool, int, int]:
-    """Returns (skip_hop, diag, win) for a (q-rank, kv-source-rank) pair.
-
-    attend(i, j) <=> (not causal or j <= i + diag) and (win < 0 or i - j <= win)
+    """Returns (skip_hop, diag, win) for a (q-rank, kv-source-rank) pair,
+    in the kernel's qpos semantics: qpos(i) = i * q_stride + diag (q_stride=1
+    for ring hops — both sides share the layout stride, which cancels);
+    attend(i, j) <=> (not causal or j <= qpos(i)) and (not win or qpos(i) - j <= win).
     """
     if striped:
         diag = 0 if rk <= rq else -1
         if lookback is not None:
-            # (i - j) * R + (rq - rk) <= L  <=>  i - j <= floor((L - rq + rk) / R)
-            win = (lookback - rq + rk) // ring_size
+            # (i-j)*R + rq-rk <= L  <=>  i-j <= floor((L-rq+rk)/R) = W
+            # kernel form: qpos - j = i - j + diag <= W + diag
+            win = (lookback - rq + rk) // ring_size + diag
         else:
-            win = -1
+            win = 0
     else:
         diag = (rq - rk) * n
-        # (rq - rk) * n + i - j <= L  <=>  i - j <= L - diag
-        win = (lookback - diag) if lookback is not None else -1
+        win = lookback if lookback is not None else 0   # qpos - j == global distance
     skip = False
-    if causal and diag < 1 - n:
+    qpos_min, qpos_max = diag, (n - 1) + diag
+    if causal and qpos_max < 0:
         skip = True        # whole shard is in the future
     if lookback is not None:
-        if win < 1 - n:
+        if qpos_min - (n - 1) > win:
             skip = True    # whole shard beyond the window
-        if causal and win < -diag:
+        if causal and win < 0:
             skip = True    # window and causal triangle do not intersect
     return skip, diag, win
 
@@ -93,6 +147,38 @@ class RingFlashAttentionHIPFunction(Function):
         ext = hip_ext.require()
         out = torch.empty_like(qb)
         lse = torch.empty(b, h, n, device=q.device, dtype=torch.float32)
+
+        strategy = _choose_strategy(use_ring, topo, n, hk, d, lookback)
+        if strategy == "allgather":
+            R, rq = topo.ring_size, topo.ring_rank
+            k_full = _gather_global_order(kb, R, striped)
+            v_full = _gather_global_order(vb, R, striped)
+            m_full = _gather_global_order(mask_u8, R, striped) if mask_u8 is not None else None
+            q_stride = R if striped else 1
+            diag = rq if striped else rq * n
+            n_total = n * R
+            qtiles = (n + 255) // 256
+            kv_split = min(16, (n_total + 127) // 128,
+                           max(1, 192 // max(1, qtiles * b * h)))
+            if kv_split > 1:
+                o_part = torch.empty(kv_split, b, h, d, n, device=q.device, dtype=torch.float32)
+                m_part = torch.empty(kv_split, b, h, n, device=q.device, dtype=torch.float32)
+                l_part = torch.empty(kv_split, b, h, n, device=q.device, dtype=torch.float32)
+                ext.attn_fwd(qb, k_full, v_full, m_full, o_part, m_part, l_part,
+                             None, None, scale, causal, diag, q_stride, 0, False,
+                             softclamp_qk_sim, softclamp_value, True, True, kv_split, 0)
+                ext.attn_fwd_merge(o_part, m_part, l_part, None, None, None,
+                                   out, lse, kv_split, b, h, d, n, True, True)
+            else:
+                ext.attn_fwd(qb, k_full, v_full, m_full, None, None, None, out, lse,
+                             scale, causal, diag, q_stride, 0, False,
+                             softclamp_qk_sim, softclamp_value, True, True, 1, 0)
+            ctx.save_for_backward(qb, kb, vb, out, lse,
+                                  mask_u8 if mask_u8 is not None else torch.empty(0))
+            ctx.params = (causal, striped, lookback, hops, softclamp_qk_sim,
+                          softclamp_value, use_ring, topo.ring_size, in_dtype,
+                          "allgather")
+            return out.to(in_dtype), lse
 
         # kv-split: fill the 256 CUs when the natural grid (q-tiles x b*h) is
         # small (flash-decoding-style partials + a softmax-correct merge)
@@ -138,7 +224,7 @@ class RingFlashAttentionHIPFunction(Function):
             if kv_split > 1:
                 ext.attn_fwd(qb, kv_t[0], kv_t[1], mk,
                              o_part, m_part, l_part, None, None,
-                             scale, causal, diag, win, lookback is not None,
+                             scale, causal, diag, 1, win, lookback is not None,
                              softclamp_qk_sim, softclamp_value,
                              is_f, is_l, kv_split, 0)
                 ext.attn_fwd_merge(o_part, m_part, l_part, o_acc, m, l,
@@ -147,14 +233,14 @@ class RingFlashAttentionHIPFunction(Function):
             else:
                 ext.attn_fwd(qb, kv_t[0], kv_t[1], mk,
                              o_acc, m, l, out, lse,
-                             scale, causal, diag, win, lookback is not None,
+                             scale, causal, diag, 1, win, lookback is not None,
                              softclamp_qk_sim, softclamp_value,
                              is_f, is_l, 1, 0)
 
         ctx.save_for_backward(qb, kb, vb, out, lse,
                               mask_u8 if mask_u8 is not None else torch.empty(0))
         ctx.params = (causal, striped, lookback, hops, softclamp_qk_sim,
-                      softclamp_value, use_ring, topo.ring_size, in_dtype)
+                      softclamp_value, use_ring, topo.ring_size, in_dtype, "ring")
         return out.to(in_dtype), lse
 
     @staticmethod
@@ -162,7 +248,7 @@ class RingFlashAttentionHIPFunction(Function):
         qb, kb, vb, out, lse, mask_u8 = ctx.saved_tensors
         mask_u8 = mask_u8 if mask_u8.numel() else None
         (causal, striped, lookback, hops, softclamp_qk_sim, softclamp_value,
-         use_ring, ring_size, in_dtype) = ctx.params
+         use_ring, ring_size, in_dtype, strategy) = ctx.params
         b, n, h, d = qb.shape
         hk = kb.shape[2]
         scale = d ** -0.5
@@ -178,6 +264,39 @@ class RingFlashAttentionHIPFunction(Function):
 
         dq = torch.zeros(b, n, h, d, device=qb.device, dtype=torch.float32)
         rq = topo.ring_rank
+
+        if strategy == "allgather":
+            R = topo.ring_size
+            k_full = _gather_global_order(kb, R, striped)
+            v_full = _gather_global_order(vb, R, striped)
+            m_full = _gather_global_order(mask_u8, R, striped) if mask_u8 is not None else None
+            n_total = n * R
+            q_stride = R if striped else 1
+            diag = rq if striped else rq * n
+            dk_full = torch.zeros(b, hk, n_total, d, device=qb.device, dtype=torch.float32)
+            dv_full = torch.zeros(b, hk, d, n_total, device=qb.device, dtype=torch.float32)
+            s2 = _get_side_stream()
+            ev = torch.cuda.Event(); ev.record()
+            with torch.cuda.stream(s2):
+                s2.wait_event(ev)
+                ext.attn_bwd(qb, k_full, v_full, dob, m_full, lse, delta,
+                             dq, dk_full, dv_full, scale, causal, diag, q_stride,
+                             0, False, softclamp_qk_sim, softclamp_value, False, 1, 1)
+            ext.attn_bwd(qb, k_full, v_full, dob, m_full, lse, delta,
+                         dq, dk_full, dv_full, scale, causal, diag, q_stride,
+                         0, False, softclamp_qk_sim, softclamp_value, False, 1, 2)
+            ev2 = torch.cuda.Event(); ev2.record(s2)
+            torch.cuda.current_stream().wait_event(ev2)
+            # ONE reduce-scatter returns each rank's dk/dv shard (summed)
+            dk_chunks = _scatter_chunks_of_global(dk_full, R, striped, dim=2)
+            dv_chunks = _scatter_chunks_of_global(dv_full, R, striped, dim=3)
+            packed = torch.cat((dk_chunks.reshape(R, -1), dv_chunks.reshape(R, -1)), dim=1)
+            own = reduce_scatter_chunks(packed)
+            half = own.numel() // 2
+            dk_home = own[:half].view(b, hk, n, d).permute(0, 2, 1, 3).contiguous()
+            dv_home = own[half:].view(b, hk, d, n).permute(0, 3, 1, 2).contiguous()
+            return (dq.to(in_dtype), dk_home.to(in_dtype), dv_home.to(in_dtype),
+                    None, None, None, None, None, None, None, None, None)
         qtiles = (n + 255) // 256
         bwd_split = min(8, max(1, 192 // max(1, qtiles * b * h)))
 
@@ -208,12 +327,12 @@ class RingFlashAttentionHIPFunction(Function):
                     s2.wait_event(ev)
                     ext.attn_bwd(qb, kv_t[0], kv_t[1], dob, mk, lse, delta,
                                  dq, dk_n, dv_n,
-                                 scale, causal, diag, win, lookback is not None,
+                                 scale, causal, diag, 1, win, lookback is not None,
                                  softclamp_qk_sim, softclamp_value, False,
                                  bwd_split, 1)   # dq only
                 ext.attn_bwd(qb, kv_t[0], kv_t[1], dob, mk, lse, delta,
                              dq, dk_n, dv_n,
-                             scale, causal, diag, win, lookback is not None,
+                             scale, causal, diag, 1, win, lookback is not None,
                              softclamp_qk_sim, softclamp_value, False,
                              bwd_split, 2)       # dk/dv only
                 ev2 = torch.cuda.Event()
@@ -273,7 +392,7 @@ class FlashAttnOffsetFunction(Function):
         out = torch.empty_like(qb)
         lse = torch.empty(b, h, n, device=q.device, dtype=torch.float32)
         ext.attn_fwd(qb, kb, vb, None, None, None, None, out, lse,
-                     scale, causal, q_offset, 0, False, False, 50.0,
+                     scale, causal, q_offset, 1, 0, False, False, 50.0,
                      True, True, 1, 0)
         ctx.save_for_backward(qb, kb, vb, out, lse)
         ctx.meta = (q_offset, causal, q.dtype)
@@ -294,7 +413,7 @@ class FlashAttnOffsetFunction(Function):
         dk_n = torch.zeros(b, hk, nk, d, device=qb.device, dtype=torch.float32)
         dv_n = torch.zeros(b, hk, d, nk, device=qb.device, dtype=torch.float32)
         ext.attn_bwd(qb, kb, vb, dob, None, lse, delta, dq, dk_n, dv_n,
-                     scale, causal, q_offset, 0, False, False, 50.0, False, 1, 0)
+                     scale, causal, q_offset, 1, 0, False, False, 50.0, False, 1, 0)
         dk = dk_n.permute(0, 2, 1, 3).contiguous()
         dv = dv_n.permute(0, 3, 1, 2).contiguous()
         return (dq.to(in_dtype), dk.to(in_dtype), dv.to(in_dtype), None, None)

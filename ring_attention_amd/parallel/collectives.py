@@ -121,3 +121,39 @@ def split_by_rank(xs: list[Tensor]) -> Tensor:
         assert len(xs) == 1
         return xs[0]
     return xs[get_rank()]
+
+
+def gather_cat(t: Tensor, dim: int) -> Tensor:
+    """Plain (non-autograd) all-gather concatenated along ``dim`` in rank order."""
+    if not is_distributed():
+        return t
+    world = get_world_size()
+    t = t.contiguous()
+    if dist.get_backend() == "nccl":
+        out = torch.empty((world,) + tuple(t.shape), device=t.device, dtype=t.dtype)
+        dist.all_gather_into_tensor(out.view(world, -1), t.view(-1))
+        chunks = list(out.unbind(0))
+    else:
+        chunks = [torch.empty_like(t) for _ in range(world)]
+        dist.all_gather(chunks, t)
+    return torch.cat(chunks, dim=dim)
+
+
+def reduce_scatter_chunks(chunks: Tensor) -> Tensor:
+    """``chunks`` (W, ...) — sum chunk r across ranks, return this rank's chunk.
+
+    RCCL reduce-scatter stripes across every xGMI link; the gloo fallback is
+    all-reduce + slice (same result).
+    """
+    if not is_distributed():
+        assert chunks.shape[0] == 1
+        return chunks[0]
+    world = get_world_size()
+    assert chunks.shape[0] == world
+    chunks = chunks.contiguous()
+    if dist.get_backend() == "nccl":
+        out = torch.empty_like(chunks[0])
+        dist.reduce_scatter_tensor(out.view(-1), chunks.view(world, -1).reshape(-1))
+        return out
+    dist.all_reduce(chunks)
+    return chunks[get_rank()].clone()

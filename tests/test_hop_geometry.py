@@ -34,11 +34,12 @@ def hip_mask(rq, rk, n, R, striped, causal, lookback):
         return torch.zeros(n, n, dtype=torch.bool)
     i = torch.arange(n)[:, None]
     j = torch.arange(n)[None, :]
+    qpos = i + diag                      # kernel semantics: qpos(i) = i*qs + diag
     allowed = torch.ones(n, n, dtype=torch.bool)
     if causal:
-        allowed &= j <= i + diag
+        allowed &= j <= qpos
     if lookback is not None:
-        allowed &= (i - j) <= win
+        allowed &= (qpos - j) <= win
     return allowed
 
 
