@@ -90,24 +90,28 @@ __device__ __forceinline__ void stage_transposed(
 }
 
 // ---------------------------------------------------------------------------
-// dq kernel: row-parallel, forward-like
+// dq kernel: row-parallel, forward-like (same pipeline as attn_fwd_kernel:
+// double-buffered LDS, one barrier per kv tile, running-pointer staging)
 // ---------------------------------------------------------------------------
 static constexpr int DQ_WAVES = 8;
 static constexpr int DQ_QROWS_WG = DQ_WAVES * 32;     // 256
-static constexpr int DQ_KVBLK = 64;
+template <int D> constexpr int dq_kvblk() { return D == 64 ? 128 : 64; }  // LDS budget
 
 template <int D>
 struct DqLds {
-    __align__(16) __bf16 k[DQ_KVBLK * D];    // [kv][d] swizzled
-    __align__(16) __bf16 v[DQ_KVBLK * D];    // [kv][d] swizzled
-    __align__(16) __bf16 kt[D * DQ_KVBLK];   // [d][kv] swizzled (pair-staged)
-    unsigned char kmask[DQ_KVBLK];
+    static constexpr int KVB = dq_kvblk<D>();
+    __align__(16) __bf16 k[2][KVB * D];    // [kv][d] swizzled
+    __align__(16) __bf16 v[2][KVB * D];    // [kv][d] swizzled
+    __align__(16) __bf16 kt[2][D * KVB];   // [d][kv] swizzled (pair-staged)
+    unsigned char kmask[2][KVB];
 };
 
 template <int D, bool SOFTCLAMP>
 __global__ __launch_bounds__(512, 1) void attn_bwd_dq_kernel(BwdParams p) {
     constexpr int DBLK = D / 32;
     constexpr int KSTEPS = D / 16;
+    constexpr int DQ_KVBLK = dq_kvblk<D>();
+    constexpr int DQ_NBLK = DQ_KVBLK / 32;
 
     __shared__ DqLds<D> lds;
 
@@ -136,7 +140,7 @@ __global__ __launch_bounds__(512, 1) void attn_bwd_dq_kernel(BwdParams p) {
         qf[ks] = *(const bf16x8*)(qbase + ks * 16 + lhi * 8);
         dof[ks] = *(const bf16x8*)(dobase + ks * 16 + lhi * 8);
     }
-    const float lse_i = p.lse[((long)b * p.h + h) * p.nq + ic];
+    const float lse_i = p.lse[((long)b * p.h + h) * p.nq + ic] * 1.4426950408889634f;
     const float delta_i = p.delta[((long)b * p.h + h) * p.nq + ic];
 
     f32x16 dq_acc[DBLK];
@@ -149,76 +153,169 @@ __global__ __launch_bounds__(512, 1) void attn_bwd_dq_kernel(BwdParams p) {
     const long wg_q_max = wg_i_max * p.q_stride + p.diag;
     const long qpos_i = i * p.q_stride + p.diag;
     const int num_kv_tiles = (int)((p.nk + DQ_KVBLK - 1) / DQ_KVBLK);
-    int zt_lo = 0, zt_hi = num_kv_tiles;
-    if (p.split > 1) {                     // grid.z splits the kv walk
+
+    int t_lo = 0, t_hi = num_kv_tiles;
+    if (p.causal)
+        t_hi = wg_q_max < 0 ? 0 : min((long)num_kv_tiles, wg_q_max / DQ_KVBLK + 1);
+    if (p.has_win) {
+        long x = wg_q_min - p.win - DQ_KVBLK + 1;
+        t_lo = x <= 0 ? 0 : (int)((x + DQ_KVBLK - 1) / DQ_KVBLK);
+        if (t_lo > t_hi) t_lo = t_hi;
+    }
+    if (p.split > 1) {
         int per = (num_kv_tiles + p.split - 1) / p.split;
-        zt_lo = blockIdx.z * per;
-        zt_hi = min(num_kv_tiles, zt_lo + per);
+        t_lo = max(t_lo, (int)(blockIdx.z * per));
+        t_hi = min(t_hi, (int)((blockIdx.z + 1) * per));
+        if (t_lo > t_hi) t_lo = t_hi;
     }
 
-    for (int t = zt_lo; t < zt_hi; ++t) {
+    // staging: K + V row chunks, K^T pairs, running pointers
+    constexpr int CH = D * 2 / 16;
+    constexpr int KCHUNKS = DQ_KVBLK * CH;
+    constexpr int KREGS = (KCHUNKS + 511) / 512;
+    constexpr int TPAIRS = (DQ_KVBLK / 2) * (D / 8);
+    constexpr int TREGS = (TPAIRS + 511) / 512;
+    const __bf16* kbase = (const __bf16*)p.k + ((long)b * p.nk) * p.hk * D + (long)hk * D;
+    const __bf16* vbase = (const __bf16*)p.v + ((long)b * p.nk) * p.hk * D + (long)hk * D;
+    const unsigned char* mbase = p.kmask ? (const unsigned char*)p.kmask + (long)b * p.nk : nullptr;
+
+    const long kv_row_stride = (long)p.hk * D;
+    const long tile_stride = DQ_KVBLK * kv_row_stride;
+    const __bf16* kptr = kbase + (long)t_lo * tile_stride
+        + (tid / CH) * kv_row_stride + (tid % CH) * 8;
+    const __bf16* vptr = vbase + (long)t_lo * tile_stride
+        + (tid / CH) * kv_row_stride + (tid % CH) * 8;
+    const __bf16* ktpa = kbase + (long)t_lo * tile_stride
+        + ((tid % (DQ_KVBLK / 2)) * 2) * kv_row_stride + (tid / (DQ_KVBLK / 2)) * 8;
+    long j0_next = (long)t_lo * DQ_KVBLK;
+
+    uint4 kst[KREGS], vst[KREGS];
+    bf16x8 kta_st[TREGS], ktb_st[TREGS];
+    unsigned char mst = 1;
+
+    auto load_tile = [&]() {
+        const long j0 = j0_next;
+        const long jmax = min(j0 + DQ_KVBLK, p.nk) - 1;
+        const bool full = jmax - j0 == DQ_KVBLK - 1;
+        #pragma unroll
+        for (int r = 0; r < KREGS; ++r) {
+            int c = tid + r * 512;
+            if (c < KCHUNKS) {
+                long off = (long)(r * (512 / CH)) * kv_row_stride;
+                bool okr = full || (j0 + c / CH) <= jmax;
+                kst[r] = okr ? *(const uint4*)(kptr + off) : uint4{0, 0, 0, 0};
+                vst[r] = okr ? *(const uint4*)(vptr + off) : uint4{0, 0, 0, 0};
+            }
+        }
+        #pragma unroll
+        for (int r = 0; r < TREGS; ++r) {
+            int c = tid + r * 512;
+            if (c < TPAIRS) {
+                const __bf16* sa = ktpa + r * (512 / (DQ_KVBLK / 2)) * 8;
+                long ja = j0 + (c % (DQ_KVBLK / 2)) * 2;
+                kta_st[r] = (full || ja <= jmax) ? *(const bf16x8*)sa : bf16x8{};
+                ktb_st[r] = (full || ja + 1 <= jmax) ? *(const bf16x8*)(sa + kv_row_stride) : bf16x8{};
+            }
+        }
+        if (mbase && tid < DQ_KVBLK)
+            mst = (j0 + tid <= jmax) ? mbase[j0 + tid] : 0;
+        kptr += tile_stride; vptr += tile_stride; ktpa += tile_stride;
+        j0_next += DQ_KVBLK;
+    };
+
+    auto write_tile = [&](int par) {
+        #pragma unroll
+        for (int r = 0; r < KREGS; ++r) {
+            int c = tid + r * 512;
+            if (c < KCHUNKS) {
+                int row = c / CH, ch = c % CH;
+                *(uint4*)(lds.k[par] + row * D + bswz(row, ch) * 8) = kst[r];
+                *(uint4*)(lds.v[par] + row * D + bswz(row, ch) * 8) = vst[r];
+            }
+        }
+        #pragma unroll
+        for (int r = 0; r < TREGS; ++r) {
+            int c = tid + r * 512;
+            if (c < TPAIRS) {
+                int jp = c % (DQ_KVBLK / 2);
+                int d0 = (c / (DQ_KVBLK / 2)) * 8;
+                #pragma unroll
+                for (int e = 0; e < 8; ++e) {
+                    int dd = d0 + e;
+                    int byte_off = dd * DQ_KVBLK * 2 + ((jp * 4) ^ ((dd & 7) << 4));
+                    __bf16 pr[2] = {kta_st[r][e], ktb_st[r][e]};
+                    *(uint32_t*)((char*)lds.kt[par] + byte_off) = *(uint32_t*)pr;
+                }
+            }
+        }
+        if (mbase && tid < DQ_KVBLK) lds.kmask[par][tid] = mst;
+    };
+
+    const float scale2 = p.scale * 1.4426950408889634f;   // exp2 domain
+    if (t_lo < t_hi) {
+        load_tile();
+        write_tile(t_lo & 1);
+        if (t_lo + 1 < t_hi) load_tile();
+    }
+
+    for (int t = t_lo; t < t_hi; ++t) {
+        const int par = t & 1;
         const long j0 = (long)t * DQ_KVBLK;
         const long jmax = min(j0 + DQ_KVBLK, p.nk) - 1;
-        if (p.causal && j0 > wg_q_max) break;
-        if (p.has_win && (wg_q_min - jmax) > p.win) continue;
         const bool full_tile =
             (jmax - j0 == DQ_KVBLK - 1) &&
             (!p.causal || jmax <= wg_q_min) &&
             (!p.has_win || (wg_q_max - j0) <= p.win) &&
             !p.kmask;
 
-        const __bf16* kbase = (const __bf16*)p.k + ((long)b * p.nk) * p.hk * D + (long)hk * D;
-        const __bf16* vbase = (const __bf16*)p.v + ((long)b * p.nk) * p.hk * D + (long)hk * D;
-        stage_rowmajor<D, DQ_KVBLK>(kbase, j0, jmax, (long)p.hk * D, lds.k, tid);
-        stage_rowmajor<D, DQ_KVBLK>(vbase, j0, jmax, (long)p.hk * D, lds.v, tid);
-        stage_transposed<D, DQ_KVBLK>(kbase, j0, jmax, (long)p.hk * D, lds.kt, tid);
-        if (p.kmask) {
-            const unsigned char* mb = (const unsigned char*)p.kmask + (long)b * p.nk;
-            for (int c = tid; c < DQ_KVBLK; c += 512)
-                lds.kmask[c] = (j0 + c <= jmax) ? mb[j0 + c] : 0;
-        }
         __syncthreads();
 
         // s^T and dp^T: lane = q, kv in regs
-        f32x16 s[2], dp[2];
+        f32x16 s[DQ_NBLK], dp[DQ_NBLK];
+        __builtin_amdgcn_s_setprio(1);
         #pragma unroll
-        for (int kb = 0; kb < 2; ++kb) {
+        for (int kb = 0; kb < DQ_NBLK; ++kb) {
             s[kb] = f32x16{}; dp[kb] = f32x16{};
             int krow = kb * 32 + l31;
             #pragma unroll
             for (int ks = 0; ks < KSTEPS; ++ks) {
                 int chunk = ks * 2 + lhi;
-                bf16x8 kfr = *(const bf16x8*)(lds.k + krow * D + bswz(krow, chunk) * 8);
-                bf16x8 vfr = *(const bf16x8*)(lds.v + krow * D + bswz(krow, chunk) * 8);
+                bf16x8 kfr = *(const bf16x8*)(lds.k[par] + krow * D + bswz(krow, chunk) * 8);
+                bf16x8 vfr = *(const bf16x8*)(lds.v[par] + krow * D + bswz(krow, chunk) * 8);
                 s[kb] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kfr, qf[ks], s[kb], 0, 0, 0);
                 dp[kb] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vfr, dof[ks], dp[kb], 0, 0, 0);
             }
         }
+        __builtin_amdgcn_s_setprio(0);
 
-        // ds^T in regs -> packed bf16 pairs
-        uint32_t pk[16];
+        if (t + 1 < t_hi) write_tile(par ^ 1);
+        if (t + 2 < t_hi) load_tile();
+
+        // ds^T in regs -> packed bf16 pairs (exp2 domain)
+        uint32_t pk[DQ_NBLK * 8];
         #pragma unroll
-        for (int x2 = 0; x2 < 16; ++x2) {
+        for (int x2 = 0; x2 < DQ_NBLK * 8; ++x2) {
             float dse[2];
             #pragma unroll
             for (int e = 0; e < 2; ++e) {
                 int kb = x2 >> 3, r = (2 * x2 + e) & 15;
-                long j = j0 + kb * 32 + (r & 3) + 8 * (r >> 2) + 4 * lhi;
-                float x = s[kb][r] * p.scale;
-                float dtanh = 1.f;
+                long jj = j0 + kb * 32 + (r & 3) + 8 * (r >> 2) + 4 * lhi;
+                float x, dtanh = 1.f;
                 if constexpr (SOFTCLAMP) {
                     float inv_v = __builtin_amdgcn_rcpf(p.softclamp_value);
-                    float th = bfast_tanhf(x * inv_v);
-                    x = p.softclamp_value * th;
+                    float th = bfast_tanhf(s[kb][r] * p.scale * inv_v);
+                    x = p.softclamp_value * th * 1.4426950408889634f;
                     dtanh = 1.f - th * th;
+                } else {
+                    x = s[kb][r] * scale2;
                 }
-                bool ok = row_valid && j <= jmax;
+                bool ok = row_valid && jj <= jmax;
                 if (!full_tile) {
-                    if (p.causal) ok = ok && (j <= qpos_i);
-                    if (p.has_win) ok = ok && (qpos_i - j <= p.win);
-                    if (p.kmask) ok = ok && lds.kmask[j - j0];
+                    if (p.causal) ok = ok && (jj <= qpos_i);
+                    if (p.has_win) ok = ok && (qpos_i - jj <= p.win);
+                    if (p.kmask) ok = ok && lds.kmask[par][jj - j0];
                 }
-                float pv = ok ? __expf(x - lse_i) : 0.f;
+                float pv = ok ? __builtin_amdgcn_exp2f(x - lse_i) : 0.f;
                 dse[e] = pv * (dp[kb][r] - delta_i) * dtanh * p.scale;
             }
             union { __hip_bfloat162 h2; uint32_t u; } cvt;
@@ -227,9 +324,9 @@ __global__ __launch_bounds__(512, 1) void attn_bwd_dq_kernel(BwdParams p) {
         }
 
         // build B-operand fragments (lane = q, k = kv contiguous): pairs +2
-        uint32_t frag[4][4];
+        uint32_t frag[DQ_NBLK * 2][4];
         #pragma unroll
-        for (int kb = 0; kb < 2; ++kb)
+        for (int kb = 0; kb < DQ_NBLK; ++kb)
             #pragma unroll
             for (int half = 0; half < 2; ++half)
                 #pragma unroll
@@ -241,18 +338,20 @@ __global__ __launch_bounds__(512, 1) void attn_bwd_dq_kernel(BwdParams p) {
                 }
 
         // dq^T[d][q] += K^T[d][kv] x ds^T[kv][q]
+        __builtin_amdgcn_s_setprio(1);
         #pragma unroll
         for (int db = 0; db < DBLK; ++db) {
             int drow = db * 32 + l31;
             #pragma unroll
-            for (int ks = 0; ks < 4; ++ks) {
+            for (int ks = 0; ks < DQ_NBLK * 2; ++ks) {
                 int chunk = ks * 2 + lhi;
-                bf16x8 ktf = *(const bf16x8*)(lds.kt + drow * DQ_KVBLK + bswz(drow, chunk) * 8);
+                bf16x8 ktf = *(const bf16x8*)(lds.kt[par] + drow * DQ_KVBLK +
+                                              bswz(drow, chunk) * 8);
                 dq_acc[db] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
                     ktf, *(const bf16x8*)frag[ks], dq_acc[db], 0, 0, 0);
             }
         }
-        __syncthreads();
+        __builtin_amdgcn_s_setprio(0);
     }
 
     if (!row_valid) return;
@@ -278,12 +377,14 @@ static constexpr int KVROWS_WG = BWD_WAVES * KVROWS_WAVE;   // 256
 
 template <int D, int QT>
 struct DkvLds {
-    __align__(16) __bf16 q[QT * D];         // [q][d]   swizzled rows
-    __align__(16) __bf16 qt[D * QT];        // [d][q]   pair-staged transpose
-    __align__(16) __bf16 do_[QT * D];       // [q][d]
-    __align__(16) __bf16 dot[D * QT];       // [d][q]
-    float lse[QT];
-    float delta[QT];
+    // double-buffered q-tile images (one barrier per q tile; staging writes
+    // and next-next tile's loads overlap the MFMAs, as in the forward kernel)
+    __align__(16) __bf16 q[2][QT * D];      // [q][d]   swizzled rows
+    __align__(16) __bf16 qt[2][D * QT];     // [d][q]   pair-staged transpose
+    __align__(16) __bf16 do_[2][QT * D];    // [q][d]
+    __align__(16) __bf16 dot[2][D * QT];    // [d][q]
+    __align__(16) float lse[2][QT];
+    __align__(16) float delta[2][QT];
 };
 
 template <int D, int QT, bool SOFTCLAMP>
@@ -357,7 +458,101 @@ __global__ __launch_bounds__(512, 1) void attn_bwd_dkv_kernel(BwdParams p) {
             t1 = min(t1, (int)((blockIdx.z + 1) * per));
         }
 
+        // ---- T14 pipeline: per-thread staging registers
+        constexpr int CH = D * 2 / 16;
+        constexpr int QCHUNKS = QT * CH;
+        constexpr int QREGS = (QCHUNKS + 511) / 512;
+        constexpr int TPAIRS = (QT / 2) * (D / 8);
+        constexpr int TREGS = (TPAIRS + 511) / 512;
+        uint4 qst[QREGS], dost[QREGS];
+        bf16x8 qtv_a[TREGS], qtv_b[TREGS], dov_a[TREGS], dov_b[TREGS];
+        float lse_st = 0.f, delta_st = 0.f;
+        long t_next = t0;
+
+        auto load_qtile = [&]() {
+            const long i0 = t_next * QT;
+            const long imax_ = min(i0 + QT, p.nq) - 1;
+            const bool full = imax_ - i0 == QT - 1;
+            #pragma unroll
+            for (int r = 0; r < QREGS; ++r) {
+                int c = tid + r * 512;
+                if (c < QCHUNKS) {
+                    long gr = i0 + c / CH;
+                    int ch = c % CH;
+                    const __bf16* src = qg + gr * p.h * D + ch * 8;
+                    const __bf16* srd = dog + gr * p.h * D + ch * 8;
+                    bool okr = full || gr <= imax_;
+                    qst[r] = okr ? *(const uint4*)src : uint4{0, 0, 0, 0};
+                    dost[r] = okr ? *(const uint4*)srd : uint4{0, 0, 0, 0};
+                }
+            }
+            #pragma unroll
+            for (int r = 0; r < TREGS; ++r) {
+                int c = tid + r * 512;
+                if (c < TPAIRS) {
+                    int jp = c % (QT / 2);
+                    int d0 = (c / (QT / 2)) * 8;
+                    long ra = i0 + jp * 2;
+                    bool oka = full || ra <= imax_;
+                    bool okb = full || ra + 1 <= imax_;
+                    const __bf16* qa_ = qg + ra * p.h * D + d0;
+                    const __bf16* da_ = dog + ra * p.h * D + d0;
+                    qtv_a[r] = oka ? *(const bf16x8*)qa_ : bf16x8{};
+                    qtv_b[r] = okb ? *(const bf16x8*)(qa_ + p.h * D) : bf16x8{};
+                    dov_a[r] = oka ? *(const bf16x8*)da_ : bf16x8{};
+                    dov_b[r] = okb ? *(const bf16x8*)(da_ + p.h * D) : bf16x8{};
+                }
+            }
+            if (tid < QT) {
+                long gi = i0 + tid;
+                bool okl = gi <= imax_;
+                lse_st = okl ? lse_row[gi] : 0.f;
+                delta_st = okl ? delta_row[gi] : 0.f;
+            }
+            ++t_next;
+        };
+
+        auto write_qtile = [&](int par) {
+            #pragma unroll
+            for (int r = 0; r < QREGS; ++r) {
+                int c = tid + r * 512;
+                if (c < QCHUNKS) {
+                    int row = c / CH, ch = c % CH;
+                    *(uint4*)(lds.q[par] + row * D + bswz(row, ch) * 8) = qst[r];
+                    *(uint4*)(lds.do_[par] + row * D + bswz(row, ch) * 8) = dost[r];
+                }
+            }
+            #pragma unroll
+            for (int r = 0; r < TREGS; ++r) {
+                int c = tid + r * 512;
+                if (c < TPAIRS) {
+                    int jp = c % (QT / 2);
+                    int d0 = (c / (QT / 2)) * 8;
+                    #pragma unroll
+                    for (int e = 0; e < 8; ++e) {
+                        int dd = d0 + e;
+                        int byte_off = dd * QT * 2 + ((jp * 4) ^ ((dd & TM) << 4));
+                        __bf16 pq[2] = {qtv_a[r][e], qtv_b[r][e]};
+                        __bf16 pd[2] = {dov_a[r][e], dov_b[r][e]};
+                        *(uint32_t*)((char*)lds.qt[par] + byte_off) = *(uint32_t*)pq;
+                        *(uint32_t*)((char*)lds.dot[par] + byte_off) = *(uint32_t*)pd;
+                    }
+                }
+            }
+            if (tid < QT) {
+                lds.lse[par][tid] = lse_st;
+                lds.delta[par][tid] = delta_st;
+            }
+        };
+
+        if (t0 < t1) {
+            load_qtile();
+            write_qtile(t0 & 1);
+            if (t0 + 1 < t1) load_qtile();
+        }
+
         for (int t = t0; t < t1; ++t) {
+            const int par = t & 1;
             const long i0 = (long)t * QT;
             const long imax = min(i0 + QT, p.nq) - 1;
             const long q_lo = i0 * p.q_stride + p.diag;
@@ -369,27 +564,35 @@ __global__ __launch_bounds__(512, 1) void attn_bwd_dkv_kernel(BwdParams p) {
                 !p.kmask;
 
             __syncthreads();
-            stage_rowmajor<D, QT>(qg, i0, imax, (long)p.h * D, lds.q, tid);
-            stage_rowmajor<D, QT>(dog, i0, imax, (long)p.h * D, lds.do_, tid);
-            stage_transposed<D, QT>(qg, i0, imax, (long)p.h * D, lds.qt, tid);
-            stage_transposed<D, QT>(dog, i0, imax, (long)p.h * D, lds.dot, tid);
-            for (int c = tid; c < QT; c += 512) {
-                long gi = i0 + c;
-                lds.lse[c] = (gi <= imax) ? lse_row[gi] : 0.f;
-                lds.delta[c] = (gi <= imax) ? delta_row[gi] : 0.f;
-            }
-            __syncthreads();
+
+            if (t + 1 < t1) write_qtile(par ^ 1);
+            if (t + 2 < t1) load_qtile();
 
             #pragma unroll
             for (int qb = 0; qb < QBLKS; ++qb) {
+                // preload this block's lse/delta for the reg-q pattern with
+                // wide reads (qloc = qb*32 + (r&3) + 8*(r>>2) + 4*lhi: regs
+                // 4g..4g+3 are consecutive -> one b128 per group of 4)
+                float lse_r[16], delta_r[16];
+                #pragma unroll
+                for (int g4 = 0; g4 < 4; ++g4) {
+                    int base = qb * 32 + 8 * g4 + 4 * lhi;
+                    f32x4 lv = *(const f32x4*)(lds.lse[par] + base);
+                    f32x4 dv_ = *(const f32x4*)(lds.delta[par] + base);
+                    #pragma unroll
+                    for (int e = 0; e < 4; ++e) {
+                        lse_r[4 * g4 + e] = lv[e] * 1.4426950408889634f;  // exp2 domain
+                        delta_r[4 * g4 + e] = dv_[e];
+                    }
+                }
                 // S2[q][kv], dP[q][kv]: lane = kv, q rows in regs
                 f32x16 s2 = f32x16{}, dp = f32x16{};
                 #pragma unroll
                 for (int ks = 0; ks < KSTEPS; ++ks) {
                     int qrow = qb * 32 + l31;
                     int chunk = ks * 2 + lhi;
-                    bf16x8 qa = *(const bf16x8*)(lds.q + qrow * D + bswz(qrow, chunk) * 8);
-                    bf16x8 da = *(const bf16x8*)(lds.do_ + qrow * D + bswz(qrow, chunk) * 8);
+                    bf16x8 qa = *(const bf16x8*)(lds.q[par] + qrow * D + bswz(qrow, chunk) * 8);
+                    bf16x8 da = *(const bf16x8*)(lds.do_[par] + qrow * D + bswz(qrow, chunk) * 8);
                     s2 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qa, kf[ks], s2, 0, 0, 0);
                     dp = __builtin_amdgcn_mfma_f32_32x32x16_bf16(da, vf[ks], dp, 0, 0, 0);
                 }
@@ -403,13 +606,14 @@ __global__ __launch_bounds__(512, 1) void attn_bwd_dkv_kernel(BwdParams p) {
                         int r = 2 * x2 + e;
                         int qloc = qb * 32 + (r & 3) + 8 * (r >> 2) + 4 * lhi;
                         long i = i0 + qloc;
-                        float x = s2[r] * p.scale;
-                        float dtanh = 1.f;
+                        float x, dtanh = 1.f;
                         if constexpr (SOFTCLAMP) {
                             float inv_v = __builtin_amdgcn_rcpf(p.softclamp_value);
-                            float th = bfast_tanhf(x * inv_v);
-                            x = p.softclamp_value * th;
+                            float th = bfast_tanhf(s2[r] * p.scale * inv_v);
+                            x = p.softclamp_value * th * 1.4426950408889634f;
                             dtanh = 1.f - th * th;
+                        } else {
+                            x = s2[r] * (p.scale * 1.4426950408889634f);
                         }
                         bool ok = col_valid && i <= imax;
                         if (!full_tile) {
@@ -418,9 +622,9 @@ __global__ __launch_bounds__(512, 1) void attn_bwd_dkv_kernel(BwdParams p) {
                             if (p.has_win) ok = ok && (qpos - j <= p.win);
                             if (p.kmask) ok = ok && kmask_own;
                         }
-                        float pv = ok ? __expf(x - lds.lse[qloc]) : 0.f;
+                        float pv = ok ? __builtin_amdgcn_exp2f(x - lse_r[r]) : 0.f;
                         pe[e] = pv;
-                        dse[e] = pv * (dp[r] - lds.delta[qloc]) * dtanh * p.scale;
+                        dse[e] = pv * (dp[r] - delta_r[r]) * dtanh * p.scale;
                     }
                     union { __hip_bfloat162 h2; uint32_t u; } c1, c2;
                     c1.h2 = __float22bfloat162_rn(float2{pe[0], pe[1]});
@@ -453,11 +657,11 @@ __global__ __launch_bounds__(512, 1) void attn_bwd_dkv_kernel(BwdParams p) {
                         int qk = qb * 2 + half;
                         int drow = db * 32 + l31;
                         int ch = qk * 2 + lhi;
-                        bf16x8 doa = *(const bf16x8*)(lds.dot + drow * QT +
+                        bf16x8 doa = *(const bf16x8*)(lds.dot[par] + drow * QT +
                                                       ((ch ^ (drow & TM))) * 8);
                         dv_acc[db] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
                             doa, *(const bf16x8*)p_frag[half], dv_acc[db], 0, 0, 0);
-                        bf16x8 qta = *(const bf16x8*)(lds.qt + drow * QT +
+                        bf16x8 qta = *(const bf16x8*)(lds.qt[par] + drow * QT +
                                                       ((ch ^ (drow & TM))) * 8);
                         dk_acc[db] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
                             *(const bf16x8*)ds_frag[half], qta, dk_acc[db], 0, 0, 0);

@@ -275,18 +275,9 @@ class RingFlashAttentionHIPFunction(Function):
             diag = rq if striped else rq * n
             dk_full = torch.zeros(b, hk, n_total, d, device=qb.device, dtype=torch.float32)
             dv_full = torch.zeros(b, hk, d, n_total, device=qb.device, dtype=torch.float32)
-            s2 = _get_side_stream()
-            ev = torch.cuda.Event(); ev.record()
-            with torch.cuda.stream(s2):
-                s2.wait_event(ev)
-                ext.attn_bwd(qb, k_full, v_full, dob, m_full, lse, delta,
-                             dq, dk_full, dv_full, scale, causal, diag, q_stride,
-                             0, False, softclamp_qk_sim, softclamp_value, False, 1, 1)
             ext.attn_bwd(qb, k_full, v_full, dob, m_full, lse, delta,
                          dq, dk_full, dv_full, scale, causal, diag, q_stride,
-                         0, False, softclamp_qk_sim, softclamp_value, False, 1, 2)
-            ev2 = torch.cuda.Event(); ev2.record(s2)
-            torch.cuda.current_stream().wait_event(ev2)
+                         0, False, softclamp_qk_sim, softclamp_value, False, 1, 0)
             # ONE reduce-scatter returns each rank's dk/dv shard (summed)
             dk_chunks = _scatter_chunks_of_global(dk_full, R, striped, dim=2)
             dv_chunks = _scatter_chunks_of_global(dv_full, R, striped, dim=3)
@@ -318,28 +309,13 @@ class RingFlashAttentionHIPFunction(Function):
             if not skip:
                 dk_n = contrib[0].view(b, hk, n, d)
                 dv_n = contrib[1].view(b, hk, d, n)
-                # dq (row-parallel) and dk/dv (column-parallel) are independent:
-                # run them on separate streams so they co-occupy the CUs
-                s2 = _get_side_stream()
-                ev = torch.cuda.Event()
-                ev.record()                      # delta/do ready on current stream
-                with torch.cuda.stream(s2):
-                    s2.wait_event(ev)
-                    ext.attn_bwd(qb, kv_t[0], kv_t[1], dob, mk, lse, delta,
-                                 dq, dk_n, dv_n,
-                                 scale, causal, diag, 1, win, lookback is not None,
-                                 softclamp_qk_sim, softclamp_value, False,
-                                 bwd_split, 1)   # dq only
+                # dq + dk/dv kernels (sequential: their LDS footprints do
+                # not co-reside, so stream-splitting buys nothing)
                 ext.attn_bwd(qb, kv_t[0], kv_t[1], dob, mk, lse, delta,
                              dq, dk_n, dv_n,
                              scale, causal, diag, 1, win, lookback is not None,
                              softclamp_qk_sim, softclamp_value, False,
-                             bwd_split, 2)       # dk/dv only
-                ev2 = torch.cuda.Event()
-                ev2.record(s2)
-                # the default stream (and thus the next hop's RCCL exchange,
-                # which waits on it) must not overwrite kv buffers dq reads
-                torch.cuda.current_stream().wait_event(ev2)
+                             bwd_split, 0)
             acc.step(contrib, info.is_last)
 
         dkv = acc.finish(hops)
