@@ -139,3 +139,31 @@ def test_transformer_world2_lookback():
 
 def test_transformer_world4():
     run_distributed(4, _transformer_case, True, False, 1, 65, None)
+
+
+def _var_batch_case(rank, world):
+    """Ranks contribute different batch sizes (reference's --batch-size-var-len)."""
+    torch.manual_seed(13)
+    model_kwargs = dict(
+        num_tokens=64, dim=32, depth=1, causal=True, dim_head=16, heads=2,
+        bucket_size=8, ring_seq_size=16, use_hip_kernel=False,
+    )
+    ring_model = RingTransformer(ring_attn=True, **model_kwargs)
+    flat_model = RingTransformer(ring_attn=False, **model_kwargs)
+    flat_model.load_state_dict(ring_model.state_dict())
+
+    torch.manual_seed(300)
+    sizes = [1 + (r % 2) for r in range(world)]           # e.g. [1, 2]
+    full_ids = torch.randint(0, 64, (sum(sizes), 32))
+    start = sum(sizes[:rank])
+    ids = full_ids[start:start + sizes[rank]]
+
+    logits = ring_model(ids)
+    ref_logits = flat_model(full_ids)
+    err = (logits - ref_logits[start:start + sizes[rank]]).abs().max().item()
+    assert err < 5e-5, f"var-batch logits err {err}"
+    return err
+
+
+def test_transformer_var_batch_world2():
+    run_distributed(2, _var_batch_case)
