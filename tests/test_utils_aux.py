@@ -59,3 +59,17 @@ def test_checkpoint_roundtrip():
         loss1 = model(ids, return_loss=True)
         loss2 = model2(ids, return_loss=True)
         assert torch.allclose(loss1, loss2)
+
+
+def test_watchdog_fires_and_recovers():
+    from ring_attention_amd.parallel.watchdog import Watchdog
+    import time
+    fired = []
+    wd = Watchdog(stall_s=0.2, check_every_s=0.05,
+                  on_stall=lambda step, el: fired.append((step, el)))
+    with wd:
+        wd.tick(1)
+        time.sleep(0.5)            # stall -> fires once
+        assert wd.stalled and len(fired) == 1 and fired[0][0] == 1
+        wd.tick(2)                 # progress resumes
+        assert not wd.stalled
