@@ -110,4 +110,16 @@ struct DecodeParams {
 
 void launch_decode_partial(const DecodeParams& p, int head_dim, hipStream_t stream);
 
+struct RotaryParams {
+    const void* x;      // bf16 (B, N, H, D)
+    const float* cos_t; // fp32 (N, D/2) host-precomputed table
+    const float* sin_t; // fp32 (N, D/2)
+    void* out;          // bf16 (B, N, H, D)
+    long rows;          // B * N * H
+    int n, h, d;
+    float sin_sign;     // +1 forward, -1 inverse (backward)
+};
+
+void launch_rotary(const RotaryParams& p, int head_dim, hipStream_t stream);
+
 }  // namespace ring_attn

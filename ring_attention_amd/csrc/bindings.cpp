@@ -158,6 +158,27 @@ void attn_bwd(
     TORCH_CHECK(hipGetLastError() == hipSuccess, "attn_bwd launch failed");
 }
 
+at::Tensor rotary_apply(at::Tensor x, at::Tensor cos_t, at::Tensor sin_t, double sin_sign) {
+    // x bf16 (B, N, H, D); cos/sin fp32 (N, D/2)
+    CHECK_BF16_CONTIG(x);
+    CHECK_F32_CONTIG(cos_t); CHECK_F32_CONTIG(sin_t);
+    const int64_t B = x.size(0), N = x.size(1), H = x.size(2), D = x.size(3);
+    TORCH_CHECK(D == 64 || D == 128, "head dim must be 64 or 128");
+    TORCH_CHECK(cos_t.size(0) == N && cos_t.size(1) == D / 2);
+    auto out = at::empty_like(x);
+    RotaryParams p{};
+    p.x = x.data_ptr();
+    p.cos_t = cos_t.data_ptr<float>();
+    p.sin_t = sin_t.data_ptr<float>();
+    p.out = out.data_ptr();
+    p.rows = B * N * H;
+    p.n = (int)N; p.h = (int)H; p.d = (int)D;
+    p.sin_sign = (float)sin_sign;
+    launch_rotary(p, (int)D, at::hip::getCurrentHIPStream());
+    TORCH_CHECK(hipGetLastError() == hipSuccess, "rotary launch failed");
+    return out;
+}
+
 std::vector<at::Tensor> decode_partial(at::Tensor q, at::Tensor k, at::Tensor v) {
     // q (B,H,1,D); k,v (B,H,N,D) bf16 -> (out fp32 (B,H,1,D), lse fp32 (B,H,1,1))
     CHECK_BF16_CONTIG(q); CHECK_BF16_CONTIG(k); CHECK_BF16_CONTIG(v);
@@ -182,4 +203,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
     mod.def("attn_fwd_merge", &ring_attn::attn_fwd_merge, "merge kv-split partials");
     mod.def("attn_bwd", &ring_attn::attn_bwd, "CDNA4 flash attention backward");
     mod.def("decode_partial", &ring_attn::decode_partial, "CDNA4 single-query decode partial");
+    mod.def("rotary_apply", &ring_attn::rotary_apply, "fused rotary embedding (table-driven)");
 }

@@ -31,13 +31,16 @@ from .rotary import RingRotaryEmbedding, apply_rotary_pos_emb
 
 
 class RMSNorm(nn.Module):
+    """x / rms(x) * gamma — via torch's fused rms_norm kernel (one pass;
+    the reference composed normalize + 2 muls, ring_attention.py:470-477)."""
+
     def __init__(self, dim: int):
         super().__init__()
-        self.scale = dim ** 0.5
+        self.dim = dim
         self.gamma = nn.Parameter(torch.ones(dim))
 
     def forward(self, x: Tensor) -> Tensor:
-        return torch.nn.functional.normalize(x, dim=-1) * self.scale * self.gamma
+        return torch.nn.functional.rms_norm(x, (self.dim,), self.gamma, eps=1e-24)
 
 
 class RingAttention(nn.Module):

@@ -55,9 +55,37 @@ def rotate_half(x: Tensor) -> Tensor:
     return torch.cat((-x2, x1), dim=-1)
 
 
+class _FusedRotary(torch.autograd.Function):
+    """One-kernel rotary on GPU (csrc/rotary.hip).  The rotation is
+    orthogonal, so backward is the same kernel with sin negated."""
+
+    @staticmethod
+    def forward(ctx, t, cos_t, sin_t):
+        from ..ops import hip_ext
+        out = hip_ext.require().rotary_apply(t.contiguous(), cos_t, sin_t, 1.0)
+        ctx.save_for_backward(cos_t, sin_t)
+        return out
+
+    @staticmethod
+    def backward(ctx, grad):
+        from ..ops import hip_ext
+        cos_t, sin_t = ctx.saved_tensors
+        g = hip_ext.require().rotary_apply(
+            grad.to(torch.bfloat16).contiguous(), cos_t, sin_t, -1.0)
+        return g, None, None
+
+
 @torch.autocast("cuda", enabled=False)
 def apply_rotary_pos_emb(freqs: Tensor, t: Tensor) -> Tensor:
     """freqs (n, d); t (b, n, h, d) -> rotated t (same dtype as input)."""
+    d = t.shape[-1]
+    if (t.is_cuda and t.dtype == torch.bfloat16 and d in (64, 128)
+            and t.dim() == 4):
+        from ..ops import hip_ext
+        if hip_ext.available():
+            half = freqs[:, :d // 2].float()
+            return _FusedRotary.apply(t, half.cos().contiguous(),
+                                      half.sin().contiguous())
     dtype = t.dtype
     t = t.float()
     f = freqs[None, :, None, :]

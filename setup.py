@@ -24,6 +24,7 @@ ext = CUDAExtension(
         os.path.join(CSRC, "attn_fwd.hip"),
         os.path.join(CSRC, "attn_bwd.hip"),
         os.path.join(CSRC, "decode.hip"),
+        os.path.join(CSRC, "rotary.hip"),
     ],
     extra_compile_args={
         "cxx": ["-O3", "-std=c++17"],

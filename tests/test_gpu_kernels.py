@@ -282,3 +282,29 @@ def test_strided_q_positions_kernel():
                          (dv_n.permute(0, 3, 1, 2).cpu(), vc.grad, "dv")):
         e = (gt - rt).abs().max().item()
         assert e / (rt.abs().max().item() + 1e-6) < 4e-2, f"{name} err {e}"
+
+
+def test_fused_rotary_parity():
+    from ring_attention_amd.models.rotary import apply_rotary_pos_emb, rotate_half
+    torch.manual_seed(11)
+    for d in (64, 128):
+        b, n, h = 2, 128, 4
+        t = torch.randn(b, n, h, d, device="cuda", dtype=torch.bfloat16,
+                        requires_grad=True)
+        freqs = torch.randn(n, d // 2, device="cuda").repeat(1, 2) * 3
+        out = apply_rotary_pos_emb(freqs, t)            # fused kernel path
+        # reference math in fp32
+        tf = t.detach().float()
+        f = freqs[None, :, None, :].float()
+        ref = tf * f.cos() + rotate_half(tf) * f.sin()
+        assert (out.float() - ref).abs().max().item() < 2e-2
+        g = torch.randn_like(out)
+        out.backward(g)
+        # backward = inverse rotation of g
+        gf = g.float()
+        # inverse rotation R^T g: [g1*c + g2*s, g2*c - g1*s]
+        g1, g2 = gf.chunk(2, dim=-1)
+        c = f.cos().chunk(2, dim=-1)[0]
+        s = f.sin().chunk(2, dim=-1)[0]
+        ref_grad = torch.cat((g1 * c + g2 * s, g2 * c - g1 * s), dim=-1)
+        assert (t.grad.float() - ref_grad).abs().max().item() < 2e-2
