@@ -32,6 +32,8 @@ def main():
     ap.add_argument("--warmup", type=int, default=5)
     ap.add_argument("--seq-per-gpu", type=int, default=8192)
     ap.add_argument("--heads", type=int, default=8)
+    ap.add_argument("--kv-heads", type=int, default=None,
+                    help="GQA: kv heads (default = heads)")
     ap.add_argument("--d-head", type=int, default=64)
     ap.add_argument("--batch", type=int, default=1)
     ap.add_argument("--causal", action="store_true")
@@ -63,10 +65,11 @@ def main():
         from ring_attention_amd.ops.ring_flash import ring_flash_attn_ as attn
 
     b, n, h, d = args.batch, args.seq_per_gpu, args.heads, args.d_head
+    hk = args.kv_heads if args.kv_heads is not None else h
     torch.manual_seed(1234 + rank)
     q = torch.randn(b, n, h, d, device=device, dtype=dtype, requires_grad=True)
-    k = torch.randn(b, n, h, d, device=device, dtype=dtype, requires_grad=True)
-    v = torch.randn(b, n, h, d, device=device, dtype=dtype, requires_grad=True)
+    k = torch.randn(b, n, hk, d, device=device, dtype=dtype, requires_grad=True)
+    v = torch.randn(b, n, hk, d, device=device, dtype=dtype, requires_grad=True)
 
     def step():
         out, _ = attn(q, k, v, causal=args.causal, ring_reduce_col=True,
@@ -129,6 +132,7 @@ def main():
                 "seq_len": n_total,
                 "seq_per_gpu": n,
                 "heads": h,
+                "kv_heads": hk,
                 "d_head": d,
                 "parallelism": f"ring{world}",
                 "flop_convention": "fwd=4*b*n_shard*n_total*d*h (/2 causal); fwd+bwd=2.5x",

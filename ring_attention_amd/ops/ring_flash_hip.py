@@ -275,9 +275,13 @@ class RingFlashAttentionHIPFunction(Function):
             diag = rq if striped else rq * n
             dk_full = torch.zeros(b, hk, n_total, d, device=qb.device, dtype=torch.float32)
             dv_full = torch.zeros(b, hk, d, n_total, device=qb.device, dtype=torch.float32)
+            # grid.z split keeps the row-parallel dq kernel filling all CUs
+            # when the local q grid is small (atomic fp32 adds, contention=split)
+            qtiles = (n + 255) // 256
+            split = min(8, max(1, 384 // max(1, qtiles * b * h)))
             ext.attn_bwd(qb, k_full, v_full, dob, m_full, lse, delta,
                          dq, dk_full, dv_full, scale, causal, diag, q_stride,
-                         0, False, softclamp_qk_sim, softclamp_value, False, 1, 0)
+                         0, False, softclamp_qk_sim, softclamp_value, False, split, 0)
             # ONE reduce-scatter returns each rank's dk/dv shard (summed)
             dk_chunks = _scatter_chunks_of_global(dk_full, R, striped, dim=2)
             dv_chunks = _scatter_chunks_of_global(dv_full, R, striped, dim=3)

@@ -1,3 +1,5 @@
+"""Diagnostic: forward kernel with staging ablated (stage-once) vs full.
+Usage (GPU box): python tools/ablate_bench.py"""
 import sys, time, torch
 sys.path.insert(0, "/root/repo")
 from ring_attention_amd.ops import hip_ext
@@ -11,11 +13,13 @@ lse = torch.empty(b, h, n, device="cuda", dtype=torch.float32)
 scale = d ** -0.5
 def run(ablate, iters=30):
     for _ in range(5):
-        ext.attn_fwd(q, k, v, None, None, None, None, out, lse, scale, False, 0, 0, False, False, 50.0, True, True, 1, ablate)
+        ext.attn_fwd(q, k, v, None, None, None, None, out, lse, scale,
+                     False, 0, 1, 0, False, False, 50.0, True, True, 1, ablate)
     torch.cuda.synchronize(); t0 = time.perf_counter()
     for _ in range(iters):
-        ext.attn_fwd(q, k, v, None, None, None, None, out, lse, scale, False, 0, 0, False, False, 50.0, True, True, 1, ablate)
+        ext.attn_fwd(q, k, v, None, None, None, None, out, lse, scale,
+                     False, 0, 1, 0, False, False, 50.0, True, True, 1, ablate)
     torch.cuda.synchronize()
     return (time.perf_counter() - t0) / iters * 1e6
-print(f"full kernel:      {run(0):8.1f} us")
-print(f"stage-once (no HBM/LDS-write steady-state): {run(1):8.1f} us")
+print(f"full kernel:  {run(0):8.1f} us")
+print(f"stage-once:   {run(1):8.1f} us")
