@@ -275,13 +275,18 @@ class RingFlashAttentionHIPFunction(Function):
             diag = rq if striped else rq * n
             dk_full = torch.zeros(b, hk, n_total, d, device=qb.device, dtype=torch.float32)
             dv_full = torch.zeros(b, hk, d, n_total, device=qb.device, dtype=torch.float32)
-            # grid.z split keeps the row-parallel dq kernel filling all CUs
-            # when the local q grid is small (atomic fp32 adds, contention=split)
+            # grid.z splits keep both kernels filling the CUs when their
+            # natural grids are small (fp32 atomics, contention = split)
             qtiles = (n + 255) // 256
-            split = min(8, max(1, 384 // max(1, qtiles * b * h)))
+            kvtiles_t = (n_total + 255) // 256
+            split_dq = min(8, max(1, 384 // max(1, qtiles * b * h)))
+            split_dkv = min(8, max(1, 384 // max(1, kvtiles_t * b * hk)))
             ext.attn_bwd(qb, k_full, v_full, dob, m_full, lse, delta,
                          dq, dk_full, dv_full, scale, causal, diag, q_stride,
-                         0, False, softclamp_qk_sim, softclamp_value, False, split, 0)
+                         0, False, softclamp_qk_sim, softclamp_value, False, split_dq, 1)
+            ext.attn_bwd(qb, k_full, v_full, dob, m_full, lse, delta,
+                         dq, dk_full, dv_full, scale, causal, diag, q_stride,
+                         0, False, softclamp_qk_sim, softclamp_value, False, split_dkv, 2)
             # ONE reduce-scatter returns each rank's dk/dv shard (summed)
             dk_chunks = _scatter_chunks_of_global(dk_full, R, striped, dim=2)
             dv_chunks = _scatter_chunks_of_global(dv_full, R, striped, dim=3)
@@ -292,8 +297,11 @@ class RingFlashAttentionHIPFunction(Function):
             dv_home = own[half:].view(b, hk, d, n).permute(0, 3, 1, 2).contiguous()
             return (dq.to(in_dtype), dk_home.to(in_dtype), dv_home.to(in_dtype),
                     None, None, None, None, None, None, None, None, None)
+        # independent grid.z splits: dq's grid is (q-tiles x b*h), dkv's is
+        # (kv-tiles x b*hk) — GQA shrinks the latter (e.g. hk=2 -> 64 WGs)
         qtiles = (n + 255) // 256
-        bwd_split = min(8, max(1, 192 // max(1, qtiles * b * h)))
+        split_dq = min(8, max(1, 384 // max(1, qtiles * b * h)))
+        split_dkv = min(8, max(1, 384 // max(1, qtiles * b * hk)))
 
         kv = torch.stack((kb, vb))
         ring_tensors = (kv,) if mask_u8 is None else (kv, mask_u8)
@@ -319,7 +327,12 @@ class RingFlashAttentionHIPFunction(Function):
                              dq, dk_n, dv_n,
                              scale, causal, diag, 1, win, lookback is not None,
                              softclamp_qk_sim, softclamp_value, False,
-                             bwd_split, 0)
+                             split_dq, 1)
+                ext.attn_bwd(qb, kv_t[0], kv_t[1], dob, mk, lse, delta,
+                             dq, dk_n, dv_n,
+                             scale, causal, diag, 1, win, lookback is not None,
+                             softclamp_qk_sim, softclamp_value, False,
+                             split_dkv, 2)
             acc.step(contrib, info.is_last)
 
         dkv = acc.finish(hops)
