@@ -308,3 +308,39 @@ def test_fused_rotary_parity():
         s = f.sin().chunk(2, dim=-1)[0]
         ref_grad = torch.cat((g1 * c + g2 * s, g2 * c - g1 * s), dim=-1)
         assert (t.grad.float() - ref_grad).abs().max().item() < 2e-2
+
+
+def test_kv_split_merge_resume_across_hops():
+    """Ring-resume + kv-split interplay: two simulated hops, each computed as
+    kv-split partials folded into the running accumulator by the merge kernel."""
+    from ring_attention_amd.ops import hip_ext
+    ext = hip_ext.require()
+    b, n, h, d = 1, 512, 2, 64
+    q, k, v = _mk(b, n, h, h, d, seed=12)
+    half = n // 2
+    scale = d ** -0.5
+    S = 2  # kv_split within each hop
+
+    out = torch.empty_like(q)
+    lse = torch.empty(b, h, n, device="cuda", dtype=torch.float32)
+    o_acc = torch.empty(b, h, d, n, device="cuda", dtype=torch.float32)
+    m = torch.empty(b, h, n, device="cuda", dtype=torch.float32)
+    l = torch.empty(b, h, n, device="cuda", dtype=torch.float32)
+    o_p = torch.empty(S, b, h, d, n, device="cuda", dtype=torch.float32)
+    m_p = torch.empty(S, b, h, n, device="cuda", dtype=torch.float32)
+    l_p = torch.empty(S, b, h, n, device="cuda", dtype=torch.float32)
+
+    for hop, (ks, vs, diag) in enumerate((
+            (k[:, :half], v[:, :half], 0),
+            (k[:, half:], v[:, half:], -half))):
+        first, last = hop == 0, hop == 1
+        ext.attn_fwd(q, ks.contiguous(), vs.contiguous(), None, o_p, m_p, l_p,
+                     None, None, scale, True, diag, 1, 0, False, False, 50.0,
+                     first, last, S, 0)
+        ext.attn_fwd_merge(o_p, m_p, l_p, o_acc, m, l,
+                           out if last else None, lse if last else None,
+                           S, b, h, d, n, first, last)
+
+    _, _, _, ref, ref_lse = _oracle(q, k, v, causal=True)
+    assert (out.float().cpu() - ref).abs().max().item() < 2e-2
+    assert (lse.cpu() - ref_lse).abs().max().item() < 2e-3
