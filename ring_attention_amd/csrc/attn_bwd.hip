@@ -291,36 +291,60 @@ __global__ __launch_bounds__(512, 1) void attn_bwd_dq_kernel(BwdParams p) {
         if (t + 1 < t_hi) write_tile(par ^ 1);
         if (t + 2 < t_hi) load_tile();
 
-        // ds^T in regs -> packed bf16 pairs (exp2 domain)
+        // ds^T in regs -> packed bf16 pairs (exp2 domain); the full-tile
+        // variant carries no per-element predicates (see fwd kernel note)
         uint32_t pk[DQ_NBLK * 8];
-        #pragma unroll
-        for (int x2 = 0; x2 < DQ_NBLK * 8; ++x2) {
-            float dse[2];
+        if (full_tile && row_valid) {
             #pragma unroll
-            for (int e = 0; e < 2; ++e) {
-                int kb = x2 >> 3, r = (2 * x2 + e) & 15;
-                long jj = j0 + kb * 32 + (r & 3) + 8 * (r >> 2) + 4 * lhi;
-                float x, dtanh = 1.f;
-                if constexpr (SOFTCLAMP) {
-                    float inv_v = __builtin_amdgcn_rcpf(p.softclamp_value);
-                    float th = bfast_tanhf(s[kb][r] * p.scale * inv_v);
-                    x = p.softclamp_value * th * 1.4426950408889634f;
-                    dtanh = 1.f - th * th;
-                } else {
-                    x = s[kb][r] * scale2;
+            for (int x2 = 0; x2 < DQ_NBLK * 8; ++x2) {
+                float dse[2];
+                #pragma unroll
+                for (int e = 0; e < 2; ++e) {
+                    int kb = x2 >> 3, r = (2 * x2 + e) & 15;
+                    float x, dtanh = 1.f;
+                    if constexpr (SOFTCLAMP) {
+                        float inv_v = __builtin_amdgcn_rcpf(p.softclamp_value);
+                        float th = bfast_tanhf(s[kb][r] * p.scale * inv_v);
+                        x = p.softclamp_value * th * 1.4426950408889634f;
+                        dtanh = 1.f - th * th;
+                    } else {
+                        x = s[kb][r] * scale2;
+                    }
+                    float pv = __builtin_amdgcn_exp2f(x - lse_i);
+                    dse[e] = pv * (dp[kb][r] - delta_i) * dtanh * p.scale;
                 }
-                bool ok = row_valid && jj <= jmax;
-                if (!full_tile) {
+                union { __hip_bfloat162 h2; uint32_t u; } cvt;
+                cvt.h2 = __float22bfloat162_rn(float2{dse[0], dse[1]});
+                pk[x2] = cvt.u;
+            }
+        } else {
+            #pragma unroll
+            for (int x2 = 0; x2 < DQ_NBLK * 8; ++x2) {
+                float dse[2];
+                #pragma unroll
+                for (int e = 0; e < 2; ++e) {
+                    int kb = x2 >> 3, r = (2 * x2 + e) & 15;
+                    long jj = j0 + kb * 32 + (r & 3) + 8 * (r >> 2) + 4 * lhi;
+                    float x, dtanh = 1.f;
+                    if constexpr (SOFTCLAMP) {
+                        float inv_v = __builtin_amdgcn_rcpf(p.softclamp_value);
+                        float th = bfast_tanhf(s[kb][r] * p.scale * inv_v);
+                        x = p.softclamp_value * th * 1.4426950408889634f;
+                        dtanh = 1.f - th * th;
+                    } else {
+                        x = s[kb][r] * scale2;
+                    }
+                    bool ok = row_valid && jj <= jmax;
                     if (p.causal) ok = ok && (jj <= qpos_i);
                     if (p.has_win) ok = ok && (qpos_i - jj <= p.win);
                     if (p.kmask) ok = ok && lds.kmask[par][jj - j0];
+                    float pv = ok ? __builtin_amdgcn_exp2f(x - lse_i) : 0.f;
+                    dse[e] = pv * (dp[kb][r] - delta_i) * dtanh * p.scale;
                 }
-                float pv = ok ? __builtin_amdgcn_exp2f(x - lse_i) : 0.f;
-                dse[e] = pv * (dp[kb][r] - delta_i) * dtanh * p.scale;
+                union { __hip_bfloat162 h2; uint32_t u; } cvt;
+                cvt.h2 = __float22bfloat162_rn(float2{dse[0], dse[1]});
+                pk[x2] = cvt.u;
             }
-            union { __hip_bfloat162 h2; uint32_t u; } cvt;
-            cvt.h2 = __float22bfloat162_rn(float2{dse[0], dse[1]});
-            pk[x2] = cvt.u;
         }
 
         // build B-operand fragments (lane = q, k = kv contiguous): pairs +2
@@ -598,39 +622,65 @@ __global__ __launch_bounds__(512, 1) void attn_bwd_dkv_kernel(BwdParams p) {
                 }
 
                 uint32_t p_pk[8], ds_pk[8];
-                #pragma unroll
-                for (int x2 = 0; x2 < 8; ++x2) {
-                    float pe[2], dse[2];
+                if (full_tile && col_valid) {
                     #pragma unroll
-                    for (int e = 0; e < 2; ++e) {
-                        int r = 2 * x2 + e;
-                        int qloc = qb * 32 + (r & 3) + 8 * (r >> 2) + 4 * lhi;
-                        long i = i0 + qloc;
-                        float x, dtanh = 1.f;
-                        if constexpr (SOFTCLAMP) {
-                            float inv_v = __builtin_amdgcn_rcpf(p.softclamp_value);
-                            float th = bfast_tanhf(s2[r] * p.scale * inv_v);
-                            x = p.softclamp_value * th * 1.4426950408889634f;
-                            dtanh = 1.f - th * th;
-                        } else {
-                            x = s2[r] * (p.scale * 1.4426950408889634f);
+                    for (int x2 = 0; x2 < 8; ++x2) {
+                        float pe[2], dse[2];
+                        #pragma unroll
+                        for (int e = 0; e < 2; ++e) {
+                            int r = 2 * x2 + e;
+                            float x, dtanh = 1.f;
+                            if constexpr (SOFTCLAMP) {
+                                float inv_v = __builtin_amdgcn_rcpf(p.softclamp_value);
+                                float th = bfast_tanhf(s2[r] * p.scale * inv_v);
+                                x = p.softclamp_value * th * 1.4426950408889634f;
+                                dtanh = 1.f - th * th;
+                            } else {
+                                x = s2[r] * (p.scale * 1.4426950408889634f);
+                            }
+                            float pv = __builtin_amdgcn_exp2f(x - lse_r[r]);
+                            pe[e] = pv;
+                            dse[e] = pv * (dp[r] - delta_r[r]) * dtanh * p.scale;
                         }
-                        bool ok = col_valid && i <= imax;
-                        if (!full_tile) {
+                        union { __hip_bfloat162 h2; uint32_t u; } c1, c2;
+                        c1.h2 = __float22bfloat162_rn(float2{pe[0], pe[1]});
+                        c2.h2 = __float22bfloat162_rn(float2{dse[0], dse[1]});
+                        p_pk[x2] = c1.u;
+                        ds_pk[x2] = c2.u;
+                    }
+                } else {
+                    #pragma unroll
+                    for (int x2 = 0; x2 < 8; ++x2) {
+                        float pe[2], dse[2];
+                        #pragma unroll
+                        for (int e = 0; e < 2; ++e) {
+                            int r = 2 * x2 + e;
+                            int qloc = qb * 32 + (r & 3) + 8 * (r >> 2) + 4 * lhi;
+                            long i = i0 + qloc;
+                            float x, dtanh = 1.f;
+                            if constexpr (SOFTCLAMP) {
+                                float inv_v = __builtin_amdgcn_rcpf(p.softclamp_value);
+                                float th = bfast_tanhf(s2[r] * p.scale * inv_v);
+                                x = p.softclamp_value * th * 1.4426950408889634f;
+                                dtanh = 1.f - th * th;
+                            } else {
+                                x = s2[r] * (p.scale * 1.4426950408889634f);
+                            }
+                            bool ok = col_valid && i <= imax;
                             long qpos = i * p.q_stride + p.diag;
                             if (p.causal) ok = ok && (j <= qpos);
                             if (p.has_win) ok = ok && (qpos - j <= p.win);
                             if (p.kmask) ok = ok && kmask_own;
+                            float pv = ok ? __builtin_amdgcn_exp2f(x - lse_r[r]) : 0.f;
+                            pe[e] = pv;
+                            dse[e] = pv * (dp[r] - delta_r[r]) * dtanh * p.scale;
                         }
-                        float pv = ok ? __builtin_amdgcn_exp2f(x - lse_r[r]) : 0.f;
-                        pe[e] = pv;
-                        dse[e] = pv * (dp[r] - delta_r[r]) * dtanh * p.scale;
+                        union { __hip_bfloat162 h2; uint32_t u; } c1, c2;
+                        c1.h2 = __float22bfloat162_rn(float2{pe[0], pe[1]});
+                        c2.h2 = __float22bfloat162_rn(float2{dse[0], dse[1]});
+                        p_pk[x2] = c1.u;
+                        ds_pk[x2] = c2.u;
                     }
-                    union { __hip_bfloat162 h2; uint32_t u; } c1, c2;
-                    c1.h2 = __float22bfloat162_rn(float2{pe[0], pe[1]});
-                    c2.h2 = __float22bfloat162_rn(float2{dse[0], dse[1]});
-                    p_pk[x2] = c1.u;
-                    ds_pk[x2] = c2.u;
                 }
 
                 // fragments (lane = kv, k = q contiguous)

@@ -287,31 +287,72 @@ __global__ __launch_bounds__(NTHREADS, 1) void attn_fwd_kernel(FwdParams p) {
             if (t + 2 < t_hi) load_tile();
         }
 
-        // ---- scale (exp2 domain), clamp, mask in place
-        float smax = MASK_VALUE_F;
-        #pragma unroll
-        for (int kb = 0; kb < NBLK; ++kb)
+        if (p.ablate == 2) {
+            // diagnostics: skip softmax VALU, feed PV garbage fragments kept
+            // alive via asm (rule 17: a skipped phase must not DCE upstream)
+            uint32_t gfrag[4];
             #pragma unroll
-            for (int r = 0; r < 16; ++r) {
-                float x;
-                if constexpr (SOFTCLAMP) {
-                    float xs = s[kb][r] * (p.scale * __builtin_amdgcn_rcpf(p.softclamp_value));
-                    xs = p.softclamp_value * fast_tanhf(xs);
-                    x = xs * LOG2E;
-                } else {
-                    x = s[kb][r] * scale2;
+            for (int c = 0; c < 4; ++c) {
+                union { float f; uint32_t u; } cv; cv.f = s[0][c];
+                gfrag[c] = cv.u;
+                asm volatile("" :: "v"(gfrag[c]));
+            }
+            #pragma unroll
+            for (int db = 0; db < DBLK; ++db) {
+                int drow = db * 32 + l31;
+                #pragma unroll
+                for (int ks = 0; ks < NBLK * 2; ++ks) {
+                    int chunk = ks * 2 + lhi;
+                    bf16x8 vf = *(const bf16x8*)(lds.vt[par] + drow * KVBLK + swz(drow, chunk) * 8);
+                    o_acc[db] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+                        vf, *(const bf16x8*)gfrag, o_acc[db], 0, 0, 0);
                 }
-                if (!full_tile) {
+            }
+            continue;
+        }
+
+        // ---- scale (exp2 domain), clamp, mask in place.  The full-tile
+        // variant must contain NO per-element conditions: a condition inside
+        // the unrolled loop is PREDICATED (cndmask per element, executed on
+        // every tile) rather than branched — hoist to one scalar branch.
+        float smax = MASK_VALUE_F;
+        if (full_tile) {
+            #pragma unroll
+            for (int kb = 0; kb < NBLK; ++kb)
+                #pragma unroll
+                for (int r = 0; r < 16; ++r) {
+                    float x;
+                    if constexpr (SOFTCLAMP) {
+                        float xs = s[kb][r] * (p.scale * __builtin_amdgcn_rcpf(p.softclamp_value));
+                        x = p.softclamp_value * fast_tanhf(xs) * LOG2E;
+                    } else {
+                        x = s[kb][r] * scale2;
+                    }
+                    s[kb][r] = x;
+                    smax = fmaxf(smax, x);
+                }
+        } else {
+            #pragma unroll
+            for (int kb = 0; kb < NBLK; ++kb)
+                #pragma unroll
+                for (int r = 0; r < 16; ++r) {
+                    float x;
+                    if constexpr (SOFTCLAMP) {
+                        float xs = s[kb][r] * (p.scale * __builtin_amdgcn_rcpf(p.softclamp_value));
+                        x = p.softclamp_value * fast_tanhf(xs) * LOG2E;
+                    } else {
+                        x = s[kb][r] * scale2;
+                    }
                     long j = j0 + kb * 32 + (r & 3) + 8 * (r >> 2) + 4 * lhi;
                     bool ok = j <= jmax;
                     if (p.causal) ok = ok && (j <= qpos_i);
                     if (p.has_win) ok = ok && (qpos_i - j <= p.win);
                     if (p.kmask) ok = ok && lds.kmask[par][j - j0];
                     if (!ok) x = MASK_VALUE_F;
+                    s[kb][r] = x;
+                    smax = fmaxf(smax, x);
                 }
-                s[kb][r] = x;
-                smax = fmaxf(smax, x);
-            }
+        }
         smax = fmaxf(smax, __shfl_xor(smax, 32));
 
         // ---- online softmax update (defer-max THR=0: exact — skip the O

@@ -23,3 +23,4 @@ def run(ablate, iters=30):
     return (time.perf_counter() - t0) / iters * 1e6
 print(f"full kernel:  {run(0):8.1f} us")
 print(f"stage-once:   {run(1):8.1f} us")
+print(f"no-softmax:   {run(2):8.1f} us")
