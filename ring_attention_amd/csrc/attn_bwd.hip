@@ -308,9 +308,9 @@ __global__ __launch_bounds__(512, 1) void attn_bwd_dq_kernel(BwdParams p) {
                         x = p.softclamp_value * th * 1.4426950408889634f;
                         dtanh = 1.f - th * th;
                     } else {
-                        x = s[kb][r] * scale2;
+                        x = __builtin_fmaf(s[kb][r], scale2, -lse_i);  // fold
                     }
-                    float pv = __builtin_amdgcn_exp2f(x - lse_i);
+                    float pv = __builtin_amdgcn_exp2f(SOFTCLAMP ? x - lse_i : x);
                     dse[e] = pv * (dp[kb][r] - delta_i) * dtanh * p.scale;
                 }
                 union { __hip_bfloat162 h2; uint32_t u; } cvt;
@@ -636,9 +636,10 @@ __global__ __launch_bounds__(512, 1) void attn_bwd_dkv_kernel(BwdParams p) {
                                 x = p.softclamp_value * th * 1.4426950408889634f;
                                 dtanh = 1.f - th * th;
                             } else {
-                                x = s2[r] * (p.scale * 1.4426950408889634f);
+                                x = __builtin_fmaf(s2[r], p.scale * 1.4426950408889634f,
+                                                   -lse_r[r]);          // fold
                             }
-                            float pv = __builtin_amdgcn_exp2f(x - lse_r[r]);
+                            float pv = __builtin_amdgcn_exp2f(SOFTCLAMP ? x - lse_r[r] : x);
                             pe[e] = pv;
                             dse[e] = pv * (dp[r] - delta_r[r]) * dtanh * p.scale;
                         }

@@ -317,20 +317,27 @@ __global__ __launch_bounds__(NTHREADS, 1) void attn_fwd_kernel(FwdParams p) {
         // every tile) rather than branched — hoist to one scalar branch.
         float smax = MASK_VALUE_F;
         if (full_tile) {
-            #pragma unroll
-            for (int kb = 0; kb < NBLK; ++kb)
+            if constexpr (!SOFTCLAMP) {
+                // exp2+fma fold: fmax runs on RAW scores (scale2 > 0 is
+                // monotone); the scale is folded into the exp argument as a
+                // single v_fma below — saves one VALU per element
                 #pragma unroll
-                for (int r = 0; r < 16; ++r) {
-                    float x;
-                    if constexpr (SOFTCLAMP) {
+                for (int kb = 0; kb < NBLK; ++kb)
+                    #pragma unroll
+                    for (int r = 0; r < 16; ++r)
+                        smax = fmaxf(smax, s[kb][r]);
+                smax *= scale2;
+            } else {
+                #pragma unroll
+                for (int kb = 0; kb < NBLK; ++kb)
+                    #pragma unroll
+                    for (int r = 0; r < 16; ++r) {
                         float xs = s[kb][r] * (p.scale * __builtin_amdgcn_rcpf(p.softclamp_value));
-                        x = p.softclamp_value * fast_tanhf(xs) * LOG2E;
-                    } else {
-                        x = s[kb][r] * scale2;
+                        float x = p.softclamp_value * fast_tanhf(xs) * LOG2E;
+                        s[kb][r] = x;
+                        smax = fmaxf(smax, x);
                     }
-                    s[kb][r] = x;
-                    smax = fmaxf(smax, x);
-                }
+            }
         } else {
             #pragma unroll
             for (int kb = 0; kb < NBLK; ++kb)
@@ -359,12 +366,16 @@ __global__ __launch_bounds__(NTHREADS, 1) void attn_fwd_kernel(FwdParams p) {
         // rescale whenever the running max did not grow on any lane)
         float m_new = fmaxf(m_run, smax);
         const bool any_growth = !__all(smax <= m_run);
+        // full tiles (no softclamp) kept RAW scores: exp2(fma(s, scale2, -m))
+        const float escale = (full_tile && !SOFTCLAMP) ? scale2 : 1.f;
         float rowsum = 0.f;
         uint32_t pk[NBLK * 8];                                      // packed bf16 pairs
         #pragma unroll
         for (int x2 = 0; x2 < NBLK * 8; ++x2) {
-            float e0 = __builtin_amdgcn_exp2f(s[x2 >> 3][(2 * x2) & 15] - m_new);
-            float e1 = __builtin_amdgcn_exp2f(s[x2 >> 3][(2 * x2 + 1) & 15] - m_new);
+            float e0 = __builtin_amdgcn_exp2f(
+                __builtin_fmaf(s[x2 >> 3][(2 * x2) & 15], escale, -m_new));
+            float e1 = __builtin_amdgcn_exp2f(
+                __builtin_fmaf(s[x2 >> 3][(2 * x2 + 1) & 15], escale, -m_new));
             rowsum += e0 + e1;
             union { __hip_bfloat162 h2; uint32_t u; } cvt;
             cvt.h2 = __float22bfloat162_rn(float2{e0, e1});
