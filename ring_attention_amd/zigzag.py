@@ -125,11 +125,25 @@ def zig_zag_attn(
         if kv_valid_len is not None:
             k_ = k_[:, :kv_valid_len]
             v_ = v_[:, :kv_valid_len]
+        head_perm = None
+        if groups > 1:
+            # reference GQA convention tiles kv heads ('b h -> b (g h)': q head
+            # qh pairs with kv head qh % hk); the kernel groups contiguously
+            # (qh // g) — permute q heads kv-major for the kernel, then back
+            # kernel position p -> kv head p // g; slot j of kv head c must
+            # hold original head c + j * hk (the reference's qh % hk pairing)
+            p = torch.arange(heads, device=q.device)
+            head_perm = (p // groups) + (p % groups) * kv_heads
+            inv_perm = torch.empty_like(head_perm)
+            inv_perm[head_perm] = torch.arange(heads, device=q.device)
+            q_ = q_.index_select(2, head_perm)
         outs = []
         for c, start in enumerate(q_chunk_starts):
             qc = q_[:, c * half:(c + 1) * half]
             outs.append(flash_attn_offset(qc, k_, v_, q_offset=start, causal=True))
         out = torch.cat(outs, dim=1)
+        if head_perm is not None:
+            out = out.index_select(2, inv_perm)
         return out.permute(0, 2, 1, 3)
     if groups > 1:
         # repeat pattern parity with the reference: 'b h n d -> b (g h) n d'

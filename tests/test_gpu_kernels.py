@@ -344,3 +344,25 @@ def test_kv_split_merge_resume_across_hops():
     _, _, _, ref, ref_lse = _oracle(q, k, v, causal=True)
     assert (out.float().cpu() - ref).abs().max().item() < 2e-2
     assert (lse.cpu() - ref_lse).abs().max().item() < 2e-3
+
+
+def test_zigzag_fast_path_gqa():
+    """GQA zig-zag fast path must use the reference head pairing (qh % hk)."""
+    from ring_attention_amd.zigzag import zig_zag_attn
+    from ring_attention_amd.ops.reference import MASK_VALUE
+    b, h, hk, n, d = 1, 6, 2, 256, 64
+    torch.manual_seed(14)
+    q = torch.randn(b, h, n, d, device="cuda", dtype=torch.bfloat16)
+    k = torch.randn(b, hk, n, d, device="cuda", dtype=torch.bfloat16)
+    v = torch.randn(b, hk, n, d, device="cuda", dtype=torch.bfloat16)
+    half = n // 2
+    out = zig_zag_attn(q, k, v, causal=True, q_chunk_starts=(0, half))
+
+    kk = k.repeat(1, h // hk, 1, 1).float()   # reference tile pairing
+    vv = v.repeat(1, h // hk, 1, 1).float()
+    sim = torch.einsum("bhid,bhjd->bhij", q.float(), kk) * d ** -0.5
+    pos = torch.arange(n, device="cuda")
+    sim = sim.masked_fill((pos[None, :] > pos[:, None])[None, None], MASK_VALUE)
+    ref = torch.einsum("bhij,bhjd->bhid", sim.softmax(-1), vv)
+    err = (out.float() - ref).abs().max().item()
+    assert err < 2e-2, f"zigzag gqa err {err}"
