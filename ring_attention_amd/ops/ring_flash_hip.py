@@ -42,6 +42,9 @@ def _choose_strategy(use_ring: bool, topo, n, hk, d, lookback, dtype_bytes=2) ->
     shards (used for sliding windows, sub-rings, and beyond-memory seqs)."""
     if not use_ring or topo.ring_size == 1:
         return "local"
+    forced = os.environ.get("RING_ATTN_FORCE_STRATEGY")
+    if forced in ("ring", "allgather") and topo.ring_size == topo.world_size:
+        return forced
     if topo.ring_size != topo.world_size:
         return "ring"        # sub-rings: per-ring communicators (future work)
     if lookback is not None:
