@@ -83,16 +83,6 @@ def _scatter_chunks_of_global(t_full, ring_size: int, striped: bool, dim: int):
     order = [sdim] + [i for i in range(view.ndim) if i != sdim]
     return view.permute(order).contiguous()
 
-_side_stream = None
-
-
-def _get_side_stream():
-    global _side_stream
-    if _side_stream is None:
-        _side_stream = torch.cuda.Stream()
-    return _side_stream
-
-
 def _hop_geometry(rq: int, rk: int, n: int, ring_size: int, striped: bool,
                   causal: bool, lookback: int | None) -> tuple[bool, int, int]:
     """Returns (skip_hop, diag, win) for a (q-rank, kv-source-rank) pair,
