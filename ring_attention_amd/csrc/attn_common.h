@@ -46,6 +46,8 @@ struct FwdParams {
     int kv_split;           // >1: grid.z splits the kv range; o_acc/m/l hold
                             // kv_split partials (merged by attn_fwd_merge)
     int ablate;             // diagnostics: 1 = stage first tile only
+    unsigned long long* ticks;  // diagnostics: per-tile segment s_memtime
+                                // stamps from block(0,0,0) wave 0 (or null)
 };
 
 struct FwdMergeParams {

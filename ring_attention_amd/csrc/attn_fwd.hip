@@ -263,6 +263,9 @@ __global__ __launch_bounds__(NTHREADS, 1) void attn_fwd_kernel(FwdParams p) {
             !p.kmask;
 
         __syncthreads();
+        const bool stamp = p.ticks && blockIdx.x == 0 && bh == 0 && tid == 0
+                           && blockIdx.z == 0;
+        if (stamp) p.ticks[t * 6 + 0] = __builtin_amdgcn_s_memtime();
 
         // ---- QK^T: S^T[kv][q] for the NBLK 32-row kv blocks
         f32x16 s[NBLK];
@@ -280,12 +283,14 @@ __global__ __launch_bounds__(NTHREADS, 1) void attn_fwd_kernel(FwdParams p) {
             }
         }
         __builtin_amdgcn_s_setprio(0);
+        if (stamp) p.ticks[t * 6 + 1] = __builtin_amdgcn_s_memtime();
 
         // stage tile t+1 into the other buffer while the MFMAs above retire
         if (p.ablate != 1) {
             if (t + 1 < t_hi) write_tile(par ^ 1);
             if (t + 2 < t_hi) load_tile();
         }
+        if (stamp) p.ticks[t * 6 + 2] = __builtin_amdgcn_s_memtime();
 
         if (p.ablate == 2) {
             // diagnostics: skip softmax VALU, feed PV garbage fragments kept
@@ -361,6 +366,7 @@ __global__ __launch_bounds__(NTHREADS, 1) void attn_fwd_kernel(FwdParams p) {
                 }
         }
         smax = fmaxf(smax, __shfl_xor(smax, 32));
+        if (stamp) p.ticks[t * 6 + 3] = __builtin_amdgcn_s_memtime();
 
         // ---- online softmax update (defer-max THR=0: exact — skip the O
         // rescale whenever the running max did not grow on any lane)
@@ -368,19 +374,25 @@ __global__ __launch_bounds__(NTHREADS, 1) void attn_fwd_kernel(FwdParams p) {
         const bool any_growth = !__all(smax <= m_run);
         // full tiles (no softclamp) kept RAW scores: exp2(fma(s, scale2, -m))
         const float escale = (full_tile && !SOFTCLAMP) ? scale2 : 1.f;
-        float rowsum = 0.f;
         uint32_t pk[NBLK * 8];                                      // packed bf16 pairs
+        float partial[NBLK * 8];
         #pragma unroll
         for (int x2 = 0; x2 < NBLK * 8; ++x2) {
             float e0 = __builtin_amdgcn_exp2f(
                 __builtin_fmaf(s[x2 >> 3][(2 * x2) & 15], escale, -m_new));
             float e1 = __builtin_amdgcn_exp2f(
                 __builtin_fmaf(s[x2 >> 3][(2 * x2 + 1) & 15], escale, -m_new));
-            rowsum += e0 + e1;
+            partial[x2] = e0 + e1;
             union { __hip_bfloat162 h2; uint32_t u; } cvt;
             cvt.h2 = __float22bfloat162_rn(float2{e0, e1});
             pk[x2] = cvt.u;
         }
+        // pairwise tree instead of a 32-deep serial dependency chain
+        #pragma unroll
+        for (int w = NBLK * 4; w >= 1; w >>= 1)
+            #pragma unroll
+            for (int x2 = 0; x2 < w; ++x2) partial[x2] += partial[x2 + w];
+        float rowsum = partial[0];
         rowsum += __shfl_xor(rowsum, 32);
         if (any_growth) {
             float alpha = __builtin_amdgcn_exp2f(m_run - m_new);
@@ -415,6 +427,7 @@ __global__ __launch_bounds__(NTHREADS, 1) void attn_fwd_kernel(FwdParams p) {
             }
         }
 
+        if (stamp) p.ticks[t * 6 + 4] = __builtin_amdgcn_s_memtime();
         // ---- PV: O^T[d][q] += V^T[d][kv] P^T[kv][q]
         __builtin_amdgcn_s_setprio(1);
         #pragma unroll
@@ -429,6 +442,7 @@ __global__ __launch_bounds__(NTHREADS, 1) void attn_fwd_kernel(FwdParams p) {
             }
         }
         __builtin_amdgcn_s_setprio(0);
+        if (stamp) p.ticks[t * 6 + 5] = __builtin_amdgcn_s_memtime();
     }
 
     // ---- epilogue

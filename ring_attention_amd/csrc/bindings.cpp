@@ -28,7 +28,8 @@ void attn_fwd(
     double scale, bool causal, int64_t diag, int64_t q_stride,
     int64_t win, bool has_win,
     bool softclamp, double softclamp_value,
-    bool is_first, bool is_last, int64_t kv_split, int64_t ablate) {
+    bool is_first, bool is_last, int64_t kv_split, int64_t ablate,
+    std::optional<at::Tensor> ticks) {
     CHECK_BF16_CONTIG(q); CHECK_BF16_CONTIG(k); CHECK_BF16_CONTIG(v);
     TORCH_CHECK(q.dim() == 4 && k.dim() == 4 && v.dim() == 4, "q/k/v must be (B,N,H,D)");
     const int64_t B = q.size(0), Nq = q.size(1), H = q.size(2), D = q.size(3);
@@ -78,6 +79,8 @@ void attn_fwd(
     p.is_first = is_first; p.is_last = is_last;
     p.kv_split = (int)kv_split;
     p.ablate = (int)ablate;
+    p.ticks = nullptr;
+    if (ticks.has_value()) p.ticks = (unsigned long long*)ticks->data_ptr();
 
     launch_attn_fwd(p, (int)D, at::hip::getCurrentHIPStream());
     TORCH_CHECK(hipGetLastError() == hipSuccess, "attn_fwd launch failed");

@@ -159,13 +159,13 @@ class RingFlashAttentionHIPFunction(Function):
                 l_part = torch.empty(kv_split, b, h, n, device=q.device, dtype=torch.float32)
                 ext.attn_fwd(qb, k_full, v_full, m_full, o_part, m_part, l_part,
                              None, None, scale, causal, diag, q_stride, 0, False,
-                             softclamp_qk_sim, softclamp_value, True, True, kv_split, 0)
+                             softclamp_qk_sim, softclamp_value, True, True, kv_split, 0, None)
                 ext.attn_fwd_merge(o_part, m_part, l_part, None, None, None,
                                    out, lse, kv_split, b, h, d, n, True, True)
             else:
                 ext.attn_fwd(qb, k_full, v_full, m_full, None, None, None, out, lse,
                              scale, causal, diag, q_stride, 0, False,
-                             softclamp_qk_sim, softclamp_value, True, True, 1, 0)
+                             softclamp_qk_sim, softclamp_value, True, True, 1, 0, None)
             ctx.save_for_backward(qb, kb, vb, out, lse,
                                   mask_u8 if mask_u8 is not None else torch.empty(0))
             ctx.params = (causal, striped, lookback, hops, softclamp_qk_sim,
@@ -219,7 +219,7 @@ class RingFlashAttentionHIPFunction(Function):
                              o_part, m_part, l_part, None, None,
                              scale, causal, diag, 1, win, lookback is not None,
                              softclamp_qk_sim, softclamp_value,
-                             is_f, is_l, kv_split, 0)
+                             is_f, is_l, kv_split, 0, None)
                 ext.attn_fwd_merge(o_part, m_part, l_part, o_acc, m, l,
                                    out if is_l else None, lse if is_l else None,
                                    kv_split, b, h, d, n, is_f, is_l)
@@ -228,7 +228,7 @@ class RingFlashAttentionHIPFunction(Function):
                              o_acc, m, l, out, lse,
                              scale, causal, diag, 1, win, lookback is not None,
                              softclamp_qk_sim, softclamp_value,
-                             is_f, is_l, 1, 0)
+                             is_f, is_l, 1, 0, None)
 
         ctx.save_for_backward(qb, kb, vb, out, lse,
                               mask_u8 if mask_u8 is not None else torch.empty(0))
@@ -379,7 +379,7 @@ class FlashAttnOffsetFunction(Function):
         lse = torch.empty(b, h, n, device=q.device, dtype=torch.float32)
         ext.attn_fwd(qb, kb, vb, None, None, None, None, out, lse,
                      scale, causal, q_offset, 1, 0, False, False, 50.0,
-                     True, True, 1, 0)
+                     True, True, 1, 0, None)
         ctx.save_for_backward(qb, kb, vb, out, lse)
         ctx.meta = (q_offset, causal, q.dtype)
         return out.to(q.dtype)

@@ -155,9 +155,9 @@ def test_resume_contract_two_passes():
     # q positions are 0..n-1, shard0 cols 0..half-1 (diag = 0), shard1 cols
     # half.. (j_local <= i - half  => diag = -half)
     ext.attn_fwd(q, k0, v0, None, o_acc, m, l, out, lse,
-                 scale, True, 0, 1, 0, False, False, 50.0, True, False, 1, 0)
+                 scale, True, 0, 1, 0, False, False, 50.0, True, False, 1, 0, None)
     ext.attn_fwd(q, k1, v1, None, o_acc, m, l, out, lse,
-                 scale, True, -half, 1, 0, False, False, 50.0, False, True, 1, 0)
+                 scale, True, -half, 1, 0, False, False, 50.0, False, True, 1, 0, None)
 
     _, _, _, ref, ref_lse = _oracle(q, k, v, causal=True)
     err = (out.float().cpu() - ref).abs().max().item()
@@ -254,7 +254,7 @@ def test_strided_q_positions_kernel():
     out = torch.empty_like(q)
     lse = torch.empty(b, h, n, device="cuda", dtype=torch.float32)
     ext.attn_fwd(q, k, v, None, None, None, None, out, lse,
-                 scale, True, rq, R, 0, False, False, 50.0, True, True, 1, 0)
+                 scale, True, rq, R, 0, False, False, 50.0, True, True, 1, 0, None)
 
     qp = torch.arange(n) * R + rq
     ref = default_attention(q.float().cpu(), k.float().cpu(), v.float().cpu(),
@@ -336,7 +336,7 @@ def test_kv_split_merge_resume_across_hops():
         first, last = hop == 0, hop == 1
         ext.attn_fwd(q, ks.contiguous(), vs.contiguous(), None, o_p, m_p, l_p,
                      None, None, scale, True, diag, 1, 0, False, False, 50.0,
-                     first, last, S, 0)
+                     first, last, S, 0, None)
         ext.attn_fwd_merge(o_p, m_p, l_p, o_acc, m, l,
                            out if last else None, lse if last else None,
                            S, b, h, d, n, first, last)

@@ -14,11 +14,11 @@ scale = d ** -0.5
 def run(ablate, iters=30):
     for _ in range(5):
         ext.attn_fwd(q, k, v, None, None, None, None, out, lse, scale,
-                     False, 0, 1, 0, False, False, 50.0, True, True, 1, ablate)
+                     False, 0, 1, 0, False, False, 50.0, True, True, 1, ablate, None)
     torch.cuda.synchronize(); t0 = time.perf_counter()
     for _ in range(iters):
         ext.attn_fwd(q, k, v, None, None, None, None, out, lse, scale,
-                     False, 0, 1, 0, False, False, 50.0, True, True, 1, ablate)
+                     False, 0, 1, 0, False, False, 50.0, True, True, 1, ablate, None)
     torch.cuda.synchronize()
     return (time.perf_counter() - t0) / iters * 1e6
 print(f"full kernel:  {run(0):8.1f} us")
