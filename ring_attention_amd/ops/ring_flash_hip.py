@@ -43,20 +43,18 @@ def _choose_strategy(use_ring: bool, topo, n, hk, d, lookback, dtype_bytes=2) ->
     if not use_ring or topo.ring_size == 1:
         return "local"
     forced = os.environ.get("RING_ATTN_FORCE_STRATEGY")
-    if forced in ("ring", "allgather") and topo.ring_size == topo.world_size:
+    if forced in ("ring", "allgather"):
         return forced
-    if topo.ring_size != topo.world_size:
-        return "ring"        # sub-rings: per-ring communicators (future work)
     if lookback is not None:
         return "ring"        # window truncates the ring walk — cheaper there
     kv_bytes = 2 * topo.ring_size * n * hk * d * dtype_bytes
     return "allgather" if kv_bytes <= _AG_BUDGET else "ring"
 
 
-def _gather_global_order(t, ring_size: int, striped: bool, dim: int = 1):
+def _gather_global_order(t, ring_size: int, striped: bool, dim: int = 1, group=None):
     """All-gather seq shards and reorder so index == GLOBAL position
     (striped shards interleave: global g = local * R + rank)."""
-    full = gather_cat(t, dim=dim)            # rank-major along dim
+    full = gather_cat(t, dim=dim, group=group)  # rank-major along dim
     if not striped or ring_size == 1:
         return full
     # rank-major (s, local) -> position-major (local, s)
@@ -144,9 +142,10 @@ class RingFlashAttentionHIPFunction(Function):
         strategy = _choose_strategy(use_ring, topo, n, hk, d, lookback)
         if strategy == "allgather":
             R, rq = topo.ring_size, topo.ring_rank
-            k_full = _gather_global_order(kb, R, striped)
-            v_full = _gather_global_order(vb, R, striped)
-            m_full = _gather_global_order(mask_u8, R, striped) if mask_u8 is not None else None
+            pg = topo.process_group()        # sub-ring communicator (or world)
+            k_full = _gather_global_order(kb, R, striped, group=pg)
+            v_full = _gather_global_order(vb, R, striped, group=pg)
+            m_full = _gather_global_order(mask_u8, R, striped, group=pg) if mask_u8 is not None else None
             q_stride = R if striped else 1
             diag = rq if striped else rq * n
             n_total = n * R
@@ -260,9 +259,10 @@ class RingFlashAttentionHIPFunction(Function):
 
         if strategy == "allgather":
             R = topo.ring_size
-            k_full = _gather_global_order(kb, R, striped)
-            v_full = _gather_global_order(vb, R, striped)
-            m_full = _gather_global_order(mask_u8, R, striped) if mask_u8 is not None else None
+            pg = topo.process_group()
+            k_full = _gather_global_order(kb, R, striped, group=pg)
+            v_full = _gather_global_order(vb, R, striped, group=pg)
+            m_full = _gather_global_order(mask_u8, R, striped, group=pg) if mask_u8 is not None else None
             n_total = n * R
             q_stride = R if striped else 1
             diag = rq if striped else rq * n
@@ -284,7 +284,7 @@ class RingFlashAttentionHIPFunction(Function):
             dk_chunks = _scatter_chunks_of_global(dk_full, R, striped, dim=2)
             dv_chunks = _scatter_chunks_of_global(dv_full, R, striped, dim=3)
             packed = torch.cat((dk_chunks.reshape(R, -1), dv_chunks.reshape(R, -1)), dim=1)
-            own = reduce_scatter_chunks(packed)
+            own = reduce_scatter_chunks(packed, group=pg)
             half = own.numel() // 2
             dk_home = own[:half].view(b, hk, n, d).permute(0, 2, 1, 3).contiguous()
             dv_home = own[half:].view(b, hk, d, n).permute(0, 3, 1, 2).contiguous()

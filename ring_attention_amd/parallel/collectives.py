@@ -123,37 +123,40 @@ def split_by_rank(xs: list[Tensor]) -> Tensor:
     return xs[get_rank()]
 
 
-def gather_cat(t: Tensor, dim: int) -> Tensor:
-    """Plain (non-autograd) all-gather concatenated along ``dim`` in rank order."""
+def gather_cat(t: Tensor, dim: int, group=None) -> Tensor:
+    """Plain (non-autograd) all-gather concatenated along ``dim`` in rank order.
+
+    ``group`` restricts the gather to a sub-ring's process group."""
     if not is_distributed():
         return t
-    world = get_world_size()
+    world = dist.get_world_size(group) if group is not None else get_world_size()
     t = t.contiguous()
     if dist.get_backend() == "nccl":
         out = torch.empty((world,) + tuple(t.shape), device=t.device, dtype=t.dtype)
-        dist.all_gather_into_tensor(out.view(world, -1), t.view(-1))
+        dist.all_gather_into_tensor(out.view(world, -1), t.view(-1), group=group)
         chunks = list(out.unbind(0))
     else:
         chunks = [torch.empty_like(t) for _ in range(world)]
-        dist.all_gather(chunks, t)
+        dist.all_gather(chunks, t, group=group)
     return torch.cat(chunks, dim=dim)
 
 
-def reduce_scatter_chunks(chunks: Tensor) -> Tensor:
+def reduce_scatter_chunks(chunks: Tensor, group=None) -> Tensor:
     """``chunks`` (W, ...) — sum chunk r across ranks, return this rank's chunk.
 
     RCCL reduce-scatter stripes across every xGMI link; the gloo fallback is
-    all-reduce + slice (same result).
-    """
+    all-reduce + slice.  ``group`` restricts to a sub-ring."""
     if not is_distributed():
         assert chunks.shape[0] == 1
         return chunks[0]
-    world = get_world_size()
+    world = dist.get_world_size(group) if group is not None else get_world_size()
+    my = dist.get_rank(group) if group is not None else get_rank()
     assert chunks.shape[0] == world
     chunks = chunks.contiguous()
     if dist.get_backend() == "nccl":
         out = torch.empty_like(chunks[0])
-        dist.reduce_scatter_tensor(out.view(-1), chunks.view(world, -1).reshape(-1))
+        dist.reduce_scatter_tensor(out.view(-1), chunks.view(world, -1).reshape(-1),
+                                   group=group)
         return out
-    dist.all_reduce(chunks)
-    return chunks[get_rank()].clone()
+    dist.all_reduce(chunks, group=group)
+    return chunks[my].clone()

@@ -14,6 +14,9 @@ import functools
 
 import torch.distributed as dist
 
+# cache: (world_size, ring_size) -> list of per-ring process groups
+_RING_GROUPS: dict[tuple[int, int], list] = {}
+
 
 def is_distributed() -> bool:
     return dist.is_initialized() and dist.get_world_size() > 1
@@ -73,6 +76,23 @@ class RingTopology:
     def left(self) -> int:
         """Global rank of the previous rank around the ring (sends what we receive)."""
         return self.global_rank_of(self.ring_rank - 1)
+
+    def process_group(self):
+        """The process group of THIS rank's ring (None = default/world group).
+
+        Sub-ring groups are created lazily but COLLECTIVELY (every rank
+        creates every ring's group in the same order, as torch requires) and
+        cached for the process lifetime."""
+        if self.ring_size == self.world_size or not dist.is_initialized():
+            return None
+        key = (self.world_size, self.ring_size)
+        if key not in _RING_GROUPS:
+            groups = []
+            for ring in range(self.world_size // self.ring_size):
+                ranks = list(range(ring * self.ring_size, (ring + 1) * self.ring_size))
+                groups.append(dist.new_group(ranks))
+            _RING_GROUPS[key] = groups
+        return _RING_GROUPS[key][self.ring_index]
 
     def source_of_hop(self, hop: int) -> int:
         """Ring rank whose original shard this rank holds after ``hop`` passes.

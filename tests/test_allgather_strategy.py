@@ -110,3 +110,34 @@ def test_gather_scatter_helpers_contiguous():
 
 def test_gather_scatter_helpers_striped():
     run_distributed(4, _helper_case, True)
+
+
+def _subring_helper_case(rank, world, ring_size):
+    """gather/reduce-scatter within sub-ring process groups (gloo)."""
+    import torch
+    from ring_attention_amd.parallel import RingTopology
+    from ring_attention_amd.parallel.collectives import gather_cat, reduce_scatter_chunks
+    topo = RingTopology(ring_size)
+    pg = topo.process_group()
+    assert (pg is None) == (ring_size == world)
+
+    # each rank contributes a distinct shard; gather within the ring only
+    t = torch.full((2, 3), float(rank))
+    full = gather_cat(t, dim=0, group=pg)
+    expect = torch.cat([torch.full((2, 3), float(topo.ring_base + s))
+                        for s in range(ring_size)], dim=0)
+    assert torch.equal(full, expect), f"rank {rank}: {full} vs {expect}"
+
+    chunks = torch.ones(ring_size, 4) * (rank + 1)
+    own = reduce_scatter_chunks(chunks, group=pg)
+    ring_sum = sum(r + 1 for r in range(topo.ring_base, topo.ring_base + ring_size))
+    assert torch.equal(own, torch.full((4,), float(ring_sum)))
+    return True
+
+
+def test_subring_collectives_w4r2():
+    run_distributed(4, _subring_helper_case, 2)
+
+
+def test_subring_collectives_w4r4():
+    run_distributed(4, _subring_helper_case, 4)
