@@ -103,11 +103,12 @@ struct DecodeParams {
     const void* q;          // bf16 (B, H, 1, D)
     const void* k;          // bf16 (B, H, N, D)
     const void* v;          // bf16 (B, H, N, D)
-    float* out;             // fp32 (B, H, 1, D)
-    float* lse;             // fp32 (B, H, 1, 1)
+    float* out;             // fp32 (S, B, H, 1, D) per-chunk partials
+    float* lse;             // fp32 (S, B, H, 1, 1)
     int b, h;
     long n;
     float scale;
+    long chunks;            // kv-split S (0 = auto)
 };
 
 void launch_decode_partial(const DecodeParams& p, int head_dim, hipStream_t stream);

@@ -183,17 +183,23 @@ at::Tensor rotary_apply(at::Tensor x, at::Tensor cos_t, at::Tensor sin_t, double
 }
 
 std::vector<at::Tensor> decode_partial(at::Tensor q, at::Tensor k, at::Tensor v) {
-    // q (B,H,1,D); k,v (B,H,N,D) bf16 -> (out fp32 (B,H,1,D), lse fp32 (B,H,1,1))
+    // q (B,H,1,D); k,v (B,H,N,D) bf16
+    // -> (out fp32 (S,B,H,1,D), lse fp32 (S,B,H,1,1)): S kv-chunk partials,
+    //    merged by the caller with the standard logsumexp combine
     CHECK_BF16_CONTIG(q); CHECK_BF16_CONTIG(k); CHECK_BF16_CONTIG(v);
     const int64_t B = q.size(0), H = q.size(1), D = q.size(3), N = k.size(2);
     TORCH_CHECK(D == 64 || D == 128, "head dim must be 64 or 128");
-    auto out = at::empty({B, H, 1, D}, q.options().dtype(at::kFloat));
-    auto lse = at::empty({B, H, 1, 1}, q.options().dtype(at::kFloat));
+    int64_t waves = B * H;
+    int64_t chunks = std::max<int64_t>(
+        1, std::min<int64_t>(N / 1024 + 1, 1024 / std::max<int64_t>(waves / 4, 1)));
+    auto out = at::empty({chunks, B, H, 1, D}, q.options().dtype(at::kFloat));
+    auto lse = at::empty({chunks, B, H, 1, 1}, q.options().dtype(at::kFloat));
     DecodeParams p{};
     p.q = q.data_ptr(); p.k = k.data_ptr(); p.v = v.data_ptr();
     p.out = out.data_ptr<float>(); p.lse = lse.data_ptr<float>();
     p.b = (int)B; p.h = (int)H; p.n = N;
     p.scale = (float)(1.0 / std::sqrt((double)D));
+    p.chunks = chunks;
     launch_decode_partial(p, (int)D, at::hip::getCurrentHIPStream());
     TORCH_CHECK(hipGetLastError() == hipSuccess, "decode launch failed");
     return {out, lse};

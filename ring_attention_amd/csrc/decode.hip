@@ -15,12 +15,19 @@ namespace ring_attn {
 
 template <int D>
 __global__ __launch_bounds__(256) void decode_partial_kernel(DecodeParams p) {
-    // one wave per (b, h); 4 waves per block
+    // one wave per (b, h); 4 waves per block; blockIdx.y splits the KV range
+    // into `chunks` partials (merged on the host) so long sequences use the
+    // whole chip instead of b*h waves
     const int wave_global = (blockIdx.x * 4) + (threadIdx.x >> 6);
     if (wave_global >= p.b * p.h) return;
     const int b = wave_global / p.h;
     const int h = wave_global % p.h;
     const int lane = threadIdx.x & 63;
+    const int chunk = blockIdx.y;
+    const long per = (p.n + gridDim.y - 1) / gridDim.y;
+    const long j_lo = chunk * per;
+    const long j_hi = min(p.n, j_lo + per);
+    const long part_off = (long)chunk * p.b * p.h;
 
     const __bf16* qp = (const __bf16*)p.q + ((long)b * p.h + h) * D;
     const __bf16* kp = (const __bf16*)p.k + ((long)b * p.h + h) * p.n * D;
@@ -37,7 +44,7 @@ __global__ __launch_bounds__(256) void decode_partial_kernel(DecodeParams p) {
     #pragma unroll
     for (int d = 0; d < D; ++d) acc[d] = 0.f;
 
-    for (long j = lane; j < p.n; j += 64) {
+    for (long j = j_lo + lane; j < j_hi; j += 64) {
         const __bf16* krow = kp + j * D;
         float s = 0.f;
         #pragma unroll
@@ -80,16 +87,21 @@ __global__ __launch_bounds__(256) void decode_partial_kernel(DecodeParams p) {
     float l_safe = fmaxf(l, 1e-38f);
     if (lane == 0) {
         float inv = 1.f / l_safe;
-        float* op = p.out + ((long)b * p.h + h) * D;
+        float* op = p.out + (part_off + (long)b * p.h + h) * D;
         #pragma unroll
         for (int d = 0; d < D; ++d) op[d] = acc[d] * inv;
-        p.lse[(long)b * p.h + h] = __logf(l_safe) + m;
+        p.lse[part_off + (long)b * p.h + h] = __logf(l_safe) + m;
     }
 }
 
 void launch_decode_partial(const DecodeParams& p, int head_dim, hipStream_t stream) {
     int waves = p.b * p.h;
-    dim3 grid((waves + 3) / 4);
+    // enough chunks to put ~1 wave per CU at this (b*h): each wave streams
+    // its KV slice; host merges the per-chunk partials
+    long target = 1024;
+    long chunks = p.chunks > 0 ? p.chunks
+                 : max(1L, min((long)(p.n / 1024 + 1), target / max(waves / 4, 1)));
+    dim3 grid((waves + 3) / 4, (unsigned)chunks);
     dim3 block(256);
     if (head_dim == 64) {
         hipLaunchKernelGGL(decode_partial_kernel<64>, grid, block, 0, stream, p);

@@ -25,7 +25,18 @@ def _local_decode_partial(q: Tensor, k: Tensor, v: Tensor) -> tuple[Tensor, Tens
     if q.is_cuda:
         from .ops import hip_ext
         if hip_ext.available():
-            return hip_ext.decode_partial(q, k, v)
+            outs, lses = hip_ext.decode_partial(
+                q.to(torch.bfloat16).contiguous(),
+                k.to(torch.bfloat16).contiguous(),
+                v.to(torch.bfloat16).contiguous())
+            # merge the S kv-chunk partials (same math as the cross-rank merge)
+            m = lses.max(dim=0).values                       # (b,h,1,1)
+            w = (lses - m[None]).exp()                       # (S,b,h,1,1)
+            den = w.sum(dim=0)
+            num = (outs * w).sum(dim=0)
+            out = num / den.clamp(min=1e-38)
+            lse = den.log() + m
+            return out, lse
     scale = q.shape[-1] ** -0.5
     sim = torch.einsum("bhid,bhjd->bhij", q.float(), k.float()) * scale
     lse = sim.logsumexp(dim=-1, keepdim=True)

@@ -166,14 +166,13 @@ def test_resume_contract_two_passes():
 
 
 def test_decode_partial():
-    from ring_attention_amd.ops import hip_ext
-    ext = hip_ext.require()
+    from ring_attention_amd.tree_decode import _local_decode_partial
     b, h, n, d = 2, 4, 1000, 64
     torch.manual_seed(7)
     q = torch.randn(b, h, 1, d, device="cuda", dtype=torch.bfloat16)
     k = torch.randn(b, h, n, d, device="cuda", dtype=torch.bfloat16)
     v = torch.randn(b, h, n, d, device="cuda", dtype=torch.bfloat16)
-    out, lse = ext.decode_partial(q, k, v)
+    out, lse = _local_decode_partial(q, k, v)
     sim = torch.einsum("bhid,bhjd->bhij", q.float(), k.float()) * d ** -0.5
     ref = torch.einsum("bhij,bhjd->bhid", sim.softmax(-1), v.float())
     ref_lse = sim.logsumexp(-1, keepdim=True)
