@@ -6,17 +6,19 @@
 // re-thought for wave64 + MFMA; no code ported):
 //
 //   * 8 waves / 512 threads per workgroup; each wave owns 32 Q rows
-//     (256-row Q tile per workgroup), KV tile = 64.
+//     (256-row Q tile per workgroup), KV tile = KVBLK (128 at d64).
 //   * swapped QK^T: S^T[kv][q] = mfma(A=K, B=Q^T) so each lane holds a full
 //     slice of ONE q row's scores -> softmax is almost entirely in-register
 //     (31 VALU max/sum + one cross-half __shfl_xor), no LDS round trip.
 //   * P -> bf16 via v_cvt_pk_bf16_f32 pairs + v_permlane32_swap to build the
 //     PV B-operand fragments in-register (T12 pattern).
 //   * K and V^T staged in LDS with an XOR-16B swizzle (conflict-free
-//     ds_read_b128 column slices).
-//   * causality, striping and lookback all reduce to two integers:
-//     attend(i,j) <=> j <= i + diag  AND  i - j <= win
-//     (host folds ring-rank offsets / stride-R striping into diag/win).
+//     ds_read_b128 column slices); 3-deep pipeline over DOUBLE-buffered LDS
+//     (one barrier per tile; staging overlaps the MFMAs).
+//   * causality, striping and lookback reduce to integers: with
+//     qpos(i) = i*q_stride + diag, attend(i,j) <=> j <= qpos(i) AND
+//     qpos(i) - j <= win (host folds ring-rank offsets / layout strides;
+//     q_stride > 1 expresses a striped q shard against gathered global KV).
 //   * resume contract: fp32 o_acc (B,H,D,Nq transposed scratch), m, l
 //     (B,H,Nq) persist between ring hops; IS_FIRST initializes instead of
 //     loading, IS_LAST normalizes and emits bf16 out (B,Nq,H,D) + lse.

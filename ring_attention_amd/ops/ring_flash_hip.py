@@ -1,21 +1,27 @@
 """Ring flash attention backed by the CDNA4 HIP kernels (the GPU compute path).
 
-Same semantics as ops/ring_flash.py (the oracle), same ring transport
-(double-buffered overlapped P2P), but each hop's compute is ONE resumable
-HIP kernel launch:
+Same semantics as ops/ring_flash.py (the oracle), with an execution-strategy
+choice per call (see _choose_strategy):
 
-- forward: attn_fwd with (o_acc fp32, m, l) persisting across hops; causality,
-  striping and lookback folded into two integers (diag, win) per hop (host
-  computes them from ring ranks — see csrc/attn_fwd.hip header).  Hops whose
-  kv shard is entirely masked are skipped host-side (kv still circulates).
-- backward: delta precomputed once; attn_bwd per hop; dq accumulates in ONE
-  fp32 buffer via kernel atomics across hops (no per-hop add); dk/dv per hop
-  go through the pipelined RingAccumulator, in the kernel's transposed
-  scratch layouts (dk (B,HK,Nk,D), dv (B,HK,D,Nk)); final transpose once.
+ALL-GATHER strategy (default when gathered K/V fits memory): one RCCL
+all-gather per tensor (stripes across all 7 xGMI links of the full mesh),
+ONE local kernel launch over the whole KV (strided-q positions express the
+striped layout), and backward ends in a single reduce-scatter for dk/dv.
+
+RING strategy (windows / sub-rings / beyond-memory): double-buffered
+overlapped P2P shard circulation; each hop is one resumable kernel launch:
+- forward: attn_fwd with (o_acc fp32, m, l) persisting across hops;
+  causality, striping and lookback fold into (diag, q_stride, win) per hop;
+  fully-masked hops are skipped host-side (kv still circulates).
+- backward: delta precomputed once; per hop a row-parallel dq kernel (plain
+  fp32 accumulation into one buffer across hops) and a column-parallel dk/dv
+  kernel whose per-hop contribution goes through the pipelined
+  RingAccumulator in the kernels' transposed scratch layouts
+  (dk (B,HK,Nk,D), dv (B,HK,D,Nk)); one final permute at home.
 
 Capability parity with the reference's ring_flash_attn_cuda
 (/root/reference/ring_attention_pytorch/ring_flash_attention_cuda.py:40-371)
-with its §2.5 bugs fixed and full comm/compute overlap.
+with its §2.5 bugs fixed.
 """
 
 from __future__ import annotations
