@@ -167,3 +167,53 @@ def _var_batch_case(rank, world):
 
 def test_transformer_var_batch_world2():
     run_distributed(2, _var_batch_case)
+
+
+def _presharded_case(rank, world, striped):
+    """Statically pre-sharded data (utils/data.py) == the model's auto-shard path."""
+    from ring_attention_amd.utils.data import shard_sequence_batch
+    torch.manual_seed(17)
+    kwargs = dict(num_tokens=64, dim=32, depth=1, causal=True, dim_head=16,
+                  heads=2, bucket_size=8, ring_seq_size=16,
+                  striped_ring_attn=striped, use_hip_kernel=False)
+    auto_model = RingTransformer(ring_attn=True, auto_shard_seq=True, **kwargs)
+    pre_model = RingTransformer(ring_attn=True, auto_shard_seq=False, **kwargs)
+    pre_model.load_state_dict(auto_model.state_dict())
+
+    torch.manual_seed(400)
+    ids = torch.randint(0, 64, (2, 29))           # padded to 32 -> 2 shards of 16
+
+    sb = shard_sequence_batch(ids, ring_seq_size=16, bucket_size=8,
+                              world=world, rank=rank, striped=striped)
+    loss_pre = pre_model(sb.ids, mask=sb.mask, labels=sb.labels, return_loss=True)
+    assert torch.isfinite(loss_pre)
+
+    # ground truth: replicated non-ring model over the full (unsharded) batch;
+    # its shard-restricted CE must equal the pre-sharded ring loss
+    flat_model = RingTransformer(ring_attn=False, **kwargs)
+    flat_model.load_state_dict(auto_model.state_dict())
+    x_full, lab_full = ids[:, :-1], ids[:, 1:]
+    logits_full = flat_model(x_full)              # (b, 28, vocab)
+    import torch.nn.functional as F
+    from ring_attention_amd.utils.sharding import stripe_permute as sp
+    n = x_full.shape[1]
+    pad = 32 - n
+    logits_pad = torch.nn.functional.pad(logits_full, (0, 0, 0, pad))
+    labels_pad = torch.nn.functional.pad(lab_full, (0, pad), value=-1)
+    if striped:
+        logits_pad = sp(logits_pad, world)
+        labels_pad = sp(labels_pad, world)
+    sl = slice(rank * 16, (rank + 1) * 16)
+    ref_loss = F.cross_entropy(logits_pad[:, sl].permute(0, 2, 1),
+                               labels_pad[:, sl], ignore_index=-1)
+    assert abs(loss_pre.item() - ref_loss.item()) < 1e-4, (
+        f"pre-sharded loss {loss_pre.item()} vs ref {ref_loss.item()}")
+    return float(loss_pre.detach())
+
+
+def test_presharded_data_world2():
+    run_distributed(2, _presharded_case, False)
+
+
+def test_presharded_data_world2_striped():
+    run_distributed(2, _presharded_case, True)
