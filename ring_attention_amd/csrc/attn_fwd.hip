@@ -44,8 +44,7 @@ namespace ring_attn {
 static constexpr int WAVES = 8;          // 8-wave WGs: the 2-waves/SIMD paired
 static constexpr int QROWS_WAVE = 32;    // regime (4-wave variant measured slower)
 static constexpr int QROWS_WG = WAVES * QROWS_WAVE;
-static constexpr int KVBLK = 128;        // kv tile; NBLK 32-row MFMA blocks
-static constexpr int NBLK = KVBLK / 32;
+template <int D> constexpr int fwd_kvblk() { return D == 64 ? 128 : 64; }  // LDS budget
 static constexpr int NTHREADS = WAVES * 64;
 
 // XOR swizzle of a 16-byte chunk index within a row (row stride D*2 bytes):
@@ -62,10 +61,11 @@ __device__ __forceinline__ float fast_tanhf(float x) {
 
 template <int D>
 struct FwdLds {
+    static constexpr int KVB = fwd_kvblk<D>();
     // double-buffered: K tile [kv][D] + V^T tile [d][kv], 16B-chunk swizzled
-    __align__(16) __bf16 k[2][KVBLK * D];
-    __align__(16) __bf16 vt[2][D * KVBLK];
-    unsigned char kmask[2][KVBLK];
+    __align__(16) __bf16 k[2][KVB * D];
+    __align__(16) __bf16 vt[2][D * KVB];
+    unsigned char kmask[2][KVB];
 };
 
 static constexpr float LOG2E = 1.4426950408889634f;
@@ -79,6 +79,8 @@ __global__ __launch_bounds__(NTHREADS, 1) void attn_fwd_kernel(FwdParams p) {
     static_assert(D % 32 == 0);
     constexpr int DBLK = D / 32;     // 32-d output blocks
     constexpr int KSTEPS = D / 16;   // QK^T k-steps
+    constexpr int KVBLK = fwd_kvblk<D>();
+    constexpr int NBLK = KVBLK / 32;
 
     __shared__ FwdLds<D> lds;
 
