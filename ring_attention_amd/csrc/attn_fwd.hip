@@ -59,6 +59,19 @@ __device__ __forceinline__ float fast_tanhf(float x) {
     return 1.f - 2.f * __builtin_amdgcn_rcpf(e + 1.f);
 }
 
+// cross-half (lane <-> lane^32) exchange via v_permlane32_swap — pure VALU,
+// no LDS round trip (a __shfl_xor(x, 32) lowers to ds_bpermute + addressing)
+__device__ __forceinline__ float cross_half(float x) {
+    union { float f; unsigned u; } c; c.f = x;
+    u32x2 r = __builtin_amdgcn_permlane32_swap(c.u, c.u, false, false);
+    // r0 = {lo: own lo, hi: own lo}; r1 = {lo: own hi, hi: own hi}
+    union { unsigned u; float f; } lo, hi; lo.u = r[0]; hi.u = r[1];
+    // lanes < 32 want the partner's value = own hi-half's value = r1.lo;
+    // lanes >= 32 want r0.hi = own lo-half's value — both are "the other
+    // half's x" exactly when we select r1 on lo lanes and r0 on hi lanes
+    return (threadIdx.x & 32) ? lo.f : hi.f;
+}
+
 template <int D>
 struct FwdLds {
     static constexpr int KVB = fwd_kvblk<D>();
@@ -369,7 +382,7 @@ __global__ __launch_bounds__(NTHREADS, 1) void attn_fwd_kernel(FwdParams p) {
                     smax = fmaxf(smax, x);
                 }
         }
-        smax = fmaxf(smax, __shfl_xor(smax, 32));
+        smax = fmaxf(smax, cross_half(smax));
         if (stamp) p.ticks[t * 6 + 3] = __builtin_amdgcn_s_memtime();
 
         // ---- online softmax update (defer-max THR=0: exact — skip the O
@@ -397,7 +410,7 @@ __global__ __launch_bounds__(NTHREADS, 1) void attn_fwd_kernel(FwdParams p) {
             #pragma unroll
             for (int x2 = 0; x2 < w; ++x2) partial[x2] += partial[x2 + w];
         float rowsum = partial[0];
-        rowsum += __shfl_xor(rowsum, 32);
+        rowsum += cross_half(rowsum);
         if (any_growth) {
             float alpha = __builtin_amdgcn_exp2f(m_run - m_new);
             l_run = l_run * alpha + rowsum;
