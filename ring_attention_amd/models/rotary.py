@@ -16,7 +16,7 @@ from __future__ import annotations
 import torch
 from torch import Tensor, nn
 
-from ..parallel import RingTopology, get_world_size, is_distributed
+from ..parallel import RingTopology, is_distributed
 
 
 class RingRotaryEmbedding(nn.Module):

@@ -8,7 +8,6 @@ path, never a silent eager fallback.
 
 from __future__ import annotations
 
-import os
 
 _ext = None
 _tried = False

@@ -10,10 +10,8 @@ Layout: q (b, n, h, d); k, v (b, n_kv, h_kv, d) with h % h_kv == 0.
 
 from __future__ import annotations
 
-import math
 
 import torch
-import torch.nn.functional as F
 from torch import Tensor
 
 MASK_VALUE = torch.finfo(torch.float32).min

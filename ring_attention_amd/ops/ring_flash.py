@@ -25,10 +25,8 @@ attend iff ``j < i`` or (``j == i`` and ``rk <= rq``).
 from __future__ import annotations
 
 import math
-from typing import Literal
 
 import torch
-import torch.nn.functional as F
 from torch import Tensor
 from torch.autograd import Function
 
