@@ -365,3 +365,26 @@ def test_zigzag_fast_path_gqa():
     ref = torch.einsum("bhij,bhjd->bhid", sim.softmax(-1), vv)
     err = (out.float() - ref).abs().max().item()
     assert err < 2e-2, f"zigzag gqa err {err}"
+
+
+def test_d128_gqa_mask_combo():
+    """d128 (QT=32 dkv, KVB=64 paths) x GQA x key-pad mask x causal, fwd+bwd."""
+    from ring_attention_amd.ops.ring_flash_hip import ring_flash_attn_hip_
+    b, n, h, hk, d = 1, 384, 4, 2, 128
+    q, k, v = _mk(b, n, h, hk, d, seed=15)
+    torch.manual_seed(16)
+    mask = torch.rand(b, n, device="cuda") > 0.25
+    mask[:, :8] = True
+    qg = q.clone().requires_grad_(True)
+    kg = k.clone().requires_grad_(True)
+    vg = v.clone().requires_grad_(True)
+    out, _ = ring_flash_attn_hip_(qg, kg, vg, mask=mask, causal=True)
+    qc, kc, vc, ref, _ = _oracle(q, k, v, mask=mask, causal=True)
+    assert (out.float().cpu() - ref).abs().max().item() < 3e-2
+    g = torch.randn_like(out)
+    out.backward(g)
+    ref.backward(g.float().cpu())
+    for gt, rt, name in ((qg.grad, qc.grad, "dq"), (kg.grad, kc.grad, "dk"),
+                         (vg.grad, vc.grad, "dv")):
+        e = (gt.float().cpu() - rt).abs().max().item()
+        assert e / (rt.abs().max().item() + 1e-6) < 5e-2, f"{name} err {e}"
