@@ -158,6 +158,7 @@ class RingFlashAttentionHIPFunction(Function):
             qtiles = (n + 255) // 256
             kv_split = min(16, (n_total + 127) // 128,
                            max(1, 192 // max(1, qtiles * b * h)))
+            kv_split = int(os.environ.get("RING_ATTN_KV_SPLIT", kv_split))
             if kv_split > 1:
                 o_part = torch.empty(kv_split, b, h, d, n, device=q.device, dtype=torch.float32)
                 m_part = torch.empty(kv_split, b, h, n, device=q.device, dtype=torch.float32)
@@ -183,6 +184,7 @@ class RingFlashAttentionHIPFunction(Function):
         qtiles = (n + 255) // 256
         kv_tiles = (n + 63) // 64
         kv_split = min(16, kv_tiles, max(1, 192 // max(1, qtiles * b * h)))
+        kv_split = int(os.environ.get("RING_ATTN_KV_SPLIT", kv_split))
 
         multi = hops > 1
         o_acc = m = l = None
