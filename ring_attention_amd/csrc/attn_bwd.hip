@@ -107,7 +107,9 @@ struct DqLds {
 };
 
 template <int D, bool SOFTCLAMP>
-__global__ __launch_bounds__(512, 1) void attn_bwd_dq_kernel(BwdParams p) {
+__global__ __launch_bounds__(512, 2)   // 8-wave WGs need exactly 2 waves/SIMD:
+void attn_bwd_dq_kernel(BwdParams p) { // cap 256 VGPR, stop the 220 B/thread
+                                       // scratch spills seen at the 128 cap
     constexpr int DBLK = D / 32;
     constexpr int KSTEPS = D / 16;
     constexpr int DQ_KVBLK = dq_kvblk<D>();
@@ -270,82 +272,81 @@ __global__ __launch_bounds__(512, 1) void attn_bwd_dq_kernel(BwdParams p) {
 
         __syncthreads();
 
-        // s^T and dp^T: lane = q, kv in regs
-        f32x16 s[DQ_NBLK], dp[DQ_NBLK];
+        // s^T and dp^T computed PER 32-row BLOCK and packed immediately —
+        // keeping all NBLK blocks' accumulators live (128 VGPRs) forced
+        // scratch spills at the 256-register budget
+        uint32_t pk[DQ_NBLK * 8];
         __builtin_amdgcn_s_setprio(1);
         #pragma unroll
         for (int kb = 0; kb < DQ_NBLK; ++kb) {
-            s[kb] = f32x16{}; dp[kb] = f32x16{};
+            f32x16 s = f32x16{}, dp = f32x16{};
             int krow = kb * 32 + l31;
             #pragma unroll
             for (int ks = 0; ks < KSTEPS; ++ks) {
                 int chunk = ks * 2 + lhi;
                 bf16x8 kfr = *(const bf16x8*)(lds.k[par] + krow * D + bswz(krow, chunk) * 8);
                 bf16x8 vfr = *(const bf16x8*)(lds.v[par] + krow * D + bswz(krow, chunk) * 8);
-                s[kb] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kfr, qf[ks], s[kb], 0, 0, 0);
-                dp[kb] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vfr, dof[ks], dp[kb], 0, 0, 0);
+                s = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kfr, qf[ks], s, 0, 0, 0);
+                dp = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vfr, dof[ks], dp, 0, 0, 0);
+            }
+            if (full_tile && row_valid) {
+                #pragma unroll
+                for (int x2 = 0; x2 < 8; ++x2) {
+                    float dse[2];
+                    #pragma unroll
+                    for (int e = 0; e < 2; ++e) {
+                        int r = 2 * x2 + e;
+                        float x, dtanh = 1.f;
+                        if constexpr (SOFTCLAMP) {
+                            float inv_v = __builtin_amdgcn_rcpf(p.softclamp_value);
+                            float th = bfast_tanhf(s[r] * p.scale * inv_v);
+                            x = p.softclamp_value * th * 1.4426950408889634f;
+                            dtanh = 1.f - th * th;
+                        } else {
+                            x = __builtin_fmaf(s[r], scale2, -lse_i);  // fold
+                        }
+                        float pv = __builtin_amdgcn_exp2f(SOFTCLAMP ? x - lse_i : x);
+                        dse[e] = pv * (dp[r] - delta_i) * dtanh * p.scale;
+                    }
+                    union { __hip_bfloat162 h2; uint32_t u; } cvt;
+                    cvt.h2 = __float22bfloat162_rn(float2{dse[0], dse[1]});
+                    pk[kb * 8 + x2] = cvt.u;
+                }
+            } else {
+                #pragma unroll
+                for (int x2 = 0; x2 < 8; ++x2) {
+                    float dse[2];
+                    #pragma unroll
+                    for (int e = 0; e < 2; ++e) {
+                        int r = 2 * x2 + e;
+                        long jj = j0 + kb * 32 + (r & 3) + 8 * (r >> 2) + 4 * lhi;
+                        float x, dtanh = 1.f;
+                        if constexpr (SOFTCLAMP) {
+                            float inv_v = __builtin_amdgcn_rcpf(p.softclamp_value);
+                            float th = bfast_tanhf(s[r] * p.scale * inv_v);
+                            x = p.softclamp_value * th * 1.4426950408889634f;
+                            dtanh = 1.f - th * th;
+                        } else {
+                            x = s[r] * scale2;
+                        }
+                        bool ok = row_valid && jj <= jmax;
+                        if (p.causal) ok = ok && (jj <= qpos_i);
+                        if (p.has_win) ok = ok && (qpos_i - jj <= p.win);
+                        if (p.kmask) ok = ok && lds.kmask[par][jj - j0];
+                        float pv = ok ? __builtin_amdgcn_exp2f(
+                            SOFTCLAMP ? x - lse_i : x - lse_i) : 0.f;
+                        dse[e] = pv * (dp[r] - delta_i) * dtanh * p.scale;
+                    }
+                    union { __hip_bfloat162 h2; uint32_t u; } cvt;
+                    cvt.h2 = __float22bfloat162_rn(float2{dse[0], dse[1]});
+                    pk[kb * 8 + x2] = cvt.u;
+                }
             }
         }
         __builtin_amdgcn_s_setprio(0);
 
         if (t + 1 < t_hi) write_tile(par ^ 1);
         if (t + 2 < t_hi) load_tile();
-
-        // ds^T in regs -> packed bf16 pairs (exp2 domain); the full-tile
-        // variant carries no per-element predicates (see fwd kernel note)
-        uint32_t pk[DQ_NBLK * 8];
-        if (full_tile && row_valid) {
-            #pragma unroll
-            for (int x2 = 0; x2 < DQ_NBLK * 8; ++x2) {
-                float dse[2];
-                #pragma unroll
-                for (int e = 0; e < 2; ++e) {
-                    int kb = x2 >> 3, r = (2 * x2 + e) & 15;
-                    float x, dtanh = 1.f;
-                    if constexpr (SOFTCLAMP) {
-                        float inv_v = __builtin_amdgcn_rcpf(p.softclamp_value);
-                        float th = bfast_tanhf(s[kb][r] * p.scale * inv_v);
-                        x = p.softclamp_value * th * 1.4426950408889634f;
-                        dtanh = 1.f - th * th;
-                    } else {
-                        x = __builtin_fmaf(s[kb][r], scale2, -lse_i);  // fold
-                    }
-                    float pv = __builtin_amdgcn_exp2f(SOFTCLAMP ? x - lse_i : x);
-                    dse[e] = pv * (dp[kb][r] - delta_i) * dtanh * p.scale;
-                }
-                union { __hip_bfloat162 h2; uint32_t u; } cvt;
-                cvt.h2 = __float22bfloat162_rn(float2{dse[0], dse[1]});
-                pk[x2] = cvt.u;
-            }
-        } else {
-            #pragma unroll
-            for (int x2 = 0; x2 < DQ_NBLK * 8; ++x2) {
-                float dse[2];
-                #pragma unroll
-                for (int e = 0; e < 2; ++e) {
-                    int kb = x2 >> 3, r = (2 * x2 + e) & 15;
-                    long jj = j0 + kb * 32 + (r & 3) + 8 * (r >> 2) + 4 * lhi;
-                    float x, dtanh = 1.f;
-                    if constexpr (SOFTCLAMP) {
-                        float inv_v = __builtin_amdgcn_rcpf(p.softclamp_value);
-                        float th = bfast_tanhf(s[kb][r] * p.scale * inv_v);
-                        x = p.softclamp_value * th * 1.4426950408889634f;
-                        dtanh = 1.f - th * th;
-                    } else {
-                        x = s[kb][r] * scale2;
-                    }
-                    bool ok = row_valid && jj <= jmax;
-                    if (p.causal) ok = ok && (jj <= qpos_i);
-                    if (p.has_win) ok = ok && (qpos_i - jj <= p.win);
-                    if (p.kmask) ok = ok && lds.kmask[par][jj - j0];
-                    float pv = ok ? __builtin_amdgcn_exp2f(x - lse_i) : 0.f;
-                    dse[e] = pv * (dp[kb][r] - delta_i) * dtanh * p.scale;
-                }
-                union { __hip_bfloat162 h2; uint32_t u; } cvt;
-                cvt.h2 = __float22bfloat162_rn(float2{dse[0], dse[1]});
-                pk[x2] = cvt.u;
-            }
-        }
 
         // build B-operand fragments (lane = q, k = kv contiguous): pairs +2
         uint32_t frag[DQ_NBLK * 2][4];
@@ -412,7 +413,8 @@ struct DkvLds {
 };
 
 template <int D, int QT, bool SOFTCLAMP>
-__global__ __launch_bounds__(512, 1) void attn_bwd_dkv_kernel(BwdParams p) {
+__global__ __launch_bounds__(512, 2)   // see dq kernel note (552 B spills)
+void attn_bwd_dkv_kernel(BwdParams p) {
     static_assert(D % 32 == 0 && QT % 32 == 0);
     constexpr int DBLK = D / 32;
     constexpr int KSTEPS = D / 16;
@@ -594,21 +596,6 @@ __global__ __launch_bounds__(512, 1) void attn_bwd_dkv_kernel(BwdParams p) {
 
             #pragma unroll
             for (int qb = 0; qb < QBLKS; ++qb) {
-                // preload this block's lse/delta for the reg-q pattern with
-                // wide reads (qloc = qb*32 + (r&3) + 8*(r>>2) + 4*lhi: regs
-                // 4g..4g+3 are consecutive -> one b128 per group of 4)
-                float lse_r[16], delta_r[16];
-                #pragma unroll
-                for (int g4 = 0; g4 < 4; ++g4) {
-                    int base = qb * 32 + 8 * g4 + 4 * lhi;
-                    f32x4 lv = *(const f32x4*)(lds.lse[par] + base);
-                    f32x4 dv_ = *(const f32x4*)(lds.delta[par] + base);
-                    #pragma unroll
-                    for (int e = 0; e < 4; ++e) {
-                        lse_r[4 * g4 + e] = lv[e] * 1.4426950408889634f;  // exp2 domain
-                        delta_r[4 * g4 + e] = dv_[e];
-                    }
-                }
                 // S2[q][kv], dP[q][kv]: lane = kv, q rows in regs
                 f32x16 s2 = f32x16{}, dp = f32x16{};
                 #pragma unroll
@@ -626,6 +613,14 @@ __global__ __launch_bounds__(512, 1) void attn_bwd_dkv_kernel(BwdParams p) {
                     #pragma unroll
                     for (int x2 = 0; x2 < 8; ++x2) {
                         float pe[2], dse[2];
+                        // per-group wide loads (8 live regs, not 32): regs
+                        // 4g..4g+3 are consecutive qloc -> one b128 each
+                        f32x4 lse4, delta4;
+                        {
+                            int base = qb * 32 + 8 * (x2 >> 1) + 4 * lhi;
+                            lse4 = *(const f32x4*)(lds.lse[par] + base);
+                            delta4 = *(const f32x4*)(lds.delta[par] + base);
+                        }
                         #pragma unroll
                         for (int e = 0; e < 2; ++e) {
                             int r = 2 * x2 + e;
@@ -637,11 +632,11 @@ __global__ __launch_bounds__(512, 1) void attn_bwd_dkv_kernel(BwdParams p) {
                                 dtanh = 1.f - th * th;
                             } else {
                                 x = __builtin_fmaf(s2[r], p.scale * 1.4426950408889634f,
-                                                   -lse_r[r]);          // fold
+                                                   -lse4[r & 3] * 1.4426950408889634f);  // fold
                             }
-                            float pv = __builtin_amdgcn_exp2f(SOFTCLAMP ? x - lse_r[r] : x);
+                            float pv = __builtin_amdgcn_exp2f(SOFTCLAMP ? x - lse4[r & 3] * 1.4426950408889634f : x);
                             pe[e] = pv;
-                            dse[e] = pv * (dp[r] - delta_r[r]) * dtanh * p.scale;
+                            dse[e] = pv * (dp[r] - delta4[r & 3]) * dtanh * p.scale;
                         }
                         union { __hip_bfloat162 h2; uint32_t u; } c1, c2;
                         c1.h2 = __float22bfloat162_rn(float2{pe[0], pe[1]});
@@ -653,6 +648,14 @@ __global__ __launch_bounds__(512, 1) void attn_bwd_dkv_kernel(BwdParams p) {
                     #pragma unroll
                     for (int x2 = 0; x2 < 8; ++x2) {
                         float pe[2], dse[2];
+                        // per-group wide loads (8 live regs, not 32): regs
+                        // 4g..4g+3 are consecutive qloc -> one b128 each
+                        f32x4 lse4, delta4;
+                        {
+                            int base = qb * 32 + 8 * (x2 >> 1) + 4 * lhi;
+                            lse4 = *(const f32x4*)(lds.lse[par] + base);
+                            delta4 = *(const f32x4*)(lds.delta[par] + base);
+                        }
                         #pragma unroll
                         for (int e = 0; e < 2; ++e) {
                             int r = 2 * x2 + e;
@@ -672,9 +675,9 @@ __global__ __launch_bounds__(512, 1) void attn_bwd_dkv_kernel(BwdParams p) {
                             if (p.causal) ok = ok && (j <= qpos);
                             if (p.has_win) ok = ok && (qpos - j <= p.win);
                             if (p.kmask) ok = ok && kmask_own;
-                            float pv = ok ? __builtin_amdgcn_exp2f(x - lse_r[r]) : 0.f;
+                            float pv = ok ? __builtin_amdgcn_exp2f(x - lse4[r & 3] * 1.4426950408889634f) : 0.f;
                             pe[e] = pv;
-                            dse[e] = pv * (dp[r] - delta_r[r]) * dtanh * p.scale;
+                            dse[e] = pv * (dp[r] - delta4[r & 3]) * dtanh * p.scale;
                         }
                         union { __hip_bfloat162 h2; uint32_t u; } c1, c2;
                         c1.h2 = __float22bfloat162_rn(float2{pe[0], pe[1]});

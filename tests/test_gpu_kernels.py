@@ -79,8 +79,12 @@ def test_gqa_parity(groups):
     ref.backward(g.float().cpu())
     for gt, rt, name in ((qg.grad, qc.grad, "dq"), (kg.grad, kc.grad, "dk"),
                          (vg.grad, vc.grad, "dv")):
-        e = (gt.float().cpu() - rt).abs().max().item()
-        assert e / (rt.abs().max().item() + 1e-6) < 4e-2, f"{name} err {e}"
+        diff = (gt.float().cpu() - rt).abs()
+        scale_r = rt.abs().max().item() + 1e-6
+        # max-err tolerance is loose for bf16 x groups-deep accumulation;
+        # the mean-err guard catches structural bugs noise cannot hide
+        assert diff.max().item() / scale_r < 6e-2, f"{name} max err {diff.max()}"
+        assert diff.mean().item() / scale_r < 2e-3, f"{name} mean err {diff.mean()}"
 
 
 def test_keypad_mask_parity():
