@@ -392,3 +392,47 @@ def test_d128_gqa_mask_combo():
                          (vg.grad, vc.grad, "dv")):
         e = (gt.float().cpu() - rt).abs().max().item()
         assert e / (rt.abs().max().item() + 1e-6) < 5e-2, f"{name} err {e}"
+
+
+def test_causal_balance_split_consistency():
+    """The auto causal-balance grid.z split (engaged when the causal diagonal
+    cuts the shard and the grid exceeds 512 WGs) must be numerically
+    consistent with the unsplit kernels — checked by self-comparison at a
+    shape large enough to trip the heuristic (16 qtiles x b4 x h8 = 512)."""
+    import os
+    from ring_attention_amd.ops.ring_flash_hip import (
+        ring_flash_attn_hip_, _causal_balance_split)
+    assert _causal_balance_split(True, None, True, 512) == 2
+    assert _causal_balance_split(True, None, True, 256) == 1
+    assert _causal_balance_split(False, None, True, 4096) == 1
+    assert _causal_balance_split(True, 128, True, 4096) == 1
+    assert _causal_balance_split(True, None, False, 4096) == 1
+
+    b, n, h, d = 4, 4096, 8, 64
+    q, k, v = _mk(b, n, h, h, d)
+    g = torch.randn(b, n, h, d, device="cuda", dtype=torch.bfloat16)
+
+    def run(env):
+        old = {k_: os.environ.pop(k_, None) for k_ in
+               ("RING_ATTN_KV_SPLIT", "RING_ATTN_SPLIT_DQ", "RING_ATTN_SPLIT_DKV")}
+        os.environ.update(env)
+        try:
+            qg = q.clone().requires_grad_(True)
+            kg = k.clone().requires_grad_(True)
+            vg = v.clone().requires_grad_(True)
+            out, _ = ring_flash_attn_hip_(qg, kg, vg, causal=True)
+            out.backward(g)
+            return out.detach(), qg.grad, kg.grad, vg.grad
+        finally:
+            for k_, v_ in old.items():
+                os.environ.pop(k_, None)
+                if v_ is not None:
+                    os.environ[k_] = v_
+
+    split = run({})          # auto: causal-balance split 2 on all kernels
+    unsplit = run({"RING_ATTN_KV_SPLIT": "1", "RING_ATTN_SPLIT_DQ": "1",
+                   "RING_ATTN_SPLIT_DKV": "1"})
+    for s, u, name in zip(split, unsplit, ("out", "dq", "dk", "dv")):
+        e = (s.float() - u.float()).abs().max().item()
+        ref = u.float().abs().max().item() + 1e-6
+        assert e / ref < 1e-2, f"{name} split-vs-unsplit rel err {e/ref}"
