@@ -389,7 +389,8 @@ void attn_bwd_dq_kernel(BwdParams p) { // cap 256 VGPR, stop the 220 B/thread
         for (int r = 0; r < 16; ++r) {
             int d = db * 32 + (r & 3) + 8 * (r >> 2) + 4 * lhi;
             if (p.split > 1) atomicAdd(dqp + d, dq_acc[db][r]);
-            else dqp[d] += dq_acc[db][r];
+            else if (p.accumulate) dqp[d] += dq_acc[db][r];
+            else dqp[d] = dq_acc[db][r];
         }
 }
 
@@ -458,6 +459,11 @@ void attn_bwd_dkv_kernel(BwdParams p) {
     const int num_q_tiles = (int)((p.nq + QT - 1) / QT);
 
     for (int g = 0; g < p.group; ++g) {
+        // group boundary barrier: the previous group's last tile is still
+        // being read from LDS by slower waves when this group's prologue
+        // write_qtile targets the same buffer (GQA-only race; widest at
+        // split>1 where a chunk is a single tile)
+        if (g > 0) __syncthreads();
         const int h = hkh * p.group + g;
         const float* lse_row = p.lse + ((long)b * p.h + h) * p.nq;
         const float* delta_row = p.delta + ((long)b * p.h + h) * p.nq;
@@ -779,6 +785,51 @@ void launch_attn_bwd_dkv(const BwdParams& p, int head_dim, hipStream_t stream) {
     } else if (head_dim == 128) {
         if (p.softclamp) hipLaunchKernelGGL((attn_bwd_dkv_kernel<128, 32, true>), grid_dkv, block, 0, stream, p);
         else hipLaunchKernelGGL((attn_bwd_dkv_kernel<128, 32, false>), grid_dkv, block, 0, stream, p);
+    } else {
+        __builtin_trap();
+    }
+}
+
+// ---------------------------------------------------------------------------
+// delta = rowsum(dO * O) preprocess, fused over the bf16 inputs
+// (replaces two f32 casts + mul + reduce torch kernels per backward)
+// ---------------------------------------------------------------------------
+template <int D>
+__global__ __launch_bounds__(256) void attn_delta_kernel(DeltaParams p) {
+    constexpr int LPR = D / 8;                 // lanes per row (16B each)
+    constexpr int RPW = 64 / LPR;              // rows per wave
+    const int lane = threadIdx.x & 63;
+    const int wave = (blockIdx.x * (blockDim.x >> 6)) + (threadIdx.x >> 6);
+    const long row = (long)wave * RPW + lane / LPR;
+    if (row >= p.rows) return;
+    const int c = lane % LPR;
+
+    const __bf16* dop = (const __bf16*)p.dout + row * D + c * 8;
+    const __bf16* op = (const __bf16*)p.out + row * D + c * 8;
+    bf16x8 a = *(const bf16x8*)dop;
+    bf16x8 b = *(const bf16x8*)op;
+    float s = 0.f;
+    #pragma unroll
+    for (int e = 0; e < 8; ++e) s += (float)a[e] * (float)b[e];
+    #pragma unroll
+    for (int off = LPR / 2; off > 0; off >>= 1) s += __shfl_xor(s, off);
+    if (c == 0) {
+        // row = (bb * n + i) * h + hh  ->  delta[(bb * h + hh) * n + i]
+        const long hh = row % p.h;
+        const long bi = row / p.h;
+        const long i = bi % p.n, bb = bi / p.n;
+        p.delta[(bb * p.h + hh) * p.n + i] = s;
+    }
+}
+
+void launch_attn_delta(const DeltaParams& p, int head_dim, hipStream_t stream) {
+    dim3 block(256);
+    if (head_dim == 64) {
+        long waves = (p.rows + 7) / 8;
+        hipLaunchKernelGGL(attn_delta_kernel<64>, dim3((waves + 3) / 4), block, 0, stream, p);
+    } else if (head_dim == 128) {
+        long waves = (p.rows + 3) / 4;
+        hipLaunchKernelGGL(attn_delta_kernel<128>, dim3((waves + 3) / 4), block, 0, stream, p);
     } else {
         __builtin_trap();
     }

@@ -99,6 +99,17 @@ struct BwdParams {
 void launch_attn_bwd_dq(const BwdParams& p, int head_dim, hipStream_t stream);
 void launch_attn_bwd_dkv(const BwdParams& p, int head_dim, hipStream_t stream);
 
+struct DeltaParams {
+    const void* dout;       // bf16 (B, N, H, D)
+    const void* out;        // bf16 (B, N, H, D)
+    float* delta;           // fp32 (B, H, N)
+    long rows;              // B*N*H
+    long n;
+    int h;
+};
+
+void launch_attn_delta(const DeltaParams& p, int head_dim, hipStream_t stream);
+
 struct DecodeParams {
     const void* q;          // bf16 (B, H, 1, D)
     const void* k;          // bf16 (B, H, N, D)

@@ -205,6 +205,21 @@ std::vector<at::Tensor> decode_partial(at::Tensor q, at::Tensor k, at::Tensor v)
     return {out, lse};
 }
 
+at::Tensor attn_delta(at::Tensor dout, at::Tensor out) {
+    // dout, out (B,N,H,D) bf16 -> delta fp32 (B,H,N) = rowsum(dout*out)
+    CHECK_BF16_CONTIG(dout); CHECK_BF16_CONTIG(out);
+    const int64_t B = dout.size(0), N = dout.size(1), H = dout.size(2), D = dout.size(3);
+    TORCH_CHECK(D == 64 || D == 128, "head dim must be 64 or 128");
+    auto delta = at::empty({B, H, N}, dout.options().dtype(at::kFloat));
+    DeltaParams p{};
+    p.dout = dout.data_ptr(); p.out = out.data_ptr();
+    p.delta = delta.data_ptr<float>();
+    p.rows = B * N * H; p.n = N; p.h = (int)H;
+    launch_attn_delta(p, (int)D, at::hip::getCurrentHIPStream());
+    TORCH_CHECK(hipGetLastError() == hipSuccess, "delta launch failed");
+    return delta;
+}
+
 }  // namespace ring_attn
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
@@ -212,5 +227,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
     mod.def("attn_fwd_merge", &ring_attn::attn_fwd_merge, "merge kv-split partials");
     mod.def("attn_bwd", &ring_attn::attn_bwd, "CDNA4 flash attention backward");
     mod.def("decode_partial", &ring_attn::decode_partial, "CDNA4 single-query decode partial");
+    mod.def("attn_delta", &ring_attn::attn_delta, "fused delta = rowsum(dO*O) preprocess");
     mod.def("rotary_apply", &ring_attn::rotary_apply, "fused rotary embedding (table-driven)");
 }

@@ -235,19 +235,18 @@ class RingFlashAttentionHIPFunction(Function):
         assert active, "every hop masked — degenerate configuration"
         first_active, last_active = active[0], active[-1]
 
-        kv = torch.stack((kb, vb))
-        ring_tensors = (kv,) if mask_u8 is None else (kv, mask_u8)
+        ring_tensors = (kb, vb) if mask_u8 is None else (kb, vb, mask_u8)
 
         for info, tensors in all_ring_pass(topo, *ring_tensors, max_hops=hops):
             skip, diag, win, ksp = plan[info.hop]
             if skip:
                 continue
-            kv_t = tensors[0]
-            mk = tensors[1] if mask_u8 is not None else None
+            k_t, v_t = tensors[0], tensors[1]
+            mk = tensors[2] if mask_u8 is not None else None
             is_f = info.hop == first_active
             is_l = info.hop == last_active
             if ksp > 1:
-                ext.attn_fwd(qb, kv_t[0], kv_t[1], mk,
+                ext.attn_fwd(qb, k_t, v_t, mk,
                              o_part, m_part, l_part, None, None,
                              scale, causal, diag, 1, win, lookback is not None,
                              softclamp_qk_sim, softclamp_value,
@@ -256,7 +255,7 @@ class RingFlashAttentionHIPFunction(Function):
                                    out if is_l else None, lse if is_l else None,
                                    ksp, b, h, d, n, is_f, is_l)
             else:
-                ext.attn_fwd(qb, kv_t[0], kv_t[1], mk,
+                ext.attn_fwd(qb, k_t, v_t, mk,
                              o_acc, m, l, out, lse,
                              scale, causal, diag, 1, win, lookback is not None,
                              softclamp_qk_sim, softclamp_value,
@@ -284,10 +283,13 @@ class RingFlashAttentionHIPFunction(Function):
         ext = hip_ext.require()
 
         dob = do.to(torch.bfloat16).contiguous()
-        # delta = rowsum(do * o) in fp32: (b, h, n)
-        delta = (dob.float() * out.float()).sum(dim=-1).permute(0, 2, 1).contiguous()
+        # delta = rowsum(do * o) in fp32: (b, h, n), fused HIP kernel
+        delta = ext.attn_delta(dob, out)
 
-        dq = torch.zeros(b, n, h, d, device=qb.device, dtype=torch.float32)
+        def _alloc_dq(zeroed):
+            fac = torch.zeros if zeroed else torch.empty
+            return fac(b, n, h, d, device=qb.device, dtype=torch.float32)
+
         rq = topo.ring_rank
 
         if strategy == "allgather":
@@ -299,8 +301,6 @@ class RingFlashAttentionHIPFunction(Function):
             n_total = n * R
             q_stride = R if striped else 1
             diag = rq if striped else rq * n
-            dk_full = torch.zeros(b, hk, n_total, d, device=qb.device, dtype=torch.float32)
-            dv_full = torch.zeros(b, hk, d, n_total, device=qb.device, dtype=torch.float32)
             # grid.z splits keep both kernels filling the CUs when their
             # natural grids are small (fp32 atomics, contention = split)
             qtiles = (n + 255) // 256
@@ -313,6 +313,10 @@ class RingFlashAttentionHIPFunction(Function):
                 causal, lookback, True, kvtiles_t * b * hk))
             split_dq = int(os.environ.get("RING_ATTN_SPLIT_DQ", split_dq))
             split_dkv = int(os.environ.get("RING_ATTN_SPLIT_DKV", split_dkv))
+            dq = _alloc_dq(zeroed=split_dq > 1)
+            fac = torch.zeros if split_dkv > 1 else torch.empty
+            dk_full = fac(b, hk, n_total, d, device=qb.device, dtype=torch.float32)
+            dv_full = fac(b, hk, d, n_total, device=qb.device, dtype=torch.float32)
             ext.attn_bwd(qb, k_full, v_full, dob, m_full, lse, delta,
                          dq, dk_full, dv_full, scale, causal, diag, q_stride,
                          0, False, softclamp_qk_sim, softclamp_value, False, split_dq, 1)
@@ -337,43 +341,55 @@ class RingFlashAttentionHIPFunction(Function):
         env_dq = os.environ.get("RING_ATTN_SPLIT_DQ")
         env_dkv = os.environ.get("RING_ATTN_SPLIT_DKV")
 
-        kv = torch.stack((kb, vb))
-        ring_tensors = (kv,) if mask_u8 is None else (kv, mask_u8)
+        # hop plan first: per-hop splits (only the diagonal hop is causally
+        # imbalanced; full hops keep the plain fill heuristic) + whether the
+        # first compute hop can write dq plainly (no zero-init, no read)
+        plan = []
+        for hop in range(hops):
+            rk_ = topo.source_of_hop(hop)
+            skip, diag, win = _hop_geometry(rq, rk_, n, topo.ring_size,
+                                            striped, causal, lookback)
+            cuts = diag < n
+            split_dq = max(split_dq_base, _causal_balance_split(
+                causal, lookback, cuts, qtiles * b * h))
+            split_dkv = max(split_dkv_base, _causal_balance_split(
+                causal, lookback, cuts, qtiles * b * hk))
+            if env_dq is not None:
+                split_dq = int(env_dq)
+            if env_dkv is not None:
+                split_dkv = int(env_dkv)
+            plan.append((skip, diag, win, split_dq, split_dkv))
+        first_active = next(i for i, p_ in enumerate(plan) if not p_[0])
+        dq_zeroed = plan[first_active][3] > 1   # atomics need a zeroed base
+        dq = _alloc_dq(zeroed=dq_zeroed)
+
+        ring_tensors = (kb, vb) if mask_u8 is None else (kb, vb, mask_u8)
         acc = RingAccumulator(topo)
 
         for info, tensors in all_ring_pass(topo, *ring_tensors, max_hops=hops):
-            rk = info.source_ring_rank
-            skip, diag, win = _hop_geometry(rq, rk, n, topo.ring_size, striped,
-                                            causal, lookback)
-            kv_t = tensors[0]
-            mk = tensors[1] if mask_u8 is not None else None
+            skip, diag, win, split_dq, split_dkv = plan[info.hop]
+            k_t, v_t = tensors[0], tensors[1]
+            mk = tensors[2] if mask_u8 is not None else None
             # circulate dk/dv flat in the kernel's native scratch layouts
             # (dk (b,hk,n,d), dv (b,hk,d,n)); elementwise accumulation is
-            # layout-agnostic, so only ONE final permute happens at home
-            contrib = torch.zeros(2, b * hk * n * d, device=qb.device,
-                                  dtype=torch.float32)
+            # layout-agnostic, so only ONE final permute happens at home.
+            # The kernels overwrite every element at split 1, so only the
+            # atomic (split>1) and skip (circulated as-is) cases zero-init.
+            fac = torch.zeros if (skip or split_dkv > 1) else torch.empty
+            contrib = fac(2, b * hk * n * d, device=qb.device,
+                          dtype=torch.float32)
             if not skip:
                 dk_n = contrib[0].view(b, hk, n, d)
                 dv_n = contrib[1].view(b, hk, d, n)
-                # per-hop splits: only the diagonal hop is causally
-                # imbalanced; full hops keep the plain fill heuristic
-                cuts = diag < n
-                split_dq = max(split_dq_base, _causal_balance_split(
-                    causal, lookback, cuts, qtiles * b * h))
-                split_dkv = max(split_dkv_base, _causal_balance_split(
-                    causal, lookback, cuts, qtiles * b * hk))
-                if env_dq is not None:
-                    split_dq = int(env_dq)
-                if env_dkv is not None:
-                    split_dkv = int(env_dkv)
+                dq_acc_flag = dq_zeroed or info.hop != first_active
                 # dq + dk/dv kernels (sequential: their LDS footprints do
                 # not co-reside, so stream-splitting buys nothing)
-                ext.attn_bwd(qb, kv_t[0], kv_t[1], dob, mk, lse, delta,
+                ext.attn_bwd(qb, k_t, v_t, dob, mk, lse, delta,
                              dq, dk_n, dv_n,
                              scale, causal, diag, 1, win, lookback is not None,
-                             softclamp_qk_sim, softclamp_value, False,
+                             softclamp_qk_sim, softclamp_value, dq_acc_flag,
                              split_dq, 1)
-                ext.attn_bwd(qb, kv_t[0], kv_t[1], dob, mk, lse, delta,
+                ext.attn_bwd(qb, k_t, v_t, dob, mk, lse, delta,
                              dq, dk_n, dv_n,
                              scale, causal, diag, 1, win, lookback is not None,
                              softclamp_qk_sim, softclamp_value, False,
@@ -446,10 +462,11 @@ class FlashAttnOffsetFunction(Function):
         scale = d ** -0.5
         ext = hip_ext.require()
         dob = do.to(torch.bfloat16).contiguous()
-        delta = (dob.float() * out.float()).sum(dim=-1).permute(0, 2, 1).contiguous()
-        dq = torch.zeros(b, n, h, d, device=qb.device, dtype=torch.float32)
-        dk_n = torch.zeros(b, hk, nk, d, device=qb.device, dtype=torch.float32)
-        dv_n = torch.zeros(b, hk, d, nk, device=qb.device, dtype=torch.float32)
+        delta = ext.attn_delta(dob, out)
+        # accumulate=False + split=1: both kernels overwrite every element
+        dq = torch.empty(b, n, h, d, device=qb.device, dtype=torch.float32)
+        dk_n = torch.empty(b, hk, nk, d, device=qb.device, dtype=torch.float32)
+        dv_n = torch.empty(b, hk, d, nk, device=qb.device, dtype=torch.float32)
         ext.attn_bwd(qb, kb, vb, dob, None, lse, delta, dq, dk_n, dv_n,
                      scale, causal, q_offset, 1, 0, False, False, 50.0, False, 1, 0)
         dk = dk_n.permute(0, 2, 1, 3).contiguous()

@@ -81,9 +81,10 @@ def test_gqa_parity(groups):
                          (vg.grad, vc.grad, "dv")):
         diff = (gt.float().cpu() - rt).abs()
         scale_r = rt.abs().max().item() + 1e-6
-        # max-err tolerance is loose for bf16 x groups-deep accumulation;
-        # the mean-err guard catches structural bugs noise cannot hide
-        assert diff.max().item() / scale_r < 6e-2, f"{name} max err {diff.max()}"
+        # (this was once 6e-2 to paper over what turned out to be a missing
+        # group-boundary __syncthreads in the dkv kernel; fixed, the error
+        # is deterministic bf16 noise well under 4e-2)
+        assert diff.max().item() / scale_r < 4e-2, f"{name} max err {diff.max()}"
         assert diff.mean().item() / scale_r < 2e-3, f"{name} mean err {diff.mean()}"
 
 
