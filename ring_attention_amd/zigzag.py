@@ -117,6 +117,19 @@ def zig_zag_attn(
         from .ops.ring_flash_hip import flash_attn_offset
         n = q.shape[-2]
         half = n // 2
+        # the gather is RANK-major: [r0 chunks (0, 2W-1), r1 chunks (1, 2W-2),
+        # ...] — the offset-causal kernel needs KV in GLOBAL position order,
+        # so reorder chunks first (differentiable; no-op at world 1)
+        world = get_world_size()
+        if world > 1:
+            bk, hkk, N, dk_ = k.shape
+            ch = N // (2 * world)
+            r = torch.arange(world, device=k.device)
+            idx = torch.empty(2 * world, dtype=torch.long, device=k.device)
+            idx[r] = 2 * r                      # global chunk r <- rank r local 0
+            idx[2 * world - 1 - r] = 2 * r + 1  # global 2W-1-r <- rank r local 1
+            k = k.reshape(bk, hkk, 2 * world, ch, dk_).index_select(2, idx).reshape(bk, hkk, N, dk_)
+            v = v.reshape(bk, hkk, 2 * world, ch, v.shape[-1]).index_select(2, idx).reshape(bk, hkk, N, v.shape[-1])
         # to (b, n, h, d) layout for the kernel; pad keys are a global-order
         # suffix, so masking them is a slice (grads zero-pad automatically)
         q_ = q.permute(0, 2, 1, 3)

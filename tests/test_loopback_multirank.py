@@ -222,3 +222,90 @@ def test_loopback_tree_decode():
         e = (out.float() - ref.float()).abs().max().item()
         s = ref.float().abs().max().item() + 1e-6
         assert e / s < 2e-2, f"rank {rank} rel err {e/s}"
+
+
+@pytest.mark.gpu
+def test_loopback_transformer_gpu():
+    """Full RingTransformer (HIP kernels, rotary, sharded CE) at world 2 on
+    one GPU via the loopback layer, against the replicated non-ring twin."""
+    from ring_attention_amd import RingTransformer
+    world, seq = 2, 1024
+    model_kwargs = dict(
+        num_tokens=256, dim=256, depth=2, causal=True, dim_head=64, heads=4,
+        ff_mult=2, num_grouped_query_heads=2, bucket_size=256,
+        ring_seq_size=seq // world, striped_ring_attn=True,
+        use_hip_kernel=True,
+    )
+    torch.manual_seed(11)
+    ring_model = RingTransformer(ring_attn=True, **model_kwargs).cuda().bfloat16()
+    flat_model = RingTransformer(ring_attn=False, **model_kwargs).cuda().bfloat16()
+    flat_model.load_state_dict(ring_model.state_dict())
+
+    torch.manual_seed(200)
+    full_ids = torch.randint(0, 256, (world, seq), device="cuda")
+
+    ref_logits = flat_model(full_ids)
+    ref_loss = flat_model(full_ids, return_loss=True)
+    ref_loss.backward()
+
+    def run(rank):
+        ids = full_ids[rank:rank + 1]
+        logits = ring_model(ids)
+        loss = ring_model(ids, return_loss=True)
+        loss.backward()        # shared params: both ranks' grads SUM in-place
+        return logits.detach(), loss.detach()
+
+    results = loopback_world(world, run)
+    scale = ref_logits.float().abs().max().item()
+    for rank, (logits, loss) in enumerate(results):
+        e = (logits.float() - ref_logits[rank:rank + 1].float()).abs().max().item()
+        assert e / scale < 3e-2, f"rank {rank} logits rel err {e/scale}"
+    loss_avg = sum(r[1].item() for r in results) / world
+    assert abs(loss_avg - ref_loss.item()) < 5e-2, (loss_avg, ref_loss.item())
+    # both rank threads accumulated into the SAME parameter tensors, so
+    # .grad already holds the cross-rank sum; /world == the DDP average
+    g_avg = ring_model.token_emb.weight.grad.float() / world
+    gerr = (g_avg - flat_model.token_emb.weight.grad.float()).abs().max().item()
+    gs = flat_model.token_emb.weight.grad.float().abs().max().item() + 1e-6
+    assert gerr / gs < 6e-2, f"emb grad rel err {gerr/gs}"
+
+
+@pytest.mark.gpu
+def test_loopback_zigzag_gpu():
+    """zig-zag CP at world 2 through the loopback layer: all-gathered KV +
+    offset-causal HIP fast path vs the plain causal kernel on the full seq."""
+    from ring_attention_amd.zigzag import (zig_zag_attn, zig_zag_pad_seq,
+                                           zig_zag_shard)
+    from ring_attention_amd.ops.ring_flash_hip import ring_flash_attn_hip_
+    world = 2
+    b, h, n, d = 2, 4, 2048, 64
+    torch.manual_seed(23)
+    q = torch.randn(b, h, n, d, device="cuda", dtype=torch.bfloat16)
+    k = torch.randn(b, h, n, d, device="cuda", dtype=torch.bfloat16)
+    v = torch.randn(b, h, n, d, device="cuda", dtype=torch.bfloat16)
+
+    # fp32 eager reference (bounds the zigzag error alone, not two bf16 runs)
+    sim = torch.einsum("bhid,bhjd->bhij", q.float(), k.float()) * d ** -0.5
+    sim = sim.masked_fill(torch.ones(n, n, device="cuda", dtype=torch.bool)
+                          .triu(1), -torch.finfo(torch.float32).max)
+    ref = torch.einsum("bhij,bhjd->bhid", sim.softmax(-1), v.float())
+
+    def run(rank):
+        qp, _ = zig_zag_pad_seq(q)
+        kp, _ = zig_zag_pad_seq(k)
+        vp, _ = zig_zag_pad_seq(v)
+        (q_loc, q_pos, _), _ = zig_zag_shard(qp)
+        (k_loc, _, _), _ = zig_zag_shard(kp)
+        (v_loc, _, _), _ = zig_zag_shard(vp)
+        c = q_loc.shape[-2] // 2
+        starts = (int(q_pos[0].item()), int(q_pos[c].item()))
+        out = zig_zag_attn(q_loc, k_loc, v_loc, causal=True,
+                           q_chunk_starts=starts, kv_valid_len=n)
+        return out.detach(), q_pos
+
+    results = loopback_world(world, run)
+    scale = ref.float().abs().max().item()
+    for rank, (out, q_pos) in enumerate(results):
+        want = ref[:, :, q_pos]
+        e = (out.float() - want.float()).abs().max().item()
+        assert e / scale < 2e-2, f"rank {rank} zigzag rel err {e/scale}"
