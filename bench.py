@@ -82,13 +82,34 @@ def main():
     for _ in range(args.warmup):
         step()
 
+    # single-GPU: replay the whole fwd+bwd step as ONE hipGraph (cuts every
+    # per-launch host round-trip; the same kernels run the same work).
+    # Collectives make capture unsafe at world>1, so only when local.
+    graph = None
+    if on_gpu and world == 1 and not os.environ.get("RING_ATTN_BENCH_NO_GRAPH"):
+        try:
+            side = torch.cuda.Stream()
+            side.wait_stream(torch.cuda.current_stream())
+            with torch.cuda.stream(side):     # allocator warmup for capture
+                step()
+            torch.cuda.current_stream().wait_stream(side)
+            torch.cuda.synchronize()
+            g_ = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(g_):
+                step()
+            g_.replay()
+            torch.cuda.synchronize()
+            graph = g_
+        except Exception:
+            graph = None
+
     if distributed:
         torch.distributed.barrier()
     if on_gpu:
         torch.cuda.synchronize()
     t0 = time.perf_counter()
     for _ in range(args.steps):
-        step()
+        graph.replay() if graph is not None else step()
     if distributed:
         torch.distributed.barrier()
     if on_gpu:
@@ -135,6 +156,7 @@ def main():
                 "kv_heads": hk,
                 "d_head": d,
                 "parallelism": f"ring{world}",
+                "hipgraph": graph is not None,
                 "flop_convention": "fwd=4*b*n_shard*n_total*d*h (/2 causal); fwd+bwd=2.5x",
             },
         }))
