@@ -106,7 +106,7 @@ struct DqLds {
     unsigned char kmask[2][KVB];
 };
 
-template <int D, bool SOFTCLAMP>
+template <int D, bool SOFTCLAMP, bool PAIRED>
 __global__ __launch_bounds__(512, 2)   // 8-wave WGs need exactly 2 waves/SIMD:
 void attn_bwd_dq_kernel(BwdParams p) { // cap 256 VGPR, stop the 220 B/thread
                                        // scratch spills seen at the 128 cap
@@ -123,11 +123,20 @@ void attn_bwd_dq_kernel(BwdParams p) { // cap 256 VGPR, stop the 220 B/thread
     const int l31 = lane & 31;
     const int lhi = lane >> 5;
 
-    const int qtile = blockIdx.x;
     const int bh = blockIdx.y;
     const int b = bh / p.h;
     const int h = bh % p.h;
     const int hk = h / p.group;
+
+    // causal pairing: WG x runs q-tiles (x, T-1-x) — uniform per-WG work
+    for (int pit = 0; pit < (PAIRED ? 2 : 1); ++pit) {
+    const int qtile = PAIRED
+        ? (pit == 0 ? (int)blockIdx.x : p.paired - 1 - (int)blockIdx.x)
+        : (int)blockIdx.x;
+    if (PAIRED && pit == 1) {
+        if (qtile == (int)blockIdx.x) break;   // odd T: middle tile once
+        __syncthreads();                       // LDS handoff between tiles
+    }
 
     const long i = (long)qtile * DQ_QROWS_WG + wid * 32 + l31;
     const bool row_valid = i < p.nq;
@@ -379,7 +388,7 @@ void attn_bwd_dq_kernel(BwdParams p) { // cap 256 VGPR, stop the 220 B/thread
         __builtin_amdgcn_s_setprio(0);
     }
 
-    if (!row_valid) return;
+    if (!row_valid) continue;
     // epilogue: dq (B, Nq, H, D) fp32; unique writer per row unless the kv
     // walk is split across grid.z (then fp32 atomics, contention = split)
     float* dqp = p.dq + ((long)b * p.nq + i) * p.h * D + (long)h * D;
@@ -392,6 +401,7 @@ void attn_bwd_dq_kernel(BwdParams p) { // cap 256 VGPR, stop the 220 B/thread
             else if (p.accumulate) dqp[d] += dq_acc[db][r];
             else dqp[d] = dq_acc[db][r];
         }
+    }  // pair loop
 }
 
 // ---------------------------------------------------------------------------
@@ -413,7 +423,7 @@ struct DkvLds {
     __align__(16) float delta[2][QT];
 };
 
-template <int D, int QT, bool SOFTCLAMP>
+template <int D, int QT, bool SOFTCLAMP, bool PAIRED>
 __global__ __launch_bounds__(512, 2)   // see dq kernel note (552 B spills)
 void attn_bwd_dkv_kernel(BwdParams p) {
     static_assert(D % 32 == 0 && QT % 32 == 0);
@@ -430,10 +440,20 @@ void attn_bwd_dkv_kernel(BwdParams p) {
     const int l31 = lane & 31;
     const int lhi = lane >> 5;
 
-    const int kvtile = blockIdx.x;
     const int bhk = blockIdx.y;
     const int b = bhk / p.hk;
     const int hkh = bhk % p.hk;
+
+    // causal pairing (mirrored): kv-tile x attends T-x q-tiles, so WG x
+    // runs kv-tiles (x, T-1-x) for uniform per-WG work
+    for (int pit = 0; pit < (PAIRED ? 2 : 1); ++pit) {
+    const int kvtile = PAIRED
+        ? (pit == 0 ? (int)blockIdx.x : p.paired - 1 - (int)blockIdx.x)
+        : (int)blockIdx.x;
+    if (PAIRED && pit == 1) {
+        if (kvtile == (int)blockIdx.x) break;  // odd T: middle tile once
+        __syncthreads();                       // LDS handoff between tiles
+    }
 
     const long j0_wg = (long)kvtile * KVROWS_WG;
     const long jmax = min(j0_wg + KVROWS_WG, p.nk) - 1;
@@ -758,18 +778,25 @@ void attn_bwd_dkv_kernel(BwdParams p) {
                 else *dst = dv_acc[db][r];
             }
     }
+    }  // pair loop
 }
 
 void launch_attn_bwd_dq(const BwdParams& p, int head_dim, hipStream_t stream) {
     dim3 block(512);
     int z = p.split > 1 ? p.split : 1;
-    dim3 grid_dq((p.nq + DQ_QROWS_WG - 1) / DQ_QROWS_WG, p.b * p.h, z);
+    long qt_ = (p.nq + DQ_QROWS_WG - 1) / DQ_QROWS_WG;
+    dim3 grid_dq(p.paired ? (qt_ + 1) / 2 : qt_, p.b * p.h, z);
+    const bool pr = p.paired > 0;
     if (head_dim == 64) {
-        if (p.softclamp) hipLaunchKernelGGL((attn_bwd_dq_kernel<64, true>), grid_dq, block, 0, stream, p);
-        else hipLaunchKernelGGL((attn_bwd_dq_kernel<64, false>), grid_dq, block, 0, stream, p);
+        if (p.softclamp) if (pr) hipLaunchKernelGGL((attn_bwd_dq_kernel<64, true, true>), grid_dq, block, 0, stream, p);
+        else hipLaunchKernelGGL((attn_bwd_dq_kernel<64, true, false>), grid_dq, block, 0, stream, p);
+        else if (pr) hipLaunchKernelGGL((attn_bwd_dq_kernel<64, false, true>), grid_dq, block, 0, stream, p);
+        else hipLaunchKernelGGL((attn_bwd_dq_kernel<64, false, false>), grid_dq, block, 0, stream, p);
     } else if (head_dim == 128) {
-        if (p.softclamp) hipLaunchKernelGGL((attn_bwd_dq_kernel<128, true>), grid_dq, block, 0, stream, p);
-        else hipLaunchKernelGGL((attn_bwd_dq_kernel<128, false>), grid_dq, block, 0, stream, p);
+        if (p.softclamp) if (pr) hipLaunchKernelGGL((attn_bwd_dq_kernel<128, true, true>), grid_dq, block, 0, stream, p);
+        else hipLaunchKernelGGL((attn_bwd_dq_kernel<128, true, false>), grid_dq, block, 0, stream, p);
+        else if (pr) hipLaunchKernelGGL((attn_bwd_dq_kernel<128, false, true>), grid_dq, block, 0, stream, p);
+        else hipLaunchKernelGGL((attn_bwd_dq_kernel<128, false, false>), grid_dq, block, 0, stream, p);
     } else {
         __builtin_trap();
     }
@@ -778,13 +805,19 @@ void launch_attn_bwd_dq(const BwdParams& p, int head_dim, hipStream_t stream) {
 void launch_attn_bwd_dkv(const BwdParams& p, int head_dim, hipStream_t stream) {
     dim3 block(512);
     int z = p.split > 1 ? p.split : 1;
-    dim3 grid_dkv((p.nk + KVROWS_WG - 1) / KVROWS_WG, p.b * p.hk, z);
+    long kt_ = (p.nk + KVROWS_WG - 1) / KVROWS_WG;
+    dim3 grid_dkv(p.paired ? (kt_ + 1) / 2 : kt_, p.b * p.hk, z);
+    const bool pr = p.paired > 0;
     if (head_dim == 64) {
-        if (p.softclamp) hipLaunchKernelGGL((attn_bwd_dkv_kernel<64, 64, true>), grid_dkv, block, 0, stream, p);
-        else hipLaunchKernelGGL((attn_bwd_dkv_kernel<64, 64, false>), grid_dkv, block, 0, stream, p);
+        if (p.softclamp) if (pr) hipLaunchKernelGGL((attn_bwd_dkv_kernel<64, 64, true, true>), grid_dkv, block, 0, stream, p);
+        else hipLaunchKernelGGL((attn_bwd_dkv_kernel<64, 64, true, false>), grid_dkv, block, 0, stream, p);
+        else if (pr) hipLaunchKernelGGL((attn_bwd_dkv_kernel<64, 64, false, true>), grid_dkv, block, 0, stream, p);
+        else hipLaunchKernelGGL((attn_bwd_dkv_kernel<64, 64, false, false>), grid_dkv, block, 0, stream, p);
     } else if (head_dim == 128) {
-        if (p.softclamp) hipLaunchKernelGGL((attn_bwd_dkv_kernel<128, 32, true>), grid_dkv, block, 0, stream, p);
-        else hipLaunchKernelGGL((attn_bwd_dkv_kernel<128, 32, false>), grid_dkv, block, 0, stream, p);
+        if (p.softclamp) if (pr) hipLaunchKernelGGL((attn_bwd_dkv_kernel<128, 32, true, true>), grid_dkv, block, 0, stream, p);
+        else hipLaunchKernelGGL((attn_bwd_dkv_kernel<128, 32, true, false>), grid_dkv, block, 0, stream, p);
+        else if (pr) hipLaunchKernelGGL((attn_bwd_dkv_kernel<128, 32, false, true>), grid_dkv, block, 0, stream, p);
+        else hipLaunchKernelGGL((attn_bwd_dkv_kernel<128, 32, false, false>), grid_dkv, block, 0, stream, p);
     } else {
         __builtin_trap();
     }

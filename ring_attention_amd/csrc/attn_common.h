@@ -46,6 +46,9 @@ struct FwdParams {
     int kv_split;           // >1: grid.z splits the kv range; o_acc/m/l hold
                             // kv_split partials (merged by attn_fwd_merge)
     int ablate;             // diagnostics: 1 = stage first tile only
+    int paired;             // causal load balance: >0 = total q-tiles T;
+                            // grid.x = ceil(T/2), each WG runs tiles
+                            // (x, T-1-x) so per-WG causal work is uniform
     unsigned long long* ticks;  // diagnostics: per-tile segment s_memtime
                                 // stamps from block(0,0,0) wave 0 (or null)
 };
@@ -91,9 +94,11 @@ struct BwdParams {
     long q_stride;
     long win;
     int has_win;
-    int accumulate;
+    int accumulate;         // dk/dv + dq: 0 = overwrite, 1 = add to existing
     int split;              // >1: grid.z splits the contraction range; dq/dk/dv
-                            // accumulated with fp32 atomics instead of plain ops         // dk/dv: 0 = overwrite, 1 = add to existing
+                            // accumulated with fp32 atomics instead of plain ops
+    int paired;             // causal balance: >0 = total walk-parallel tiles T;
+                            // grid.x = ceil(T/2), WG x runs tiles (x, T-1-x)
 };
 
 void launch_attn_bwd_dq(const BwdParams& p, int head_dim, hipStream_t stream);

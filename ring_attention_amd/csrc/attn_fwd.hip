@@ -87,7 +87,7 @@ static constexpr float LN2 = 0.6931471805599453f;
 // ---------------------------------------------------------------------------
 // forward kernel
 // ---------------------------------------------------------------------------
-template <int D, bool SOFTCLAMP>
+template <int D, bool SOFTCLAMP, bool PAIRED>
 __global__ __launch_bounds__(NTHREADS, 1) void attn_fwd_kernel(FwdParams p) {
     static_assert(D % 32 == 0);
     constexpr int DBLK = D / 32;     // 32-d output blocks
@@ -103,11 +103,21 @@ __global__ __launch_bounds__(NTHREADS, 1) void attn_fwd_kernel(FwdParams p) {
     const int l31 = lane & 31;
     const int lhi = lane >> 5;        // 0 or 1
 
-    const int qtile = blockIdx.x;
     const int bh = blockIdx.y;
     const int b = bh / p.h;
     const int h = bh % p.h;
     const int hk = h / p.group;       // kv head
+
+    // causal pairing: WG x runs q-tiles (x, T-1-x) — per-WG work is the
+    // uniform T+1 tiles instead of the 2:1 triangle imbalance
+    for (int pit = 0; pit < (PAIRED ? 2 : 1); ++pit) {
+    const int qtile = PAIRED
+        ? (pit == 0 ? (int)blockIdx.x : p.paired - 1 - (int)blockIdx.x)
+        : (int)blockIdx.x;
+    if (PAIRED && pit == 1) {
+        if (qtile == (int)blockIdx.x) break;   // odd T: middle tile once
+        __syncthreads();                       // LDS handoff between tiles
+    }
 
     const long irow0 = (long)qtile * QROWS_WG + wid * QROWS_WAVE;  // this wave's first q row
     const long i = irow0 + l31;                                     // this lane's q row
@@ -463,7 +473,7 @@ __global__ __launch_bounds__(NTHREADS, 1) void attn_fwd_kernel(FwdParams p) {
     }
 
     // ---- epilogue
-    if (!row_valid) return;
+    if (!row_valid) continue;
 
     if (split_mode) {
         // write this split's unnormalized partial (merged by attn_fwd_merge)
@@ -479,7 +489,7 @@ __global__ __launch_bounds__(NTHREADS, 1) void attn_fwd_kernel(FwdParams p) {
                 int d = db * 32 + (r & 3) + 8 * (r >> 2) + 4 * lhi;
                 oa[(long)d * p.nq + i] = o_acc[db][r];
             }
-        return;
+        continue;
     }
 
     if (p.is_last) {
@@ -514,6 +524,7 @@ __global__ __launch_bounds__(NTHREADS, 1) void attn_fwd_kernel(FwdParams p) {
                 oa[(long)d * p.nq + i] = o_acc[db][r];
             }
     }
+    }  // pair loop
 }
 
 // ---------------------------------------------------------------------------
@@ -593,14 +604,21 @@ void launch_attn_fwd_merge(const FwdMergeParams& p, int head_dim, hipStream_t st
 }
 
 void launch_attn_fwd(const FwdParams& p, int head_dim, hipStream_t stream) {
-    dim3 grid((p.nq + QROWS_WG - 1) / QROWS_WG, p.b * p.h, p.kv_split > 1 ? p.kv_split : 1);
+    long qtiles = (p.nq + QROWS_WG - 1) / QROWS_WG;
+    dim3 grid(p.paired ? (qtiles + 1) / 2 : qtiles, p.b * p.h,
+              p.kv_split > 1 ? p.kv_split : 1);
     dim3 block(NTHREADS);
+    const bool pr = p.paired > 0;
     if (head_dim == 64) {
-        if (p.softclamp) hipLaunchKernelGGL((attn_fwd_kernel<64, true>), grid, block, 0, stream, p);
-        else hipLaunchKernelGGL((attn_fwd_kernel<64, false>), grid, block, 0, stream, p);
+        if (p.softclamp) if (pr) hipLaunchKernelGGL((attn_fwd_kernel<64, true, true>), grid, block, 0, stream, p);
+        else hipLaunchKernelGGL((attn_fwd_kernel<64, true, false>), grid, block, 0, stream, p);
+        else if (pr) hipLaunchKernelGGL((attn_fwd_kernel<64, false, true>), grid, block, 0, stream, p);
+        else hipLaunchKernelGGL((attn_fwd_kernel<64, false, false>), grid, block, 0, stream, p);
     } else if (head_dim == 128) {
-        if (p.softclamp) hipLaunchKernelGGL((attn_fwd_kernel<128, true>), grid, block, 0, stream, p);
-        else hipLaunchKernelGGL((attn_fwd_kernel<128, false>), grid, block, 0, stream, p);
+        if (p.softclamp) if (pr) hipLaunchKernelGGL((attn_fwd_kernel<128, true, true>), grid, block, 0, stream, p);
+        else hipLaunchKernelGGL((attn_fwd_kernel<128, true, false>), grid, block, 0, stream, p);
+        else if (pr) hipLaunchKernelGGL((attn_fwd_kernel<128, false, true>), grid, block, 0, stream, p);
+        else hipLaunchKernelGGL((attn_fwd_kernel<128, false, false>), grid, block, 0, stream, p);
     } else {
         // unsupported head dim is a host-side error (checked in bindings)
         __builtin_trap();

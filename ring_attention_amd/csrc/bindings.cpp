@@ -5,6 +5,7 @@
 // stream so they compose with RCCL comm streams and hipGraph capture.
 
 #include <torch/extension.h>
+#include <cstdlib>
 #include <ATen/hip/HIPContext.h>
 #include <c10/hip/HIPStream.h>
 
@@ -81,6 +82,19 @@ void attn_fwd(
     p.ablate = (int)ablate;
     p.ticks = nullptr;
     if (ticks.has_value()) p.ticks = (unsigned long long*)ticks->data_ptr();
+    // causal pairing (uniform per-WG work) whenever the diagonal cuts this
+    // kv range and nothing incompatible is on (splits balance differently;
+    // windows are already uniform; ticks index per tile)
+    p.paired = 0;
+    {
+        const long T = (Nq + 255) / 256;      // QROWS_WG
+        // pairing halves grid.x — only engage when the paired grid still
+        // fills the 256 CUs (measured: 8k/8h loses, 16k+ wins +8-24%)
+        if (causal && !has_win && kv_split <= 1 && diag < Nk
+            && ((T + 1) / 2) * B * H >= 256 && !ticks.has_value()
+            && ablate == 0 && !std::getenv("RING_ATTN_NO_PAIR"))
+            p.paired = (int)T;
+    }
 
     launch_attn_fwd(p, (int)D, at::hip::getCurrentHIPStream());
     TORCH_CHECK(hipGetLastError() == hipSuccess, "attn_fwd launch failed");
@@ -154,10 +168,21 @@ void attn_bwd(
     p.accumulate = accumulate;
     p.split = (int)split;
 
-    if (which == 0 || which == 1)
+    // causal pairing: uniform per-WG work when the diagonal cuts this range.
+    // dq pairs over Q tiles, dkv over KV tiles (different counts when the
+    // kv range is gathered), so set per launch.
+    const bool pair_ok = causal && !has_win && split <= 1 && diag < Nk
+                         && !std::getenv("RING_ATTN_NO_PAIR");
+    if (which == 0 || which == 1) {
+        const long Tq = (Nq + 255) / 256;
+        p.paired = (pair_ok && ((Tq + 1) / 2) * B * H >= 256) ? (int)Tq : 0;
         launch_attn_bwd_dq(p, (int)D, at::hip::getCurrentHIPStream());
-    if (which == 0 || which == 2)
+    }
+    if (which == 0 || which == 2) {
+        const long Tk = (Nk + 255) / 256;
+        p.paired = (pair_ok && ((Tk + 1) / 2) * B * HK >= 256) ? (int)Tk : 0;
         launch_attn_bwd_dkv(p, (int)D, at::hip::getCurrentHIPStream());
+    }
     TORCH_CHECK(hipGetLastError() == hipSuccess, "attn_bwd launch failed");
 }
 

@@ -118,17 +118,16 @@ def _hop_geometry(rq: int, rk: int, n: int, ring_size: int, striped: bool,
 
 
 def _causal_balance_split(causal, lookback, diag_cuts, grid_wgs):
-    """Extra grid.z split for causally-imbalanced grids.
+    """Causal load-balance hook (now a no-op).
 
-    With a causal diagonal cutting through the shard, per-WG work varies
-    linearly (the tail WG walks the full kv range), so once the grid exceeds
-    ~2x the 256 CUs a 2-way split shortens the critical path faster than the
-    merge/atomic overhead costs (measured: +19% at 16k/GPU causal, +8% at
-    32k; a loss at 8k where the grid is exactly 1 WG/CU — hence the 512
-    threshold).  Sliding-window shards are near-uniform, so no split.
+    Round-1 history: a 2-way grid.z split here gave +19%/+8% at 16k/32k
+    causal, but it is SUPERSEDED by in-kernel paired-tile scheduling — the
+    C++ binding makes WG x run tiles (x, T-1-x) whenever the paired grid
+    still fills the 256 CUs, which measured strictly better wherever either
+    engages (c16k 184 vs 228 TF, c32k 231 vs 251) and costs no partial
+    buffers or atomics.  RING_ATTN_SPLIT_* env overrides still force
+    grid.z splits; RING_ATTN_NO_PAIR disables pairing.
     """
-    if causal and lookback is None and diag_cuts and grid_wgs >= 512:
-        return 2
     return 1
 
 

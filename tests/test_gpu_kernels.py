@@ -395,19 +395,13 @@ def test_d128_gqa_mask_combo():
         assert e / (rt.abs().max().item() + 1e-6) < 5e-2, f"{name} err {e}"
 
 
-def test_causal_balance_split_consistency():
-    """The auto causal-balance grid.z split (engaged when the causal diagonal
-    cuts the shard and the grid exceeds 512 WGs) must be numerically
-    consistent with the unsplit kernels — checked by self-comparison at a
-    shape large enough to trip the heuristic (16 qtiles x b4 x h8 = 512)."""
+def test_causal_pairing_and_split_consistency():
+    """Causal paired-tile scheduling (in-binding: WG x runs tiles (x, T-1-x)
+    when the paired grid fills the CUs — here 16 qtiles x b4 x h8 -> paired
+    grid 256) and the grid.z split env overrides must all be numerically
+    consistent with the plain unpaired/unsplit kernels."""
     import os
-    from ring_attention_amd.ops.ring_flash_hip import (
-        ring_flash_attn_hip_, _causal_balance_split)
-    assert _causal_balance_split(True, None, True, 512) == 2
-    assert _causal_balance_split(True, None, True, 256) == 1
-    assert _causal_balance_split(False, None, True, 4096) == 1
-    assert _causal_balance_split(True, 128, True, 4096) == 1
-    assert _causal_balance_split(True, None, False, 4096) == 1
+    from ring_attention_amd.ops.ring_flash_hip import ring_flash_attn_hip_
 
     b, n, h, d = 4, 4096, 8, 64
     q, k, v = _mk(b, n, h, h, d)
@@ -415,7 +409,8 @@ def test_causal_balance_split_consistency():
 
     def run(env):
         old = {k_: os.environ.pop(k_, None) for k_ in
-               ("RING_ATTN_KV_SPLIT", "RING_ATTN_SPLIT_DQ", "RING_ATTN_SPLIT_DKV")}
+               ("RING_ATTN_KV_SPLIT", "RING_ATTN_SPLIT_DQ",
+                "RING_ATTN_SPLIT_DKV", "RING_ATTN_NO_PAIR")}
         os.environ.update(env)
         try:
             qg = q.clone().requires_grad_(True)
@@ -430,10 +425,12 @@ def test_causal_balance_split_consistency():
                 if v_ is not None:
                     os.environ[k_] = v_
 
-    split = run({})          # auto: causal-balance split 2 on all kernels
-    unsplit = run({"RING_ATTN_KV_SPLIT": "1", "RING_ATTN_SPLIT_DQ": "1",
-                   "RING_ATTN_SPLIT_DKV": "1"})
-    for s, u, name in zip(split, unsplit, ("out", "dq", "dk", "dv")):
-        e = (s.float() - u.float()).abs().max().item()
-        ref = u.float().abs().max().item() + 1e-6
-        assert e / ref < 1e-2, f"{name} split-vs-unsplit rel err {e/ref}"
+    paired = run({})                            # auto: pairing engages
+    plain = run({"RING_ATTN_NO_PAIR": "1"})     # unpaired, unsplit
+    forced = run({"RING_ATTN_NO_PAIR": "1", "RING_ATTN_KV_SPLIT": "2",
+                  "RING_ATTN_SPLIT_DQ": "2", "RING_ATTN_SPLIT_DKV": "2"})
+    for variant, tag in ((paired, "paired"), (forced, "split2")):
+        for s, u, name in zip(variant, plain, ("out", "dq", "dk", "dv")):
+            e = (s.float() - u.float()).abs().max().item()
+            ref = u.float().abs().max().item() + 1e-6
+            assert e / ref < 1e-2, f"{name} {tag}-vs-plain rel err {e/ref}"
