@@ -174,10 +174,12 @@ void attn_bwd_dq_kernel(BwdParams p) { // cap 256 VGPR, stop the 220 B/thread
         if (t_lo > t_hi) t_lo = t_hi;
     }
     if (p.split > 1) {
-        int per = (num_kv_tiles + p.split - 1) / p.split;
-        t_lo = max(t_lo, (int)(blockIdx.z * per));
-        t_hi = min(t_hi, (int)((blockIdx.z + 1) * per));
-        if (t_lo > t_hi) t_lo = t_hi;
+        // fractional split of this WG's own valid range (see attn_fwd)
+        int valid = t_hi - t_lo;
+        int per = (valid + p.split - 1) / p.split;
+        int base = t_lo;
+        t_lo = base + min(valid, (int)(blockIdx.z * per));
+        t_hi = base + min(valid, (int)((blockIdx.z + 1) * per));
     }
 
     // staging: K + V row chunks, K^T pairs, running pointers
@@ -504,10 +506,14 @@ void attn_bwd_dkv_kernel(BwdParams p) {
                 t1 = (int)min((long)num_q_tiles,
                               i_max_needed < 0 ? 0 : i_max_needed / QT + 1);
         }
-        if (p.split > 1) {                 // grid.z splits the q walk
-            int per = (num_q_tiles + p.split - 1) / p.split;
-            t0 = max(t0, (int)(blockIdx.z * per));
-            t1 = min(t1, (int)((blockIdx.z + 1) * per));
+        if (p.split > 1) {
+            // grid.z takes a fractional share of this WG's own valid q walk
+            // (a global-range split is skewed against the causal trapezoid)
+            int valid = t1 > t0 ? t1 - t0 : 0;
+            int per = (valid + p.split - 1) / p.split;
+            int base = t0;
+            t0 = base + min(valid, (int)(blockIdx.z * per));
+            t1 = base + min(valid, (int)((blockIdx.z + 1) * per));
         }
 
         // ---- T14 pipeline: per-thread staging registers

@@ -176,11 +176,14 @@ __global__ __launch_bounds__(NTHREADS, 1) void attn_fwd_kernel(FwdParams p) {
         t_lo = x <= 0 ? 0 : (int)((x + KVBLK - 1) / KVBLK);
         if (t_lo > t_hi) t_lo = t_hi;
     }
-    if (split_mode) {                     // this z's tile-aligned kv share
-        int per_split = (num_kv_tiles + p.kv_split - 1) / p.kv_split;
-        t_lo = max(t_lo, zsplit * per_split);
-        t_hi = min(t_hi, (zsplit + 1) * per_split);
-        if (t_lo > t_hi) t_lo = t_hi;
+    if (split_mode) {
+        // this z's share of THIS WG's valid tile range (fractional split:
+        // a global-range split is skewed against the causal trapezoid)
+        int valid = t_hi - t_lo;
+        int per_split = (valid + p.kv_split - 1) / p.kv_split;
+        int base = t_lo;
+        t_lo = base + min(valid, zsplit * per_split);
+        t_hi = base + min(valid, (zsplit + 1) * per_split);
     }
 
     // ---- T14 async-stage split: per-thread staging registers

@@ -88,10 +88,12 @@ void attn_fwd(
     p.paired = 0;
     {
         const long T = (Nq + 255) / 256;      // QROWS_WG
-        // pairing halves grid.x — only engage when the paired grid still
-        // fills the 256 CUs (measured: 8k/8h loses, 16k+ wins +8-24%)
-        if (causal && !has_win && kv_split <= 1 && diag < Nk
-            && ((T + 1) / 2) * B * H >= 256 && !ticks.has_value()
+        // pairing halves grid.x — only engage when the paired grid (times
+        // any grid.z split) still fills the 256 CUs (measured: engaging
+        // below that idles half the chip and loses)
+        const long z = kv_split > 1 ? kv_split : 1;
+        if (causal && !has_win && diag < Nk
+            && ((T + 1) / 2) * B * H * z >= 256 && !ticks.has_value()
             && ablate == 0 && !std::getenv("RING_ATTN_NO_PAIR"))
             p.paired = (int)T;
     }
@@ -171,16 +173,17 @@ void attn_bwd(
     // causal pairing: uniform per-WG work when the diagonal cuts this range.
     // dq pairs over Q tiles, dkv over KV tiles (different counts when the
     // kv range is gathered), so set per launch.
-    const bool pair_ok = causal && !has_win && split <= 1 && diag < Nk
+    const bool pair_ok = causal && !has_win && diag < Nk
                          && !std::getenv("RING_ATTN_NO_PAIR");
+    const long z = split > 1 ? split : 1;
     if (which == 0 || which == 1) {
         const long Tq = (Nq + 255) / 256;
-        p.paired = (pair_ok && ((Tq + 1) / 2) * B * H >= 256) ? (int)Tq : 0;
+        p.paired = (pair_ok && ((Tq + 1) / 2) * B * H * z >= 256) ? (int)Tq : 0;
         launch_attn_bwd_dq(p, (int)D, at::hip::getCurrentHIPStream());
     }
     if (which == 0 || which == 2) {
         const long Tk = (Nk + 255) / 256;
-        p.paired = (pair_ok && ((Tk + 1) / 2) * B * HK >= 256) ? (int)Tk : 0;
+        p.paired = (pair_ok && ((Tk + 1) / 2) * B * HK * z >= 256) ? (int)Tk : 0;
         launch_attn_bwd_dkv(p, (int)D, at::hip::getCurrentHIPStream());
     }
     TORCH_CHECK(hipGetLastError() == hipSuccess, "attn_bwd launch failed");

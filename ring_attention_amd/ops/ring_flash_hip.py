@@ -127,7 +127,15 @@ def _causal_balance_split(causal, lookback, diag_cuts, grid_wgs):
     engages (c16k 184 vs 228 TF, c32k 231 vs 251) and costs no partial
     buffers or atomics.  RING_ATTN_SPLIT_* env overrides still force
     grid.z splits; RING_ATTN_NO_PAIR disables pairing.
+
+    One case remains python-side: grids in [256, 512) WGs (e.g. the 8k/GPU
+    causal shape, 32 qtiles x 8 heads) — pairing alone halves the grid below
+    the CU count, so return a 2-way grid.z split; the binding then pairs ON
+    TOP of it (paired grid x split = 256+ busy WGs, uniform work, halved
+    critical path).
     """
+    if causal and lookback is None and diag_cuts and 256 <= grid_wgs < 512:
+        return 2
     return 1
 
 
