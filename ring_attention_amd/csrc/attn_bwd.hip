@@ -129,12 +129,13 @@ void attn_bwd_dq_kernel(BwdParams p) { // cap 256 VGPR, stop the 220 B/thread
     const int hk = h / p.group;
 
     // causal pairing: WG x runs q-tiles (x, T-1-x) — uniform per-WG work
-    for (int pit = 0; pit < (PAIRED ? 2 : 1); ++pit) {
+    const int n_pit = PAIRED
+        ? (p.paired - 1 - (int)blockIdx.x == (int)blockIdx.x ? 1 : 2) : 1;
+    for (int pit = 0; pit < n_pit; ++pit) {
     const int qtile = PAIRED
         ? (pit == 0 ? (int)blockIdx.x : p.paired - 1 - (int)blockIdx.x)
         : (int)blockIdx.x;
     if (PAIRED && pit == 1) {
-        if (qtile == (int)blockIdx.x) break;   // odd T: middle tile once
         __syncthreads();                       // LDS handoff between tiles
     }
 
@@ -448,12 +449,13 @@ void attn_bwd_dkv_kernel(BwdParams p) {
 
     // causal pairing (mirrored): kv-tile x attends T-x q-tiles, so WG x
     // runs kv-tiles (x, T-1-x) for uniform per-WG work
-    for (int pit = 0; pit < (PAIRED ? 2 : 1); ++pit) {
+    const int n_pit = PAIRED
+        ? (p.paired - 1 - (int)blockIdx.x == (int)blockIdx.x ? 1 : 2) : 1;
+    for (int pit = 0; pit < n_pit; ++pit) {
     const int kvtile = PAIRED
         ? (pit == 0 ? (int)blockIdx.x : p.paired - 1 - (int)blockIdx.x)
         : (int)blockIdx.x;
     if (PAIRED && pit == 1) {
-        if (kvtile == (int)blockIdx.x) break;  // odd T: middle tile once
         __syncthreads();                       // LDS handoff between tiles
     }
 

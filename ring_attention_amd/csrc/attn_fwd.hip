@@ -110,12 +110,13 @@ __global__ __launch_bounds__(NTHREADS, 1) void attn_fwd_kernel(FwdParams p) {
 
     // causal pairing: WG x runs q-tiles (x, T-1-x) — per-WG work is the
     // uniform T+1 tiles instead of the 2:1 triangle imbalance
-    for (int pit = 0; pit < (PAIRED ? 2 : 1); ++pit) {
+    const int n_pit = PAIRED
+        ? (p.paired - 1 - (int)blockIdx.x == (int)blockIdx.x ? 1 : 2) : 1;
+    for (int pit = 0; pit < n_pit; ++pit) {
     const int qtile = PAIRED
         ? (pit == 0 ? (int)blockIdx.x : p.paired - 1 - (int)blockIdx.x)
         : (int)blockIdx.x;
     if (PAIRED && pit == 1) {
-        if (qtile == (int)blockIdx.x) break;   // odd T: middle tile once
         __syncthreads();                       // LDS handoff between tiles
     }
 
