@@ -134,7 +134,7 @@ void attn_bwd_dq_kernel(BwdParams p) { // cap 256 VGPR, stop the 220 B/thread
     for (int pit = 0; pit < n_pit; ++pit) {
     const int qtile = PAIRED
         ? (pit == 0 ? (int)blockIdx.x : p.paired - 1 - (int)blockIdx.x)
-        : (int)blockIdx.x;
+        : (p.desc ? p.desc[(long)blockIdx.x * 3] : (int)blockIdx.x);
     if (PAIRED && pit == 1) {
         __syncthreads();                       // LDS handoff between tiles
     }
@@ -181,6 +181,10 @@ void attn_bwd_dq_kernel(BwdParams p) { // cap 256 VGPR, stop the 220 B/thread
         int base = t_lo;
         t_lo = base + min(valid, (int)(blockIdx.z * per));
         t_hi = base + min(valid, (int)((blockIdx.z + 1) * per));
+    }
+    if (!PAIRED && p.desc) {     // descriptor mode: exact unit bounds
+        t_lo = p.desc[(long)blockIdx.x * 3 + 1];
+        t_hi = p.desc[(long)blockIdx.x * 3 + 2];
     }
 
     // staging: K + V row chunks, K^T pairs, running pointers
@@ -400,7 +404,7 @@ void attn_bwd_dq_kernel(BwdParams p) { // cap 256 VGPR, stop the 220 B/thread
         #pragma unroll
         for (int r = 0; r < 16; ++r) {
             int d = db * 32 + (r & 3) + 8 * (r >> 2) + 4 * lhi;
-            if (p.split > 1) atomicAdd(dqp + d, dq_acc[db][r]);
+            if (p.split > 1 || p.desc) atomicAdd(dqp + d, dq_acc[db][r]);
             else if (p.accumulate) dqp[d] += dq_acc[db][r];
             else dqp[d] = dq_acc[db][r];
         }
@@ -454,7 +458,7 @@ void attn_bwd_dkv_kernel(BwdParams p) {
     for (int pit = 0; pit < n_pit; ++pit) {
     const int kvtile = PAIRED
         ? (pit == 0 ? (int)blockIdx.x : p.paired - 1 - (int)blockIdx.x)
-        : (int)blockIdx.x;
+        : (p.desc ? p.desc[(long)blockIdx.x * 3] : (int)blockIdx.x);
     if (PAIRED && pit == 1) {
         __syncthreads();                       // LDS handoff between tiles
     }
@@ -510,12 +514,16 @@ void attn_bwd_dkv_kernel(BwdParams p) {
         }
         if (p.split > 1) {
             // grid.z takes a fractional share of this WG's own valid q walk
-            // (a global-range split is skewed against the causal trapezoid)
+            // (a global-range slice is skewed against the causal trapezoid)
             int valid = t1 > t0 ? t1 - t0 : 0;
             int per = (valid + p.split - 1) / p.split;
             int base = t0;
             t0 = base + min(valid, (int)(blockIdx.z * per));
             t1 = base + min(valid, (int)((blockIdx.z + 1) * per));
+        }
+        if (!PAIRED && p.desc) {   // descriptor mode: exact unit bounds
+            t0 = p.desc[(long)blockIdx.x * 3 + 1];
+            t1 = p.desc[(long)blockIdx.x * 3 + 2];
         }
 
         // ---- T14 pipeline: per-thread staging registers
@@ -769,7 +777,7 @@ void attn_bwd_dkv_kernel(BwdParams p) {
                 int d = db * 32 + l31;
                 if (kvrow <= jmax) {
                     float* dst = dkb + kvrow * D + d;
-                    if (p.split > 1) atomicAdd(dst, dk_acc[db][r]);
+                    if (p.split > 1 || p.desc) atomicAdd(dst, dk_acc[db][r]);
                     else if (p.accumulate) *dst += dk_acc[db][r];
                     else *dst = dk_acc[db][r];
                 }
@@ -781,7 +789,7 @@ void attn_bwd_dkv_kernel(BwdParams p) {
             for (int r = 0; r < 16; ++r) {
                 int d = db * 32 + (r & 3) + 8 * (r >> 2) + 4 * lhi;
                 float* dst = dvb + (long)d * p.nk + j;
-                if (p.split > 1) atomicAdd(dst, dv_acc[db][r]);
+                if (p.split > 1 || p.desc) atomicAdd(dst, dv_acc[db][r]);
                 else if (p.accumulate) *dst += dv_acc[db][r];
                 else *dst = dv_acc[db][r];
             }
@@ -793,7 +801,9 @@ void launch_attn_bwd_dq(const BwdParams& p, int head_dim, hipStream_t stream) {
     dim3 block(512);
     int z = p.split > 1 ? p.split : 1;
     long qt_ = (p.nq + DQ_QROWS_WG - 1) / DQ_QROWS_WG;
-    dim3 grid_dq(p.paired ? (qt_ + 1) / 2 : qt_, p.b * p.h, z);
+    if (p.desc) qt_ = p.n_units;
+    else if (p.paired) qt_ = (qt_ + 1) / 2;
+    dim3 grid_dq(qt_, p.b * p.h, z);
     const bool pr = p.paired > 0;
     if (head_dim == 64) {
         if (p.softclamp) if (pr) hipLaunchKernelGGL((attn_bwd_dq_kernel<64, true, true>), grid_dq, block, 0, stream, p);
@@ -814,7 +824,9 @@ void launch_attn_bwd_dkv(const BwdParams& p, int head_dim, hipStream_t stream) {
     dim3 block(512);
     int z = p.split > 1 ? p.split : 1;
     long kt_ = (p.nk + KVROWS_WG - 1) / KVROWS_WG;
-    dim3 grid_dkv(p.paired ? (kt_ + 1) / 2 : kt_, p.b * p.hk, z);
+    if (p.desc) kt_ = p.n_units;
+    else if (p.paired) kt_ = (kt_ + 1) / 2;
+    dim3 grid_dkv(kt_, p.b * p.hk, z);
     const bool pr = p.paired > 0;
     if (head_dim == 64) {
         if (p.softclamp) if (pr) hipLaunchKernelGGL((attn_bwd_dkv_kernel<64, 64, true, true>), grid_dkv, block, 0, stream, p);

@@ -99,6 +99,11 @@ struct BwdParams {
                             // accumulated with fp32 atomics instead of plain ops
     int paired;             // causal balance: >0 = total walk-parallel tiles T;
                             // grid.x = ceil(T/2), WG x runs tiles (x, T-1-x)
+    const int* desc;        // descriptor scheduling (or null): grid.x units,
+                            // desc[3u] = tile, desc[3u+1] = t_lo,
+                            // desc[3u+2] = t_hi — constant work per unit;
+                            // outputs accumulate with fp32 atomics
+    long n_units;           // rows in desc (grid.x when desc mode)
 };
 
 void launch_attn_bwd_dq(const BwdParams& p, int head_dim, hipStream_t stream);

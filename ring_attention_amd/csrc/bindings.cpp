@@ -142,7 +142,9 @@ void attn_bwd(
     double scale, bool causal, int64_t diag, int64_t q_stride,
     int64_t win, bool has_win,
     bool softclamp, double softclamp_value, bool accumulate, int64_t split,
-    int64_t which) {   // 0 = both, 1 = dq only, 2 = dk/dv only
+    int64_t which,     // 0 = both, 1 = dq only, 2 = dk/dv only
+    std::optional<at::Tensor> desc_dq,    // int32 (U,3): tile, t_lo, t_hi
+    std::optional<at::Tensor> desc_dkv) {
     CHECK_BF16_CONTIG(q); CHECK_BF16_CONTIG(k); CHECK_BF16_CONTIG(v); CHECK_BF16_CONTIG(dout);
     CHECK_F32_CONTIG(lse); CHECK_F32_CONTIG(delta);
     CHECK_F32_CONTIG(dq); CHECK_F32_CONTIG(dk); CHECK_F32_CONTIG(dv);
@@ -176,14 +178,28 @@ void attn_bwd(
     const bool pair_ok = causal && !has_win && diag < Nk
                          && !std::getenv("RING_ATTN_NO_PAIR");
     const long z = split > 1 ? split : 1;
+    auto set_desc = [&](const std::optional<at::Tensor>& dsc) {
+        p.desc = nullptr; p.n_units = 0;
+        if (dsc.has_value()) {
+            TORCH_CHECK(dsc->is_contiguous() && dsc->scalar_type() == at::kInt
+                        && dsc->dim() == 2 && dsc->size(1) == 3,
+                        "desc must be contiguous int32 (U,3)");
+            p.desc = dsc->data_ptr<int>();
+            p.n_units = dsc->size(0);
+        }
+    };
     if (which == 0 || which == 1) {
         const long Tq = (Nq + 255) / 256;
-        p.paired = (pair_ok && ((Tq + 1) / 2) * B * H * z >= 256) ? (int)Tq : 0;
+        set_desc(desc_dq);
+        p.paired = (!p.desc && pair_ok && ((Tq + 1) / 2) * B * H * z >= 256)
+                       ? (int)Tq : 0;
         launch_attn_bwd_dq(p, (int)D, at::hip::getCurrentHIPStream());
     }
     if (which == 0 || which == 2) {
         const long Tk = (Nk + 255) / 256;
-        p.paired = (pair_ok && ((Tk + 1) / 2) * B * HK * z >= 256) ? (int)Tk : 0;
+        set_desc(desc_dkv);
+        p.paired = (!p.desc && pair_ok && ((Tk + 1) / 2) * B * HK * z >= 256)
+                       ? (int)Tk : 0;
         launch_attn_bwd_dkv(p, (int)D, at::hip::getCurrentHIPStream());
     }
     TORCH_CHECK(hipGetLastError() == hipSuccess, "attn_bwd launch failed");

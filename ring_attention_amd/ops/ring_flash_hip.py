@@ -139,6 +139,69 @@ def _causal_balance_split(causal, lookback, diag_cuts, grid_wgs):
     return 1
 
 
+_DESC_CACHE = {}
+
+
+def _walk_descriptors(kind, d, nq, nk, diag, q_stride, bh, device):
+    """Constant-work unit descriptors for a causally-cut tile walk.
+
+    Returns int32 (U, 3) [tile, t_lo, t_hi] — or None when the walk is
+    empty.  `kind` "dq" = q-tiles (256 rows) walking kv tiles of the dq
+    kernel's KVBLK; "dkv" = kv-tiles walking q tiles of QT.  The bounds
+    replicate the kernels' causal tile math exactly (no window — desc mode
+    is only engaged without lookback).  Uniform unit work removes the
+    causal trapezoid imbalance at ANY grid size, with fp32-atomic
+    accumulation instead of pairing/splits.
+    """
+    key = (kind, d, nq, nk, diag, q_stride, bh, str(device))
+    hit = _DESC_CACHE.get(key)
+    if hit is not None:
+        return hit
+    if kind == "dq":
+        T = (nq + 255) // 256
+        W = 128 if d == 64 else 64          # dq_kvblk
+        n_w = (nk + W - 1) // W
+        bounds = []
+        for x in range(T):
+            imax = min((x + 1) * 256, nq) - 1
+            qmax = imax * q_stride + diag
+            hi = 0 if qmax < 0 else min(n_w, qmax // W + 1)
+            bounds.append((x, 0, hi))
+    else:
+        T = (nk + 255) // 256
+        W = 64 if d == 64 else 32           # dkv QT
+        n_w = (nq + W - 1) // W
+        bounds = []
+        for x in range(T):
+            i_min = -(-(x * 256 - diag) // q_stride)   # ceil div
+            lo = 0 if i_min <= 0 else min(n_w, i_min // W)
+            bounds.append((x, lo, n_w))
+    total = sum(hi - lo for _, lo, hi in bounds)
+    if total <= 0:
+        desc = None
+    else:
+        unit = max(1, -(-total // max(32, 768 // max(1, bh))))
+        rows = []
+        for x, lo, hi in bounds:
+            t = lo
+            while t < hi:
+                rows.append((x, t, min(hi, t + unit)))
+                t += unit
+        desc = torch.tensor(rows, dtype=torch.int32, device=device)
+    _DESC_CACHE[key] = desc
+    return desc
+
+
+def _use_desc(causal, lookback, diag, nk, grid_wgs):
+    """Engage descriptor units only where measured to win: a causally-cut
+    walk whose natural grid UNDER-FILLS the chip (GQA dkv: kv-tiles x b x hk
+    < 256 WGs — gqa16k 191 -> 204 TF).  Everywhere else the per-unit atomic
+    output traffic loses to paired-tile scheduling (c16k 230 vs 203,
+    causal8k 146 vs 129 — measured both ways)."""
+    return (causal and lookback is None and diag < nk and grid_wgs < 256
+            and not os.environ.get("RING_ATTN_NO_DESC"))
+
+
 class RingFlashAttentionHIPFunction(Function):
     @staticmethod
     def forward(ctx, q, k, v, mask, causal, bucket_size, ring_reduce, striped,
@@ -320,16 +383,24 @@ class RingFlashAttentionHIPFunction(Function):
                 causal, lookback, True, kvtiles_t * b * hk))
             split_dq = int(os.environ.get("RING_ATTN_SPLIT_DQ", split_dq))
             split_dkv = int(os.environ.get("RING_ATTN_SPLIT_DKV", split_dkv))
-            dq = _alloc_dq(zeroed=split_dq > 1)
-            fac = torch.zeros if split_dkv > 1 else torch.empty
+            ddq = ddkv = None
+            if _use_desc(causal, lookback, diag, n_total, kvtiles_t * b * hk):
+                ddkv = _walk_descriptors("dkv", d, n, n_total, diag, q_stride,
+                                         b * hk, qb.device)
+            if ddkv is not None:
+                split_dkv = 1
+            dq = _alloc_dq(zeroed=split_dq > 1 or ddq is not None)
+            fac = torch.zeros if (split_dkv > 1 or ddkv is not None) else torch.empty
             dk_full = fac(b, hk, n_total, d, device=qb.device, dtype=torch.float32)
             dv_full = fac(b, hk, d, n_total, device=qb.device, dtype=torch.float32)
             ext.attn_bwd(qb, k_full, v_full, dob, m_full, lse, delta,
                          dq, dk_full, dv_full, scale, causal, diag, q_stride,
-                         0, False, softclamp_qk_sim, softclamp_value, False, split_dq, 1)
+                         0, False, softclamp_qk_sim, softclamp_value, False, split_dq, 1,
+                         ddq, None)
             ext.attn_bwd(qb, k_full, v_full, dob, m_full, lse, delta,
                          dq, dk_full, dv_full, scale, causal, diag, q_stride,
-                         0, False, softclamp_qk_sim, softclamp_value, False, split_dkv, 2)
+                         0, False, softclamp_qk_sim, softclamp_value, False, split_dkv, 2,
+                         None, ddkv)
             # ONE reduce-scatter returns each rank's dk/dv shard (summed)
             dk_chunks = _scatter_chunks_of_global(dk_full, R, striped, dim=2)
             dv_chunks = _scatter_chunks_of_global(dv_full, R, striped, dim=3)
@@ -365,24 +436,33 @@ class RingFlashAttentionHIPFunction(Function):
                 split_dq = int(env_dq)
             if env_dkv is not None:
                 split_dkv = int(env_dkv)
-            plan.append((skip, diag, win, split_dq, split_dkv))
+            ddq = ddkv = None
+            if not skip and _use_desc(causal, lookback, diag, n,
+                                      qtiles * b * hk):
+                ddkv = _walk_descriptors("dkv", d, n, n, diag, 1, b * hk,
+                                         qb.device)
+                if ddkv is not None:
+                    split_dkv = 1
+            plan.append((skip, diag, win, split_dq, split_dkv, ddq, ddkv))
         first_active = next(i for i, p_ in enumerate(plan) if not p_[0])
-        dq_zeroed = plan[first_active][3] > 1   # atomics need a zeroed base
+        # atomics (grid.z split or descriptor units) need a zeroed base
+        dq_zeroed = plan[first_active][3] > 1 or plan[first_active][5] is not None
         dq = _alloc_dq(zeroed=dq_zeroed)
 
         ring_tensors = (kb, vb) if mask_u8 is None else (kb, vb, mask_u8)
         acc = RingAccumulator(topo)
 
         for info, tensors in all_ring_pass(topo, *ring_tensors, max_hops=hops):
-            skip, diag, win, split_dq, split_dkv = plan[info.hop]
+            skip, diag, win, split_dq, split_dkv, ddq, ddkv = plan[info.hop]
             k_t, v_t = tensors[0], tensors[1]
             mk = tensors[2] if mask_u8 is not None else None
             # circulate dk/dv flat in the kernel's native scratch layouts
             # (dk (b,hk,n,d), dv (b,hk,d,n)); elementwise accumulation is
             # layout-agnostic, so only ONE final permute happens at home.
             # The kernels overwrite every element at split 1, so only the
-            # atomic (split>1) and skip (circulated as-is) cases zero-init.
-            fac = torch.zeros if (skip or split_dkv > 1) else torch.empty
+            # atomic (split/descriptor) and skip (circulated) cases zero-init.
+            fac = (torch.zeros if (skip or split_dkv > 1 or ddkv is not None)
+                   else torch.empty)
             contrib = fac(2, b * hk * n * d, device=qb.device,
                           dtype=torch.float32)
             if not skip:
@@ -395,12 +475,12 @@ class RingFlashAttentionHIPFunction(Function):
                              dq, dk_n, dv_n,
                              scale, causal, diag, 1, win, lookback is not None,
                              softclamp_qk_sim, softclamp_value, dq_acc_flag,
-                             split_dq, 1)
+                             split_dq, 1, ddq, None)
                 ext.attn_bwd(qb, k_t, v_t, dob, mk, lse, delta,
                              dq, dk_n, dv_n,
                              scale, causal, diag, 1, win, lookback is not None,
                              softclamp_qk_sim, softclamp_value, False,
-                             split_dkv, 2)
+                             split_dkv, 2, None, ddkv)
             acc.step(contrib, info.is_last)
 
         dkv = acc.finish(hops)
@@ -470,12 +550,17 @@ class FlashAttnOffsetFunction(Function):
         ext = hip_ext.require()
         dob = do.to(torch.bfloat16).contiguous()
         delta = ext.attn_delta(dob, out)
-        # accumulate=False + split=1: both kernels overwrite every element
+        ddq = ddkv = None
+        kvt_ = (nk + 255) // 256
+        if _use_desc(causal, None, q_offset, nk, kvt_ * b * hk):
+            ddkv = _walk_descriptors("dkv", d, n, nk, q_offset, 1, b * hk, qb.device)
+        fac = torch.zeros if ddkv is not None else torch.empty
         dq = torch.empty(b, n, h, d, device=qb.device, dtype=torch.float32)
-        dk_n = torch.empty(b, hk, nk, d, device=qb.device, dtype=torch.float32)
-        dv_n = torch.empty(b, hk, d, nk, device=qb.device, dtype=torch.float32)
+        dk_n = fac(b, hk, nk, d, device=qb.device, dtype=torch.float32)
+        dv_n = fac(b, hk, d, nk, device=qb.device, dtype=torch.float32)
         ext.attn_bwd(qb, kb, vb, dob, None, lse, delta, dq, dk_n, dv_n,
-                     scale, causal, q_offset, 1, 0, False, False, 50.0, False, 1, 0)
+                     scale, causal, q_offset, 1, 0, False, False, 50.0, False, 1, 0,
+                     ddq, ddkv)
         dk = dk_n.permute(0, 2, 1, 3).contiguous()
         dv = dv_n.permute(0, 3, 1, 2).contiguous()
         return (dq.to(in_dtype), dk.to(in_dtype), dv.to(in_dtype), None, None)

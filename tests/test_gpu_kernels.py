@@ -273,7 +273,8 @@ def test_strided_q_positions_kernel():
     dk_n = torch.zeros(b, h, N, d, device="cuda", dtype=torch.float32)
     dv_n = torch.zeros(b, h, d, N, device="cuda", dtype=torch.float32)
     ext.attn_bwd(q, k, v, do, None, lse, delta, dq, dk_n, dv_n,
-                 scale, True, rq, R, 0, False, False, 50.0, False, 1, 0)
+                 scale, True, rq, R, 0, False, False, 50.0, False, 1, 0,
+                 None, None)
 
     qc = q.float().cpu().requires_grad_(True)
     kc = k.float().cpu().requires_grad_(True)
@@ -410,7 +411,8 @@ def test_causal_pairing_and_split_consistency():
     def run(env):
         old = {k_: os.environ.pop(k_, None) for k_ in
                ("RING_ATTN_KV_SPLIT", "RING_ATTN_SPLIT_DQ",
-                "RING_ATTN_SPLIT_DKV", "RING_ATTN_NO_PAIR")}
+                "RING_ATTN_SPLIT_DKV", "RING_ATTN_NO_PAIR",
+                "RING_ATTN_NO_DESC")}
         os.environ.update(env)
         try:
             qg = q.clone().requires_grad_(True)
@@ -425,11 +427,13 @@ def test_causal_pairing_and_split_consistency():
                 if v_ is not None:
                     os.environ[k_] = v_
 
-    paired = run({})                            # auto: pairing engages
-    plain = run({"RING_ATTN_NO_PAIR": "1"})     # unpaired, unsplit
-    forced = run({"RING_ATTN_NO_PAIR": "1", "RING_ATTN_KV_SPLIT": "2",
+    desc = run({})                              # auto: descriptor units (bwd)
+    paired = run({"RING_ATTN_NO_DESC": "1"})    # paired-tile scheduling
+    plain = run({"RING_ATTN_NO_DESC": "1", "RING_ATTN_NO_PAIR": "1"})
+    forced = run({"RING_ATTN_NO_DESC": "1", "RING_ATTN_NO_PAIR": "1",
+                  "RING_ATTN_KV_SPLIT": "2",
                   "RING_ATTN_SPLIT_DQ": "2", "RING_ATTN_SPLIT_DKV": "2"})
-    for variant, tag in ((paired, "paired"), (forced, "split2")):
+    for variant, tag in ((desc, "desc"), (paired, "paired"), (forced, "split2")):
         for s, u, name in zip(variant, plain, ("out", "dq", "dk", "dv")):
             e = (s.float() - u.float()).abs().max().item()
             ref = u.float().abs().max().item() + 1e-6
