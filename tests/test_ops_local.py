@@ -112,3 +112,49 @@ def test_plan_ring_shard():
     # long sequence forces bigger shards
     padded, shard, chunks = plan_ring_shard(1000, 16, 8, 4)
     assert padded >= 1000 and 4 % chunks == 0 and shard % 8 == 0
+
+
+def test_walk_descriptor_coverage():
+    """Descriptor units must tile each walk exactly: per tile, the union of
+    [t_lo, t_hi) ranges equals the kernel's valid range, disjoint and in
+    order; unit work is bounded by the chunk size."""
+    from ring_attention_amd.ops.ring_flash_hip import _walk_descriptors, _DESC_CACHE
+    for kind in ("dq", "dkv"):
+        for d in (64, 128):
+            for (nq, nk, diag, stride) in [(4096, 4096, 0, 1), (1000, 1000, 0, 1),
+                                           (2048, 4096, 2048, 1), (512, 2048, 1, 4),
+                                           (4096, 4096, -1, 1), (256, 8192, 8191, 1)]:
+                desc = _walk_descriptors(kind, d, nq, nk, diag, stride, 8, "cpu")
+                if kind == "dq":
+                    T = (nq + 255) // 256
+                    W = 128 if d == 64 else 64
+                    n_w = (nk + W - 1) // W
+                    expect = {}
+                    for x in range(T):
+                        qmax = (min((x + 1) * 256, nq) - 1) * stride + diag
+                        hi = 0 if qmax < 0 else min(n_w, qmax // W + 1)
+                        if hi > 0:
+                            expect[x] = (0, hi)
+                else:
+                    T = (nk + 255) // 256
+                    W = 64 if d == 64 else 32
+                    n_w = (nq + W - 1) // W
+                    expect = {}
+                    for x in range(T):
+                        i_min = -(-(x * 256 - diag) // stride)
+                        lo = 0 if i_min <= 0 else min(n_w, i_min // W)
+                        if n_w > lo:
+                            expect[x] = (lo, n_w)
+                if desc is None:
+                    assert not expect, (kind, d, nq, nk, diag)
+                    continue
+                got = {}
+                for tile, lo, hi in desc.tolist():
+                    assert lo < hi
+                    if tile in got:
+                        assert got[tile][1] == lo, "units out of order / gap"
+                        got[tile] = (got[tile][0], hi)
+                    else:
+                        got[tile] = (lo, hi)
+                assert got == expect, (kind, d, nq, nk, diag, stride)
+    _DESC_CACHE.clear()
