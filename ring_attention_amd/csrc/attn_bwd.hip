@@ -430,6 +430,9 @@ struct DkvLds {
     __align__(16) float delta[2][QT];
 };
 
+template <class F>
+__device__ __attribute__((noinline)) void dkv_noinline_call(F&& f) { f(); }
+
 template <int D, int QT, bool SOFTCLAMP, bool PAIRED>
 __global__ __launch_bounds__(512, 2)   // see dq kernel note (552 B spills)
 void attn_bwd_dkv_kernel(BwdParams p) {
@@ -463,14 +466,17 @@ void attn_bwd_dkv_kernel(BwdParams p) {
         __syncthreads();                       // LDS handoff between tiles
     }
 
+    auto dkv_body = [&]() {
+    const BwdParams P = p;   // register-local copy: the noinline frame would
+                             // otherwise re-read fields through scratch
     const long j0_wg = (long)kvtile * KVROWS_WG;
-    const long jmax = min(j0_wg + KVROWS_WG, p.nk) - 1;
+    const long jmax = min(j0_wg + KVROWS_WG, P.nk) - 1;
     const long j = j0_wg + wid * KVROWS_WAVE + l31;
     const bool col_valid = j <= jmax;
     const long jc = col_valid ? j : j0_wg;
 
-    const __bf16* kb = (const __bf16*)p.k + ((long)b * p.nk + jc) * p.hk * D + (long)hkh * D;
-    const __bf16* vb = (const __bf16*)p.v + ((long)b * p.nk + jc) * p.hk * D + (long)hkh * D;
+    const __bf16* kb = (const __bf16*)P.k + ((long)b * P.nk + jc) * P.hk * D + (long)hkh * D;
+    const __bf16* vb = (const __bf16*)P.v + ((long)b * P.nk + jc) * P.hk * D + (long)hkh * D;
     bf16x8 kf[KSTEPS], vf[KSTEPS];
     #pragma unroll
     for (int ks = 0; ks < KSTEPS; ++ks) {
@@ -478,52 +484,52 @@ void attn_bwd_dkv_kernel(BwdParams p) {
         vf[ks] = *(const bf16x8*)(vb + ks * 16 + lhi * 8);
     }
     unsigned char kmask_own = 1;
-    if (p.kmask) kmask_own = col_valid ? ((const unsigned char*)p.kmask)[(long)b * p.nk + j] : 0;
+    if (P.kmask) kmask_own = col_valid ? ((const unsigned char*)P.kmask)[(long)b * P.nk + j] : 0;
 
     f32x16 dv_acc[DBLK], dk_acc[DBLK];
     #pragma unroll
     for (int db = 0; db < DBLK; ++db) { dv_acc[db] = f32x16{}; dk_acc[db] = f32x16{}; }
 
-    const int num_q_tiles = (int)((p.nq + QT - 1) / QT);
+    const int num_q_tiles = (int)((P.nq + QT - 1) / QT);
 
-    for (int g = 0; g < p.group; ++g) {
+    for (int g = 0; g < P.group; ++g) {
         // group boundary barrier: the previous group's last tile is still
         // being read from LDS by slower waves when this group's prologue
         // write_qtile targets the same buffer (GQA-only race; widest at
         // split>1 where a chunk is a single tile)
         if (g > 0) __syncthreads();
-        const int h = hkh * p.group + g;
-        const float* lse_row = p.lse + ((long)b * p.h + h) * p.nq;
-        const float* delta_row = p.delta + ((long)b * p.h + h) * p.nq;
-        const __bf16* qg = (const __bf16*)p.q + ((long)b * p.nq) * p.h * D + (long)h * D;
-        const __bf16* dog = (const __bf16*)p.dout + ((long)b * p.nq) * p.h * D + (long)h * D;
+        const int h = hkh * P.group + g;
+        const float* lse_row = P.lse + ((long)b * P.h + h) * P.nq;
+        const float* delta_row = P.delta + ((long)b * P.h + h) * P.nq;
+        const __bf16* qg = (const __bf16*)P.q + ((long)b * P.nq) * P.h * D + (long)h * D;
+        const __bf16* dog = (const __bf16*)P.dout + ((long)b * P.nq) * P.h * D + (long)h * D;
 
         int t0 = 0, t1 = num_q_tiles;
-        if (p.causal) {
+        if (P.causal) {
             // need qpos(i) >= j0_wg  =>  i >= (j0_wg - diag) / q_stride
-            long i_min_needed = (j0_wg - p.diag + p.q_stride - 1) / p.q_stride;
+            long i_min_needed = (j0_wg - P.diag + P.q_stride - 1) / P.q_stride;
             if (i_min_needed > 0) t0 = (int)(i_min_needed / QT);
         }
-        if (p.has_win) {
+        if (P.has_win) {
             // need qpos(i) <= jmax + win  =>  i <= (jmax + win - diag) / q_stride
-            long num = jmax + p.win - p.diag;
-            long i_max_needed = num < 0 ? -1 : num / p.q_stride;
+            long num = jmax + P.win - P.diag;
+            long i_max_needed = num < 0 ? -1 : num / P.q_stride;
             if (i_max_needed < (long)num_q_tiles * QT)
                 t1 = (int)min((long)num_q_tiles,
                               i_max_needed < 0 ? 0 : i_max_needed / QT + 1);
         }
-        if (p.split > 1) {
+        if (P.split > 1) {
             // grid.z takes a fractional share of this WG's own valid q walk
             // (a global-range slice is skewed against the causal trapezoid)
             int valid = t1 > t0 ? t1 - t0 : 0;
-            int per = (valid + p.split - 1) / p.split;
+            int per = (valid + P.split - 1) / P.split;
             int base = t0;
             t0 = base + min(valid, (int)(blockIdx.z * per));
             t1 = base + min(valid, (int)((blockIdx.z + 1) * per));
         }
-        if (!PAIRED && p.desc) {   // descriptor mode: exact unit bounds
-            t0 = p.desc[(long)blockIdx.x * 3 + 1];
-            t1 = p.desc[(long)blockIdx.x * 3 + 2];
+        if (!PAIRED && P.desc) {   // descriptor mode: exact unit bounds
+            t0 = P.desc[(long)blockIdx.x * 3 + 1];
+            t1 = P.desc[(long)blockIdx.x * 3 + 2];
         }
 
         // ---- T14 pipeline: per-thread staging registers
@@ -539,7 +545,7 @@ void attn_bwd_dkv_kernel(BwdParams p) {
 
         auto load_qtile = [&]() {
             const long i0 = t_next * QT;
-            const long imax_ = min(i0 + QT, p.nq) - 1;
+            const long imax_ = min(i0 + QT, P.nq) - 1;
             const bool full = imax_ - i0 == QT - 1;
             #pragma unroll
             for (int r = 0; r < QREGS; ++r) {
@@ -547,8 +553,8 @@ void attn_bwd_dkv_kernel(BwdParams p) {
                 if (c < QCHUNKS) {
                     long gr = i0 + c / CH;
                     int ch = c % CH;
-                    const __bf16* src = qg + gr * p.h * D + ch * 8;
-                    const __bf16* srd = dog + gr * p.h * D + ch * 8;
+                    const __bf16* src = qg + gr * P.h * D + ch * 8;
+                    const __bf16* srd = dog + gr * P.h * D + ch * 8;
                     bool okr = full || gr <= imax_;
                     qst[r] = okr ? *(const uint4*)src : uint4{0, 0, 0, 0};
                     dost[r] = okr ? *(const uint4*)srd : uint4{0, 0, 0, 0};
@@ -563,12 +569,12 @@ void attn_bwd_dkv_kernel(BwdParams p) {
                     long ra = i0 + jp * 2;
                     bool oka = full || ra <= imax_;
                     bool okb = full || ra + 1 <= imax_;
-                    const __bf16* qa_ = qg + ra * p.h * D + d0;
-                    const __bf16* da_ = dog + ra * p.h * D + d0;
+                    const __bf16* qa_ = qg + ra * P.h * D + d0;
+                    const __bf16* da_ = dog + ra * P.h * D + d0;
                     qtv_a[r] = oka ? *(const bf16x8*)qa_ : bf16x8{};
-                    qtv_b[r] = okb ? *(const bf16x8*)(qa_ + p.h * D) : bf16x8{};
+                    qtv_b[r] = okb ? *(const bf16x8*)(qa_ + P.h * D) : bf16x8{};
                     dov_a[r] = oka ? *(const bf16x8*)da_ : bf16x8{};
-                    dov_b[r] = okb ? *(const bf16x8*)(da_ + p.h * D) : bf16x8{};
+                    dov_b[r] = okb ? *(const bf16x8*)(da_ + P.h * D) : bf16x8{};
                 }
             }
             if (tid < QT) {
@@ -622,14 +628,14 @@ void attn_bwd_dkv_kernel(BwdParams p) {
         for (int t = t0; t < t1; ++t) {
             const int par = t & 1;
             const long i0 = (long)t * QT;
-            const long imax = min(i0 + QT, p.nq) - 1;
-            const long q_lo = i0 * p.q_stride + p.diag;
-            const long q_hi = imax * p.q_stride + p.diag;
+            const long imax = min(i0 + QT, P.nq) - 1;
+            const long q_lo = i0 * P.q_stride + P.diag;
+            const long q_hi = imax * P.q_stride + P.diag;
             const bool full_tile =
                 (imax - i0 == QT - 1) &&
-                (!p.causal || q_lo >= jmax) &&
-                (!p.has_win || ((q_hi - j0_wg) <= p.win)) &&
-                !p.kmask;
+                (!P.causal || q_lo >= jmax) &&
+                (!P.has_win || ((q_hi - j0_wg) <= P.win)) &&
+                !P.kmask;
 
             __syncthreads();
 
@@ -668,17 +674,17 @@ void attn_bwd_dkv_kernel(BwdParams p) {
                             int r = 2 * x2 + e;
                             float x, dtanh = 1.f;
                             if constexpr (SOFTCLAMP) {
-                                float inv_v = __builtin_amdgcn_rcpf(p.softclamp_value);
-                                float th = bfast_tanhf(s2[r] * p.scale * inv_v);
-                                x = p.softclamp_value * th * 1.4426950408889634f;
+                                float inv_v = __builtin_amdgcn_rcpf(P.softclamp_value);
+                                float th = bfast_tanhf(s2[r] * P.scale * inv_v);
+                                x = P.softclamp_value * th * 1.4426950408889634f;
                                 dtanh = 1.f - th * th;
                             } else {
-                                x = __builtin_fmaf(s2[r], p.scale * 1.4426950408889634f,
+                                x = __builtin_fmaf(s2[r], P.scale * 1.4426950408889634f,
                                                    -lse4[r & 3] * 1.4426950408889634f);  // fold
                             }
                             float pv = __builtin_amdgcn_exp2f(SOFTCLAMP ? x - lse4[r & 3] * 1.4426950408889634f : x);
                             pe[e] = pv;
-                            dse[e] = pv * (dp[r] - delta4[r & 3]) * dtanh * p.scale;
+                            dse[e] = pv * (dp[r] - delta4[r & 3]) * dtanh * P.scale;
                         }
                         union { __hip_bfloat162 h2; uint32_t u; } c1, c2;
                         c1.h2 = __float22bfloat162_rn(float2{pe[0], pe[1]});
@@ -705,21 +711,21 @@ void attn_bwd_dkv_kernel(BwdParams p) {
                             long i = i0 + qloc;
                             float x, dtanh = 1.f;
                             if constexpr (SOFTCLAMP) {
-                                float inv_v = __builtin_amdgcn_rcpf(p.softclamp_value);
-                                float th = bfast_tanhf(s2[r] * p.scale * inv_v);
-                                x = p.softclamp_value * th * 1.4426950408889634f;
+                                float inv_v = __builtin_amdgcn_rcpf(P.softclamp_value);
+                                float th = bfast_tanhf(s2[r] * P.scale * inv_v);
+                                x = P.softclamp_value * th * 1.4426950408889634f;
                                 dtanh = 1.f - th * th;
                             } else {
-                                x = s2[r] * (p.scale * 1.4426950408889634f);
+                                x = s2[r] * (P.scale * 1.4426950408889634f);
                             }
                             bool ok = col_valid && i <= imax;
-                            long qpos = i * p.q_stride + p.diag;
-                            if (p.causal) ok = ok && (j <= qpos);
-                            if (p.has_win) ok = ok && (qpos - j <= p.win);
-                            if (p.kmask) ok = ok && kmask_own;
+                            long qpos = i * P.q_stride + P.diag;
+                            if (P.causal) ok = ok && (j <= qpos);
+                            if (P.has_win) ok = ok && (qpos - j <= P.win);
+                            if (P.kmask) ok = ok && kmask_own;
                             float pv = ok ? __builtin_amdgcn_exp2f(x - lse4[r & 3] * 1.4426950408889634f) : 0.f;
                             pe[e] = pv;
-                            dse[e] = pv * (dp[r] - delta4[r & 3]) * dtanh * p.scale;
+                            dse[e] = pv * (dp[r] - delta4[r & 3]) * dtanh * P.scale;
                         }
                         union { __hip_bfloat162 h2; uint32_t u; } c1, c2;
                         c1.h2 = __float22bfloat162_rn(float2{pe[0], pe[1]});
@@ -768,7 +774,7 @@ void attn_bwd_dkv_kernel(BwdParams p) {
 
     // write dk (B,HK,Nk,D) and dv^T (B,HK,D,Nk)
     if (col_valid) {
-        float* dkb = p.dk + (((long)b * p.hk + hkh) * p.nk) * D;
+        float* dkb = P.dk + (((long)b * P.hk + hkh) * P.nk) * D;
         #pragma unroll
         for (int db = 0; db < DBLK; ++db)
             #pragma unroll
@@ -777,23 +783,30 @@ void attn_bwd_dkv_kernel(BwdParams p) {
                 int d = db * 32 + l31;
                 if (kvrow <= jmax) {
                     float* dst = dkb + kvrow * D + d;
-                    if (p.split > 1 || p.desc) atomicAdd(dst, dk_acc[db][r]);
-                    else if (p.accumulate) *dst += dk_acc[db][r];
+                    if (P.split > 1 || P.desc) atomicAdd(dst, dk_acc[db][r]);
+                    else if (P.accumulate) *dst += dk_acc[db][r];
                     else *dst = dk_acc[db][r];
                 }
             }
-        float* dvb = p.dv + (((long)b * p.hk + hkh) * D) * p.nk;
+        float* dvb = P.dv + (((long)b * P.hk + hkh) * D) * P.nk;
         #pragma unroll
         for (int db = 0; db < DBLK; ++db)
             #pragma unroll
             for (int r = 0; r < 16; ++r) {
                 int d = db * 32 + (r & 3) + 8 * (r >> 2) + 4 * lhi;
-                float* dst = dvb + (long)d * p.nk + j;
-                if (p.split > 1 || p.desc) atomicAdd(dst, dv_acc[db][r]);
-                else if (p.accumulate) *dst += dv_acc[db][r];
+                float* dst = dvb + (long)d * P.nk + j;
+                if (P.split > 1 || P.desc) atomicAdd(dst, dv_acc[db][r]);
+                else if (P.accumulate) *dst += dv_acc[db][r];
                 else *dst = dv_acc[db][r];
             }
     }
+    };
+    // the noinline frame confines register allocation to the body — at d64
+    // it cuts the hot-loop spills roughly in half for BOTH instantiations
+    // (measured headline +4%, causal +17-24%); d128's plain path is better
+    // inlined (direct call), so only d64 and the paired form take the call
+    if constexpr (D == 64 || PAIRED) dkv_noinline_call(dkv_body);
+    else dkv_body();
     }  // pair loop
 }
 
