@@ -847,10 +847,10 @@ void launch_attn_bwd_dkv(const BwdParams& p, int head_dim, hipStream_t stream) {
         else if (pr) hipLaunchKernelGGL((attn_bwd_dkv_kernel<64, 64, false, true>), grid_dkv, block, 0, stream, p);
         else hipLaunchKernelGGL((attn_bwd_dkv_kernel<64, 64, false, false>), grid_dkv, block, 0, stream, p);
     } else if (head_dim == 128) {
-        if (p.softclamp) if (pr) hipLaunchKernelGGL((attn_bwd_dkv_kernel<128, 32, true, true>), grid_dkv, block, 0, stream, p);
-        else hipLaunchKernelGGL((attn_bwd_dkv_kernel<128, 32, true, false>), grid_dkv, block, 0, stream, p);
-        else if (pr) hipLaunchKernelGGL((attn_bwd_dkv_kernel<128, 32, false, true>), grid_dkv, block, 0, stream, p);
-        else hipLaunchKernelGGL((attn_bwd_dkv_kernel<128, 32, false, false>), grid_dkv, block, 0, stream, p);
+        if (p.softclamp) if (pr) hipLaunchKernelGGL((attn_bwd_dkv_kernel<128, 64, true, true>), grid_dkv, block, 0, stream, p);
+        else hipLaunchKernelGGL((attn_bwd_dkv_kernel<128, 64, true, false>), grid_dkv, block, 0, stream, p);
+        else if (pr) hipLaunchKernelGGL((attn_bwd_dkv_kernel<128, 64, false, true>), grid_dkv, block, 0, stream, p);
+        else hipLaunchKernelGGL((attn_bwd_dkv_kernel<128, 64, false, false>), grid_dkv, block, 0, stream, p);
     } else {
         __builtin_trap();
     }

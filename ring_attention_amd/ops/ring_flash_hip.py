@@ -169,7 +169,7 @@ def _walk_descriptors(kind, d, nq, nk, diag, q_stride, bh, device):
             bounds.append((x, 0, hi))
     else:
         T = (nk + 255) // 256
-        W = 64 if d == 64 else 32           # dkv QT
+        W = 64                              # dkv QT (both head dims)
         n_w = (nq + W - 1) // W
         bounds = []
         for x in range(T):
