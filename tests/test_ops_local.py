@@ -137,7 +137,7 @@ def test_walk_descriptor_coverage():
                             expect[x] = (0, hi)
                 else:
                     T = (nk + 255) // 256
-                    W = 64 if d == 64 else 32
+                    W = 64                      # dkv QT (both head dims)
                     n_w = (nq + W - 1) // W
                     expect = {}
                     for x in range(T):
