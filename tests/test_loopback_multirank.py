@@ -110,6 +110,8 @@ def test_loopback_subring_oracle_cpu():
     ("ring", False, False, 4, None),
     ("ring", True, True, 4, None),
     ("ring", True, False, 4, 1024),          # lookback truncates the walk
+    ("ring", True, True, 4, 1024),           # striped + lookback window math
+    ("allgather", True, True, 2, None),      # striped + GQA gather order
 ])
 def test_loopback_hip_multirank(strategy, causal, striped, hk, lookback):
     from ring_attention_amd.ops.ring_flash_hip import ring_flash_attn_hip_
