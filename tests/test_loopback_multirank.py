@@ -311,3 +311,25 @@ def test_loopback_zigzag_gpu():
         want = ref[:, :, q_pos]
         e = (out.float() - want.float()).abs().max().item()
         assert e / scale < 2e-2, f"rank {rank} zigzag rel err {e/scale}"
+
+
+def test_loopback_tree_decode_cpu():
+    """Tree-decode's two-round collective merge (MAX lse + one packed SUM)
+    on the CPU eager partial — validates the cross-rank math without a GPU."""
+    from ring_attention_amd.tree_decode import tree_attn_decode
+    world = 4
+    b, h, d, n = 2, 3, 16, 256
+    torch.manual_seed(29)
+    q = torch.randn(b, h, 1, d)
+    k = torch.randn(b, h, n, d)
+    v = torch.randn(b, h, n, d)
+    ref = tree_attn_decode(q, k, v, shard_kv_seq=False)
+
+    def run(rank):
+        ks = k.chunk(world, dim=-2)[rank]
+        vs = v.chunk(world, dim=-2)[rank]
+        return tree_attn_decode(q, ks, vs, shard_kv_seq=False)
+
+    for rank, out in enumerate(loopback_world(world, run)):
+        e = (out - ref).abs().max().item()
+        assert e < 1e-5, f"rank {rank} err {e}"
