@@ -19,6 +19,15 @@ overlapped P2P shard circulation; each hop is one resumable kernel launch:
   RingAccumulator in the kernels' transposed scratch layouts
   (dk (B,HK,Nk,D), dv (B,HK,D,Nk)); one final permute at home.
 
+Causal load balance (both strategies): the C++ binding engages paired-tile
+scheduling (WG x runs tiles (x, T-1-x) — uniform work) whenever the halved
+grid still fills the 256 CUs, combined with FRACTIONAL grid.z splits (each
+chunk takes a share of its WG's own valid range) for 256-512-WG grids;
+under-filled GQA dkv grids instead use host-built constant-work descriptor
+units (_walk_descriptors) with fp32-atomic accumulation.  Env overrides:
+RING_ATTN_NO_PAIR, RING_ATTN_NO_DESC, RING_ATTN_KV_SPLIT,
+RING_ATTN_SPLIT_DQ/DKV, RING_ATTN_FORCE_STRATEGY, RING_ATTN_AG_BUDGET.
+
 Capability parity with the reference's ring_flash_attn_cuda
 (/root/reference/ring_attention_pytorch/ring_flash_attention_cuda.py:40-371)
 with its §2.5 bugs fixed.
