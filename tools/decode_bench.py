@@ -3,7 +3,9 @@
 import sys, time, torch
 sys.path.insert(0, "/root/repo")
 from ring_attention_amd.tree_decode import tree_attn_decode
-b, h, n, d = 1, 8, 131072, 64
+import sys
+b, h, d = 1, 8, 64
+n = int(sys.argv[1]) if len(sys.argv) > 1 else 131072
 torch.manual_seed(0)
 q = torch.randn(b, h, 1, d, device="cuda", dtype=torch.bfloat16)
 k = torch.randn(b, h, n, d, device="cuda", dtype=torch.bfloat16)
