@@ -17,9 +17,12 @@ from __future__ import annotations
 
 import argparse
 import os
+import sys
 import time
 
 import torch
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 from ring_attention_amd import tree_attn_decode
 from ring_attention_amd.parallel import get_rank, get_world_size, is_distributed
@@ -46,10 +49,14 @@ def main():
     b, h, d = args.batch, args.heads, args.d_head
     n_local = args.kv_len // world
 
-    # this rank's shard of the KV cache (in service: filled by prefill)
-    torch.manual_seed(1234 + rank)
-    k_cache = torch.randn(b, h, n_local, d, device=device, dtype=dtype)
-    v_cache = torch.randn(b, h, n_local, d, device=device, dtype=dtype)
+    # this rank's shard of a COMMON synthetic KV cache (in service: filled
+    # by prefill) — outputs are identical for any world size
+    torch.manual_seed(1234)
+    k_full = torch.randn(b, h, args.kv_len, d, device=device, dtype=dtype)
+    v_full = torch.randn(b, h, args.kv_len, d, device=device, dtype=dtype)
+    k_cache = k_full[:, :, rank * n_local:(rank + 1) * n_local].contiguous()
+    v_cache = v_full[:, :, rank * n_local:(rank + 1) * n_local].contiguous()
+    del k_full, v_full
 
     torch.manual_seed(7)
     q = torch.randn(b, h, 1, d, device=device, dtype=dtype)
