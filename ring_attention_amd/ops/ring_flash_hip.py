@@ -204,9 +204,10 @@ def _walk_descriptors(kind, d, nq, nk, diag, q_stride, bh, device):
 def _use_desc(causal, lookback, diag, nk, grid_wgs):
     """Engage descriptor units only where measured to win: a causally-cut
     walk whose natural grid UNDER-FILLS the chip (GQA dkv: kv-tiles x b x hk
-    < 256 WGs — gqa16k 191 -> 204 TF).  Everywhere else the per-unit atomic
-    output traffic loses to paired-tile scheduling (c16k 230 vs 203,
-    causal8k 146 vs 129 — measured both ways)."""
+    < 256 WGs — GQA causal 16k 185 -> 216 TF with dkv-only descriptors).
+    Everywhere else the per-unit atomic output traffic loses to paired-tile
+    scheduling (c16k 230 vs 203, causal8k 146 vs 129 when applied to all
+    kernels — measured both ways)."""
     return (causal and lookback is None and diag < nk and grid_wgs < 256
             and not os.environ.get("RING_ATTN_NO_DESC"))
 
