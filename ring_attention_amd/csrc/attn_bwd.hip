@@ -126,7 +126,7 @@ void attn_bwd_dq_kernel(BwdParams p) { // cap 256 VGPR, stop the 220 B/thread
     const int bh = blockIdx.y;
     const int b = bh / p.h;
     const int h = bh % p.h;
-    const int hk = h / p.group;
+    const int hk = h % p.hk;   // reference tile GQA pairing
 
     // causal pairing: WG x runs q-tiles (x, T-1-x) — uniform per-WG work
     const int n_pit = PAIRED
@@ -498,7 +498,7 @@ void attn_bwd_dkv_kernel(BwdParams p) {
         // write_qtile targets the same buffer (GQA-only race; widest at
         // split>1 where a chunk is a single tile)
         if (g > 0) __syncthreads();
-        const int h = hkh * P.group + g;
+        const int h = hkh + g * P.hk;   // reference tile GQA: group g's q head
         const float* lse_row = P.lse + ((long)b * P.h + h) * P.nq;
         const float* delta_row = P.delta + ((long)b * P.h + h) * P.nq;
         const __bf16* qg = (const __bf16*)P.q + ((long)b * P.nq) * P.h * D + (long)h * D;

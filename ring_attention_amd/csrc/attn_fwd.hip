@@ -106,7 +106,7 @@ __global__ __launch_bounds__(NTHREADS, 1) void attn_fwd_kernel(FwdParams p) {
     const int bh = blockIdx.y;
     const int b = bh / p.h;
     const int h = bh % p.h;
-    const int hk = h / p.group;       // kv head
+    const int hk = h % p.hk;          // kv head (reference tile GQA: qh pairs qh % hk)
 
     // causal pairing: WG x runs q-tiles (x, T-1-x) — per-WG work is the
     // uniform T+1 tiles instead of the 2:1 triangle imbalance

@@ -45,8 +45,10 @@ def default_attention(
 
     q, k, v = q.float(), k.float(), v.float()
     if groups > 1:
-        k = k.repeat_interleave(groups, dim=2)
-        v = v.repeat_interleave(groups, dim=2)
+        # reference GQA convention ('... h d -> ... (g h) d', ring_attention.py:86-89):
+        # q head qh pairs kv head qh % hk (tile, not repeat_interleave)
+        k = k.repeat(1, 1, groups, 1)
+        v = v.repeat(1, 1, groups, 1)
 
     scale = d ** -0.5
     sim = torch.einsum("bihd,bjhd->bhij", q, k) * scale

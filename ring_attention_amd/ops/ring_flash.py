@@ -165,8 +165,8 @@ class RingFlashAttentionFunction(Function):
             mask_t = tensors[1].bool() if mask is not None else None
             k_t, v_t = kv_t[0], kv_t[1]
             if groups > 1:
-                k_t = k_t.repeat_interleave(groups, dim=2)
-                v_t = v_t.repeat_interleave(groups, dim=2)
+                k_t = k_t.repeat(1, 1, groups, 1)   # tile: qh pairs qh % hk
+                v_t = v_t.repeat(1, 1, groups, 1)
 
             rk = info.source_ring_rank
             rq = topo.ring_rank
@@ -235,8 +235,8 @@ class RingFlashAttentionFunction(Function):
             mask_t = tensors[1].bool() if mask is not None else None
             k_src, v_src = kv_t[0], kv_t[1]                      # (b, n, hk, d)
             if groups > 1:
-                k_t = k_src.repeat_interleave(groups, dim=2)
-                v_t = v_src.repeat_interleave(groups, dim=2)
+                k_t = k_src.repeat(1, 1, groups, 1)   # tile: qh pairs qh % hk
+                v_t = v_src.repeat(1, 1, groups, 1)
             else:
                 k_t, v_t = k_src, v_src
 
@@ -276,8 +276,8 @@ class RingFlashAttentionFunction(Function):
                     dq[:, sli] += torch.einsum("bhij,bjhd->bihd", ds, kj)
                     dk_part = torch.einsum("bhij,bihd->bjhd", ds, qi)
                     if groups > 1:
-                        dv_part = dv_part.view(b, bucket_size, hk, groups, d).sum(dim=3)
-                        dk_part = dk_part.view(b, bucket_size, hk, groups, d).sum(dim=3)
+                        dv_part = dv_part.view(b, bucket_size, groups, hk, d).sum(dim=2)
+                        dk_part = dk_part.view(b, bucket_size, groups, hk, d).sum(dim=2)
                     dv_c[:, slj] += dv_part
                     dk_c[:, slj] += dk_part
 
@@ -358,8 +358,8 @@ def _eager_with_lse(q, k, v, mask, causal, qp, kp, lookback, sc, scv):
     groups = h // hk
     kf, vf = k.float(), v.float()
     if groups > 1:
-        kf = kf.repeat_interleave(groups, dim=2)
-        vf = vf.repeat_interleave(groups, dim=2)
+        kf = kf.repeat(1, 1, groups, 1)   # tile: qh pairs qh % hk
+        vf = vf.repeat(1, 1, groups, 1)
     sim = torch.einsum("bihd,bjhd->bhij", q.float(), kf) * d ** -0.5
     if sc:
         sim = softclamp(sim, scv)

@@ -14,8 +14,39 @@ import functools
 
 import torch.distributed as dist
 
-# cache: (world_size, ring_size) -> list of per-ring process groups
-_RING_GROUPS: dict[tuple[int, int], list] = {}
+# cache: (default-group id, world_size, ring_size) -> list of per-ring process
+# groups.  Keyed on the default group's identity so a destroy_process_group +
+# re-init with the same sizes does not return stale dead group handles.
+_RING_GROUPS: dict[tuple[int, int, int], list] = {}
+
+
+def _default_group_id() -> int:
+    if not dist.is_initialized():
+        return 0
+    return id(dist.distributed_c10d._get_default_group())
+
+
+def init_ring_groups(ring_size: int) -> None:
+    """Eagerly create the sub-ring process groups for ``ring_size``.
+
+    ``dist.new_group`` is itself a collective: EVERY rank must create EVERY
+    ring's group in the same order relative to all other collectives.  Call
+    this once right after ``init_process_group`` (before any forward) when
+    using ``ring_size < world_size`` — ``RingTopology`` also triggers it at
+    construction (the start of the first forward), but an explicit startup
+    call is the robust pattern for complex training loops that may issue
+    other collectives concurrently.
+    """
+    world = get_world_size()
+    if ring_size >= world or not dist.is_initialized():
+        return
+    key = (_default_group_id(), world, ring_size)
+    if key not in _RING_GROUPS:
+        groups = []
+        for ring in range(world // ring_size):
+            ranks = list(range(ring * ring_size, (ring + 1) * ring_size))
+            groups.append(dist.new_group(ranks))
+        _RING_GROUPS[key] = groups
 
 
 def is_distributed() -> bool:
@@ -63,6 +94,14 @@ class RingTopology:
         self.ring_index = rank // ring_size          # which sub-ring this rank belongs to
         self.ring_rank = rank % ring_size            # position within the sub-ring
         self.ring_base = self.ring_index * ring_size  # global rank of ring position 0
+        # eager sub-ring group creation: new_group is collective, so creating
+        # at first topology construction (start of forward, before this
+        # layer's other collectives) keeps the creation order deterministic
+        # across ranks; init_ring_groups() at startup is the belt-and-braces
+        # pattern (only when sizes describe THIS process's world)
+        if (world_size is None and rank is None and ring_size < world
+                and dist.is_initialized()):
+            init_ring_groups(ring_size)
 
     def global_rank_of(self, ring_rank: int) -> int:
         return self.ring_base + (ring_rank % self.ring_size)
@@ -80,18 +119,14 @@ class RingTopology:
     def process_group(self):
         """The process group of THIS rank's ring (None = default/world group).
 
-        Sub-ring groups are created lazily but COLLECTIVELY (every rank
-        creates every ring's group in the same order, as torch requires) and
-        cached for the process lifetime."""
+        Groups are created eagerly at topology construction (or by an explicit
+        ``init_ring_groups`` call at startup) and cached per default-group
+        identity, so re-initialized process groups never see stale handles."""
         if self.ring_size == self.world_size or not dist.is_initialized():
             return None
-        key = (self.world_size, self.ring_size)
+        key = (_default_group_id(), self.world_size, self.ring_size)
         if key not in _RING_GROUPS:
-            groups = []
-            for ring in range(self.world_size // self.ring_size):
-                ranks = list(range(ring * self.ring_size, (ring + 1) * self.ring_size))
-                groups.append(dist.new_group(ranks))
-            _RING_GROUPS[key] = groups
+            init_ring_groups(self.ring_size)
         return _RING_GROUPS[key][self.ring_index]
 
     def source_of_hop(self, hop: int) -> int:

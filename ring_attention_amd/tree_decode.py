@@ -58,6 +58,11 @@ def tree_attn_decode(
     dtype = q.dtype
     b, h, one, d = q.shape
 
+    # capture the value dim from the FULL v before chunking: a rank whose
+    # shard is empty must still pack a (b,h,1,1+dv)-shaped all-reduce buffer,
+    # and dv may differ from q's head dim d
+    dim_v = v.shape[-1] if v is not None else d
+
     if shard_kv_seq:
         assert k is not None
         rank, world = get_rank(), get_world_size()
@@ -66,11 +71,9 @@ def tree_attn_decode(
         k, v = (ks[rank], vs[rank]) if rank < len(ks) else (None, None)
 
     if v is not None:
-        dim_v = v.shape[-1]
         local_out, lse = _local_decode_partial(q, k, v)
     else:
         # seq shorter than world: this rank holds nothing
-        dim_v = d
         local_out = q.new_zeros((b, h, one, dim_v), dtype=torch.float32)
         lse = torch.full((b, h, one, 1), -torch.finfo(torch.float32).max,
                          device=q.device, dtype=torch.float32)

@@ -138,30 +138,42 @@ def zig_zag_attn(
         if kv_valid_len is not None:
             k_ = k_[:, :kv_valid_len]
             v_ = v_[:, :kv_valid_len]
-        head_perm = None
-        if groups > 1:
-            # reference GQA convention tiles kv heads ('b h -> b (g h)': q head
-            # qh pairs with kv head qh % hk); the kernel groups contiguously
-            # (qh // g) — permute q heads kv-major for the kernel, then back
-            # kernel position p -> kv head p // g; slot j of kv head c must
-            # hold original head c + j * hk (the reference's qh % hk pairing)
-            p = torch.arange(heads, device=q.device)
-            head_perm = (p // groups) + (p % groups) * kv_heads
-            inv_perm = torch.empty_like(head_perm)
-            inv_perm[head_perm] = torch.arange(heads, device=q.device)
-            q_ = q_.index_select(2, head_perm)
+        # GQA: the HIP kernels pair q head qh with kv head qh % hk natively
+        # (the reference tile convention, ring_attention.py:86-89) — no head
+        # permutation needed
         outs = []
         for c, start in enumerate(q_chunk_starts):
             qc = q_[:, c * half:(c + 1) * half]
             outs.append(flash_attn_offset(qc, k_, v_, q_offset=start, causal=True))
         out = torch.cat(outs, dim=1)
-        if head_perm is not None:
-            out = out.index_select(2, inv_perm)
         return out.permute(0, 2, 1, 3)
     if groups > 1:
         # repeat pattern parity with the reference: 'b h n d -> b (g h) n d'
         k = k.repeat(1, groups, 1, 1)
         v = v.repeat(1, groups, 1, 1)
+
+    if causal and attn_mask is None:
+        # portable path must honor causal too (same semantics as the fast
+        # path), not silently compute full attention: derive the mask from the
+        # chunk-start positions.  KV after the gather is RANK-major
+        # [r0:(0, 2W-1), r1:(1, 2W-2), ...] — build its global positions.
+        if q_chunk_starts is None:
+            raise ValueError(
+                "zig_zag_attn(causal=True) on the portable path needs "
+                "q_chunk_starts (from zig_zag_shard's query_positions) or an "
+                "explicit attn_mask encoding causality")
+        n, N = q.shape[-2], k.shape[-2]
+        half = n // 2
+        world = get_world_size()
+        ch = N // (2 * world)
+        r = torch.arange(world, device=q.device)
+        kpos = torch.stack((r * ch, (2 * world - 1 - r) * ch), dim=1).reshape(-1)
+        kpos = (kpos[:, None] + torch.arange(ch, device=q.device)[None, :]).reshape(-1)
+        qpos = torch.cat([q_chunk_starts[0] + torch.arange(half, device=q.device),
+                          q_chunk_starts[1] + torch.arange(n - half, device=q.device)])
+        attn_mask = kpos[None, :] <= qpos[:, None]
+        if kv_valid_len is not None:
+            attn_mask = attn_mask & (kpos[None, :] < kv_valid_len)
 
     dtype = q.dtype
     scale = q.shape[-1] ** -0.5

@@ -67,8 +67,8 @@ def _ring_case(rank, world, causal, striped, groups, mask_on, bucket_size, lookb
         from ring_attention_amd.ops.reference import MASK_VALUE
         import torch.nn.functional as F
         scale = d ** -0.5
-        kk = k2.repeat_interleave(groups, dim=2) if groups > 1 else k2
-        vv = v2.repeat_interleave(groups, dim=2) if groups > 1 else v2
+        kk = k2.repeat(1, 1, groups, 1) if groups > 1 else k2   # reference tile GQA
+        vv = v2.repeat(1, 1, groups, 1) if groups > 1 else v2
         sim = torch.einsum("bihd,bjhd->bhij", q2.float(), kk.float()) * scale
         if mask is not None:
             sim = sim.masked_fill(~mask[:, None, None, :], MASK_VALUE)
