@@ -23,7 +23,12 @@ _RING_GROUPS: dict[tuple[int, int, int], list] = {}
 def _default_group_id() -> int:
     if not dist.is_initialized():
         return 0
-    return id(dist.distributed_c10d._get_default_group())
+    try:
+        return id(dist.distributed_c10d._get_default_group())
+    except Exception:
+        # fake/monkey-patched dist (tests) may report initialized without a
+        # real default group — treat as one stable epoch
+        return 0
 
 
 def init_ring_groups(ring_size: int) -> None:
