@@ -73,6 +73,10 @@ void launch_attn_fwd_merge(const FwdMergeParams& p, int head_dim, hipStream_t st
 
 void launch_attn_fwd(const FwdParams& p, int head_dim, hipStream_t stream);
 
+// v2: one-wave-per-SIMD pipelined forward (attn_fwd_v2.hip); returns false
+// when the config is not covered (caller falls back to v1)
+bool launch_attn_fwd_v2(const FwdParams& p, int head_dim, hipStream_t stream);
+
 struct BwdParams {
     const void* q;          // bf16 (B, Nq, H, D)
     const void* k;          // bf16 (B, Nk, HK, D)

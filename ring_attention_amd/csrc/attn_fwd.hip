@@ -32,6 +32,7 @@
 #include <hip/hip_runtime.h>
 #include <hip/hip_bf16.h>
 #include <cstdint>
+#include <cstdlib>
 
 #include "attn_common.h"
 
@@ -608,6 +609,13 @@ void launch_attn_fwd_merge(const FwdMergeParams& p, int head_dim, hipStream_t st
 }
 
 void launch_attn_fwd(const FwdParams& p, int head_dim, hipStream_t stream) {
+    // v2 (one-wave-per-SIMD pipeline) opt-in via RING_ATTN_FWD_V2=1;
+    // RING_ATTN_FWD_V2=0 forces v1.  Default currently v1 (flip after the
+    // promotion criterion: v2 beats v1 with full-flag oracle parity).
+    static const char* v2env = std::getenv("RING_ATTN_FWD_V2");
+    if (v2env && v2env[0] == '1') {
+        if (launch_attn_fwd_v2(p, head_dim, stream)) return;
+    }
     long qtiles = (p.nq + QROWS_WG - 1) / QROWS_WG;
     dim3 grid(p.paired ? (qtiles + 1) / 2 : qtiles, p.b * p.h,
               p.kv_split > 1 ? p.kv_split : 1);

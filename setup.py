@@ -22,6 +22,7 @@ ext = CUDAExtension(
     sources=[
         os.path.join(CSRC, "bindings.cpp"),
         os.path.join(CSRC, "attn_fwd.hip"),
+        os.path.join(CSRC, "attn_fwd_v2.hip"),
         os.path.join(CSRC, "attn_bwd.hip"),
         os.path.join(CSRC, "decode.hip"),
         os.path.join(CSRC, "rotary.hip"),
