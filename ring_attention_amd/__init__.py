@@ -30,6 +30,8 @@ __all__ = [
     "ring_flash_attn_",
     "ring_flash_attn_hip",
     "ring_flash_attn_hip_",
+    "flash_attn",
+    "flash_attn_offset",
     "RingAttention",
     "RingTransformer",
     "RingRotaryEmbedding",
@@ -45,7 +47,8 @@ __all__ = [
 
 def __getattr__(name):
     # HIP-backed function imports lazily (requires the built extension on GPU)
-    if name in ("ring_flash_attn_hip", "ring_flash_attn_hip_"):
+    if name in ("ring_flash_attn_hip", "ring_flash_attn_hip_", "flash_attn",
+                "flash_attn_offset"):
         from .ops import ring_flash_hip
         return getattr(ring_flash_hip, name)
     raise AttributeError(f"module {__name__!r} has no attribute {name!r}")

@@ -39,7 +39,10 @@
 
 namespace ring_attn {
 
-__device__ __forceinline__ int bswz(int row, int chunk) { return chunk ^ (row & 7); }
+template <int CH>
+__device__ __forceinline__ int bswz(int row, int chunk) {
+    return chunk ^ (row & (CH < 8 ? CH - 1 : 7));   // see swz in attn_fwd.hip
+}
 
 __device__ __forceinline__ float bfast_tanhf(float x) {
     float e = __builtin_amdgcn_exp2f(x * 2.885390081777927f);
@@ -61,7 +64,7 @@ __device__ __forceinline__ void stage_rowmajor(
         long gr = row0 + row;
         uint4 val = (gr <= rowmax) ? *(const uint4*)(gbase + gr * row_stride + ch * 8)
                                    : uint4{0, 0, 0, 0};
-        *(uint4*)(lds_rm + row * D + bswz(row, ch) * 8) = val;
+        *(uint4*)(lds_rm + row * D + bswz<D / 8>(row, ch) * 8) = val;
     }
 }
 
@@ -106,7 +109,7 @@ struct DqLds {
     unsigned char kmask[2][KVB];
 };
 
-template <int D, bool SOFTCLAMP, bool PAIRED>
+template <int D, bool SOFTCLAMP, bool PAIRED, bool BIAS = false>
 __global__ __launch_bounds__(512, 2)   // 8-wave WGs need exactly 2 waves/SIMD:
 void attn_bwd_dq_kernel(BwdParams p) { // cap 256 VGPR, stop the 220 B/thread
                                        // scratch spills seen at the 128 cap
@@ -247,8 +250,8 @@ void attn_bwd_dq_kernel(BwdParams p) { // cap 256 VGPR, stop the 220 B/thread
             int c = tid + r * 512;
             if (c < KCHUNKS) {
                 int row = c / CH, ch = c % CH;
-                *(uint4*)(lds.k[par] + row * D + bswz(row, ch) * 8) = kst[r];
-                *(uint4*)(lds.v[par] + row * D + bswz(row, ch) * 8) = vst[r];
+                *(uint4*)(lds.k[par] + row * D + bswz<D / 8>(row, ch) * 8) = kst[r];
+                *(uint4*)(lds.v[par] + row * D + bswz<D / 8>(row, ch) * 8) = vst[r];
             }
         }
         #pragma unroll
@@ -280,7 +283,7 @@ void attn_bwd_dq_kernel(BwdParams p) { // cap 256 VGPR, stop the 220 B/thread
         const int par = t & 1;
         const long j0 = (long)t * DQ_KVBLK;
         const long jmax = min(j0 + DQ_KVBLK, p.nk) - 1;
-        const bool full_tile =
+        const bool full_tile = !BIAS &&
             (jmax - j0 == DQ_KVBLK - 1) &&
             (!p.causal || jmax <= wg_q_min) &&
             (!p.has_win || (wg_q_max - j0) <= p.win) &&
@@ -300,8 +303,8 @@ void attn_bwd_dq_kernel(BwdParams p) { // cap 256 VGPR, stop the 220 B/thread
             #pragma unroll
             for (int ks = 0; ks < KSTEPS; ++ks) {
                 int chunk = ks * 2 + lhi;
-                bf16x8 kfr = *(const bf16x8*)(lds.k[par] + krow * D + bswz(krow, chunk) * 8);
-                bf16x8 vfr = *(const bf16x8*)(lds.v[par] + krow * D + bswz(krow, chunk) * 8);
+                bf16x8 kfr = *(const bf16x8*)(lds.k[par] + krow * D + bswz<D / 8>(krow, chunk) * 8);
+                bf16x8 vfr = *(const bf16x8*)(lds.v[par] + krow * D + bswz<D / 8>(krow, chunk) * 8);
                 s = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kfr, qf[ks], s, 0, 0, 0);
                 dp = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vfr, dof[ks], dp, 0, 0, 0);
             }
@@ -346,6 +349,14 @@ void attn_bwd_dq_kernel(BwdParams p) { // cap 256 VGPR, stop the 220 B/thread
                             x = s[r] * scale2;
                         }
                         bool ok = row_valid && jj <= jmax;
+                        if constexpr (BIAS) {
+                            if (ok) {
+                                const long bi = p.bias_mat
+                                    ? (((long)b * p.h + h) * p.nq + ic) * p.nk + jj
+                                    : ((long)b * p.h + h) * p.nk + jj;
+                                x += p.bias[bi] * 1.4426950408889634f;
+                            }
+                        }
                         if (p.causal) ok = ok && (jj <= qpos_i);
                         if (p.has_win) ok = ok && (qpos_i - jj <= p.win);
                         if (p.kmask) ok = ok && lds.kmask[par][jj - j0];
@@ -387,7 +398,7 @@ void attn_bwd_dq_kernel(BwdParams p) { // cap 256 VGPR, stop the 220 B/thread
             for (int ks = 0; ks < DQ_NBLK * 2; ++ks) {
                 int chunk = ks * 2 + lhi;
                 bf16x8 ktf = *(const bf16x8*)(lds.kt[par] + drow * DQ_KVBLK +
-                                              bswz(drow, chunk) * 8);
+                                              bswz<DQ_KVBLK / 8>(drow, chunk) * 8);
                 dq_acc[db] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
                     ktf, *(const bf16x8*)frag[ks], dq_acc[db], 0, 0, 0);
             }
@@ -433,7 +444,7 @@ struct DkvLds {
 template <class F>
 __device__ __attribute__((noinline)) void dkv_noinline_call(F&& f) { f(); }
 
-template <int D, int QT, bool SOFTCLAMP, bool PAIRED>
+template <int D, int QT, bool SOFTCLAMP, bool PAIRED, bool BIAS = false>
 __global__ __launch_bounds__(512, 2)   // see dq kernel note (552 B spills)
 void attn_bwd_dkv_kernel(BwdParams p) {
     static_assert(D % 32 == 0 && QT % 32 == 0);
@@ -592,8 +603,8 @@ void attn_bwd_dkv_kernel(BwdParams p) {
                 int c = tid + r * 512;
                 if (c < QCHUNKS) {
                     int row = c / CH, ch = c % CH;
-                    *(uint4*)(lds.q[par] + row * D + bswz(row, ch) * 8) = qst[r];
-                    *(uint4*)(lds.do_[par] + row * D + bswz(row, ch) * 8) = dost[r];
+                    *(uint4*)(lds.q[par] + row * D + bswz<D / 8>(row, ch) * 8) = qst[r];
+                    *(uint4*)(lds.do_[par] + row * D + bswz<D / 8>(row, ch) * 8) = dost[r];
                 }
             }
             #pragma unroll
@@ -631,7 +642,7 @@ void attn_bwd_dkv_kernel(BwdParams p) {
             const long imax = min(i0 + QT, P.nq) - 1;
             const long q_lo = i0 * P.q_stride + P.diag;
             const long q_hi = imax * P.q_stride + P.diag;
-            const bool full_tile =
+            const bool full_tile = !BIAS &&
                 (imax - i0 == QT - 1) &&
                 (!P.causal || q_lo >= jmax) &&
                 (!P.has_win || ((q_hi - j0_wg) <= P.win)) &&
@@ -650,8 +661,8 @@ void attn_bwd_dkv_kernel(BwdParams p) {
                 for (int ks = 0; ks < KSTEPS; ++ks) {
                     int qrow = qb * 32 + l31;
                     int chunk = ks * 2 + lhi;
-                    bf16x8 qa = *(const bf16x8*)(lds.q[par] + qrow * D + bswz(qrow, chunk) * 8);
-                    bf16x8 da = *(const bf16x8*)(lds.do_[par] + qrow * D + bswz(qrow, chunk) * 8);
+                    bf16x8 qa = *(const bf16x8*)(lds.q[par] + qrow * D + bswz<D / 8>(qrow, chunk) * 8);
+                    bf16x8 da = *(const bf16x8*)(lds.do_[par] + qrow * D + bswz<D / 8>(qrow, chunk) * 8);
                     s2 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qa, kf[ks], s2, 0, 0, 0);
                     dp = __builtin_amdgcn_mfma_f32_32x32x16_bf16(da, vf[ks], dp, 0, 0, 0);
                 }
@@ -719,6 +730,14 @@ void attn_bwd_dkv_kernel(BwdParams p) {
                                 x = s2[r] * (P.scale * 1.4426950408889634f);
                             }
                             bool ok = col_valid && i <= imax;
+                            if constexpr (BIAS) {
+                                if (ok) {
+                                    const long bi = P.bias_mat
+                                        ? (((long)b * P.h + h) * P.nq + i) * P.nk + j
+                                        : ((long)b * P.h + h) * P.nk + j;
+                                    x += P.bias[bi] * 1.4426950408889634f;
+                                }
+                            }
                             long qpos = i * P.q_stride + P.diag;
                             if (P.causal) ok = ok && (j <= qpos);
                             if (P.has_win) ok = ok && (qpos - j <= P.win);
@@ -818,6 +837,19 @@ void launch_attn_bwd_dq(const BwdParams& p, int head_dim, hipStream_t stream) {
     else if (p.paired) qt_ = (qt_ + 1) / 2;
     dim3 grid_dq(qt_, p.b * p.h, z);
     const bool pr = p.paired > 0;
+    if (p.bias) {   // bias runs unpaired (bindings force paired=0)
+        if (head_dim == 64) {
+            if (p.softclamp) hipLaunchKernelGGL((attn_bwd_dq_kernel<64, true, false, true>), grid_dq, block, 0, stream, p);
+            else hipLaunchKernelGGL((attn_bwd_dq_kernel<64, false, false, true>), grid_dq, block, 0, stream, p);
+        } else if (head_dim == 128) {
+            if (p.softclamp) hipLaunchKernelGGL((attn_bwd_dq_kernel<128, true, false, true>), grid_dq, block, 0, stream, p);
+            else hipLaunchKernelGGL((attn_bwd_dq_kernel<128, false, false, true>), grid_dq, block, 0, stream, p);
+        } else {
+            if (p.softclamp) hipLaunchKernelGGL((attn_bwd_dq_kernel<32, true, false, true>), grid_dq, block, 0, stream, p);
+            else hipLaunchKernelGGL((attn_bwd_dq_kernel<32, false, false, true>), grid_dq, block, 0, stream, p);
+        }
+        return;
+    }
     if (head_dim == 64) {
         if (p.softclamp) if (pr) hipLaunchKernelGGL((attn_bwd_dq_kernel<64, true, true>), grid_dq, block, 0, stream, p);
         else hipLaunchKernelGGL((attn_bwd_dq_kernel<64, true, false>), grid_dq, block, 0, stream, p);
@@ -828,6 +860,11 @@ void launch_attn_bwd_dq(const BwdParams& p, int head_dim, hipStream_t stream) {
         else hipLaunchKernelGGL((attn_bwd_dq_kernel<128, true, false>), grid_dq, block, 0, stream, p);
         else if (pr) hipLaunchKernelGGL((attn_bwd_dq_kernel<128, false, true>), grid_dq, block, 0, stream, p);
         else hipLaunchKernelGGL((attn_bwd_dq_kernel<128, false, false>), grid_dq, block, 0, stream, p);
+    } else if (head_dim == 32) {
+        if (p.softclamp) if (pr) hipLaunchKernelGGL((attn_bwd_dq_kernel<32, true, true>), grid_dq, block, 0, stream, p);
+        else hipLaunchKernelGGL((attn_bwd_dq_kernel<32, true, false>), grid_dq, block, 0, stream, p);
+        else if (pr) hipLaunchKernelGGL((attn_bwd_dq_kernel<32, false, true>), grid_dq, block, 0, stream, p);
+        else hipLaunchKernelGGL((attn_bwd_dq_kernel<32, false, false>), grid_dq, block, 0, stream, p);
     } else {
         __builtin_trap();
     }
@@ -841,6 +878,19 @@ void launch_attn_bwd_dkv(const BwdParams& p, int head_dim, hipStream_t stream) {
     else if (p.paired) kt_ = (kt_ + 1) / 2;
     dim3 grid_dkv(kt_, p.b * p.hk, z);
     const bool pr = p.paired > 0;
+    if (p.bias) {   // bias runs unpaired (bindings force paired=0)
+        if (head_dim == 64) {
+            if (p.softclamp) hipLaunchKernelGGL((attn_bwd_dkv_kernel<64, 64, true, false, true>), grid_dkv, block, 0, stream, p);
+            else hipLaunchKernelGGL((attn_bwd_dkv_kernel<64, 64, false, false, true>), grid_dkv, block, 0, stream, p);
+        } else if (head_dim == 128) {
+            if (p.softclamp) hipLaunchKernelGGL((attn_bwd_dkv_kernel<128, 64, true, false, true>), grid_dkv, block, 0, stream, p);
+            else hipLaunchKernelGGL((attn_bwd_dkv_kernel<128, 64, false, false, true>), grid_dkv, block, 0, stream, p);
+        } else {
+            if (p.softclamp) hipLaunchKernelGGL((attn_bwd_dkv_kernel<32, 64, true, false, true>), grid_dkv, block, 0, stream, p);
+            else hipLaunchKernelGGL((attn_bwd_dkv_kernel<32, 64, false, false, true>), grid_dkv, block, 0, stream, p);
+        }
+        return;
+    }
     if (head_dim == 64) {
         if (p.softclamp) if (pr) hipLaunchKernelGGL((attn_bwd_dkv_kernel<64, 64, true, true>), grid_dkv, block, 0, stream, p);
         else hipLaunchKernelGGL((attn_bwd_dkv_kernel<64, 64, true, false>), grid_dkv, block, 0, stream, p);
@@ -851,6 +901,11 @@ void launch_attn_bwd_dkv(const BwdParams& p, int head_dim, hipStream_t stream) {
         else hipLaunchKernelGGL((attn_bwd_dkv_kernel<128, 64, true, false>), grid_dkv, block, 0, stream, p);
         else if (pr) hipLaunchKernelGGL((attn_bwd_dkv_kernel<128, 64, false, true>), grid_dkv, block, 0, stream, p);
         else hipLaunchKernelGGL((attn_bwd_dkv_kernel<128, 64, false, false>), grid_dkv, block, 0, stream, p);
+    } else if (head_dim == 32) {
+        if (p.softclamp) if (pr) hipLaunchKernelGGL((attn_bwd_dkv_kernel<32, 64, true, true>), grid_dkv, block, 0, stream, p);
+        else hipLaunchKernelGGL((attn_bwd_dkv_kernel<32, 64, true, false>), grid_dkv, block, 0, stream, p);
+        else if (pr) hipLaunchKernelGGL((attn_bwd_dkv_kernel<32, 64, false, true>), grid_dkv, block, 0, stream, p);
+        else hipLaunchKernelGGL((attn_bwd_dkv_kernel<32, 64, false, false>), grid_dkv, block, 0, stream, p);
     } else {
         __builtin_trap();
     }
@@ -896,6 +951,9 @@ void launch_attn_delta(const DeltaParams& p, int head_dim, hipStream_t stream) {
     } else if (head_dim == 128) {
         long waves = (p.rows + 3) / 4;
         hipLaunchKernelGGL(attn_delta_kernel<128>, dim3((waves + 3) / 4), block, 0, stream, p);
+    } else if (head_dim == 32) {
+        long waves = (p.rows + 15) / 16;
+        hipLaunchKernelGGL(attn_delta_kernel<32>, dim3((waves + 3) / 4), block, 0, stream, p);
     } else {
         __builtin_trap();
     }

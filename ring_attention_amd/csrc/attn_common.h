@@ -51,6 +51,10 @@ struct FwdParams {
                             // (x, T-1-x) so per-WG causal work is uniform
     unsigned long long* ticks;  // diagnostics: per-tile segment s_memtime
                                 // stamps from block(0,0,0) wave 0 (or null)
+    const float* bias;      // additive attention bias (natural-log domain)
+                            // or null; (B,H,Nk) vector / (B,H,Nq,Nk) matrix
+                            // (reference triton_flash_attn.py:1047-1063)
+    int bias_mat;           // 1 = matrix (per-row) bias
 };
 
 struct FwdMergeParams {
@@ -103,6 +107,8 @@ struct BwdParams {
                             // accumulated with fp32 atomics instead of plain ops
     int paired;             // causal balance: >0 = total walk-parallel tiles T;
                             // grid.x = ceil(T/2), WG x runs tiles (x, T-1-x)
+    const float* bias;      // additive bias (see FwdParams) or null
+    int bias_mat;
     const int* desc;        // descriptor scheduling (or null): grid.x units,
                             // desc[3u] = tile, desc[3u+1] = t_lo,
                             // desc[3u+2] = t_hi — constant work per unit;

@@ -51,7 +51,12 @@ static constexpr int NTHREADS = WAVES * 64;
 // XOR swizzle of a 16-byte chunk index within a row (row stride D*2 bytes):
 // chunk' = chunk ^ (row & 7).  Applied identically on the staging write and
 // the fragment read, so the LDS image is consistent (both-sides rule).
-__device__ __forceinline__ int swz(int row, int chunk) { return chunk ^ (row & 7); }
+template <int CH>
+__device__ __forceinline__ int swz(int row, int chunk) {
+    // swizzle mask must stay inside the row: CH chunks per row (CH-1 when
+    // CH <= 8; wider rows keep the measured 8-slot spread)
+    return chunk ^ (row & (CH < 8 ? CH - 1 : 7));
+}
 
 // fast tanh from builtins: tanh(x) = 1 - 2/(exp2(2x*log2e) + 1); avoids the
 // libm tanhf whose inlined body injects v_div_* sequences into the hot loop
@@ -253,7 +258,7 @@ __global__ __launch_bounds__(NTHREADS, 1) void attn_fwd_kernel(FwdParams p) {
             int c = tid + r * NTHREADS;
             if (c < KCHUNKS) {
                 int row = c / CH_PER_ROW, ch = c % CH_PER_ROW;
-                *(uint4*)(lds.k[par] + row * D + swz(row, ch) * 8) = kst[r];
+                *(uint4*)(lds.k[par] + row * D + swz<D / 8>(row, ch) * 8) = kst[r];
             }
         }
         #pragma unroll
@@ -292,7 +297,7 @@ __global__ __launch_bounds__(NTHREADS, 1) void attn_fwd_kernel(FwdParams p) {
             (jmax - j0 == KVBLK - 1) &&
             (!p.causal || jmax <= wg_q_min) &&
             (!p.has_win || (wg_q_max - j0) <= p.win) &&
-            !p.kmask;
+            !p.kmask && !p.bias;
 
         __syncthreads();
         const bool stamp = p.ticks && blockIdx.x == 0 && bh == 0 && tid == 0
@@ -310,7 +315,7 @@ __global__ __launch_bounds__(NTHREADS, 1) void attn_fwd_kernel(FwdParams p) {
             #pragma unroll
             for (int ks = 0; ks < KSTEPS; ++ks) {
                 int chunk = ks * 2 + lhi;
-                bf16x8 kf = *(const bf16x8*)(lds.k[par] + krow * D + swz(krow, chunk) * 8);
+                bf16x8 kf = *(const bf16x8*)(lds.k[par] + krow * D + swz<D / 8>(krow, chunk) * 8);
                 s[kb] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf, qf[ks], s[kb], 0, 0, 0);
             }
         }
@@ -340,7 +345,7 @@ __global__ __launch_bounds__(NTHREADS, 1) void attn_fwd_kernel(FwdParams p) {
                 #pragma unroll
                 for (int ks = 0; ks < NBLK * 2; ++ks) {
                     int chunk = ks * 2 + lhi;
-                    bf16x8 vf = *(const bf16x8*)(lds.vt[par] + drow * KVBLK + swz(drow, chunk) * 8);
+                    bf16x8 vf = *(const bf16x8*)(lds.vt[par] + drow * KVBLK + swz<KVBLK / 8>(drow, chunk) * 8);
                     o_acc[db] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
                         vf, *(const bf16x8*)gfrag, o_acc[db], 0, 0, 0);
                 }
@@ -389,6 +394,15 @@ __global__ __launch_bounds__(NTHREADS, 1) void attn_fwd_kernel(FwdParams p) {
                     }
                     long j = j0 + kb * 32 + (r & 3) + 8 * (r >> 2) + 4 * lhi;
                     bool ok = j <= jmax;
+                    if (p.bias && ok) {
+                        // reference semantics: sim = qk*scale + bias
+                        // (natural log), applied after softclamp; our
+                        // softmax runs in the exp2 domain
+                        const long bi = p.bias_mat
+                            ? (((long)b * p.h + h) * p.nq + i_clamped) * p.nk + j
+                            : ((long)b * p.h + h) * p.nk + j;
+                        x += p.bias[bi] * LOG2E;
+                    }
                     if (p.causal) ok = ok && (j <= qpos_i);
                     if (p.has_win) ok = ok && (qpos_i - j <= p.win);
                     if (p.kmask) ok = ok && lds.kmask[par][j - j0];
@@ -404,6 +418,11 @@ __global__ __launch_bounds__(NTHREADS, 1) void attn_fwd_kernel(FwdParams p) {
         // rescale whenever the running max did not grow on any lane)
         float m_new = fmaxf(m_run, smax);
         const bool any_growth = !__all(smax <= m_run);
+        // all-masked rows: m_new == MASK -> exp2(x - m_new) would be 1 for
+        // every masked element (out = mean(V) instead of 0).  Clamping the
+        // exp-domain max keeps those exps at 0 while leaving real rows
+        // (|scores| << 1e37) untouched.
+        const float m_exp = fmaxf(m_new, -1.7e38f);
         // full tiles (no softclamp) kept RAW scores: exp2(fma(s, scale2, -m))
         const float escale = (full_tile && !SOFTCLAMP) ? scale2 : 1.f;
         uint32_t pk[NBLK * 8];                                      // packed bf16 pairs
@@ -411,9 +430,9 @@ __global__ __launch_bounds__(NTHREADS, 1) void attn_fwd_kernel(FwdParams p) {
         #pragma unroll
         for (int x2 = 0; x2 < NBLK * 8; ++x2) {
             float e0 = __builtin_amdgcn_exp2f(
-                __builtin_fmaf(s[x2 >> 3][(2 * x2) & 15], escale, -m_new));
+                __builtin_fmaf(s[x2 >> 3][(2 * x2) & 15], escale, -m_exp));
             float e1 = __builtin_amdgcn_exp2f(
-                __builtin_fmaf(s[x2 >> 3][(2 * x2 + 1) & 15], escale, -m_new));
+                __builtin_fmaf(s[x2 >> 3][(2 * x2 + 1) & 15], escale, -m_exp));
             partial[x2] = e0 + e1;
             union { __hip_bfloat162 h2; uint32_t u; } cvt;
             cvt.h2 = __float22bfloat162_rn(float2{e0, e1});
@@ -468,7 +487,7 @@ __global__ __launch_bounds__(NTHREADS, 1) void attn_fwd_kernel(FwdParams p) {
             #pragma unroll
             for (int ks = 0; ks < NBLK * 2; ++ks) {
                 int chunk = ks * 2 + lhi;
-                bf16x8 vf = *(const bf16x8*)(lds.vt[par] + drow * KVBLK + swz(drow, chunk) * 8);
+                bf16x8 vf = *(const bf16x8*)(lds.vt[par] + drow * KVBLK + swz<KVBLK / 8>(drow, chunk) * 8);
                 o_acc[db] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
                     vf, *(const bf16x8*)frag[ks], o_acc[db], 0, 0, 0);
             }
@@ -603,6 +622,8 @@ void launch_attn_fwd_merge(const FwdMergeParams& p, int head_dim, hipStream_t st
     dim3 block(256);
     if (head_dim == 64) {
         hipLaunchKernelGGL(attn_fwd_merge_kernel<64>, grid, block, 0, stream, p);
+    } else if (head_dim == 32) {
+        hipLaunchKernelGGL(attn_fwd_merge_kernel<32>, grid, block, 0, stream, p);
     } else {
         hipLaunchKernelGGL(attn_fwd_merge_kernel<128>, grid, block, 0, stream, p);
     }
@@ -631,6 +652,11 @@ void launch_attn_fwd(const FwdParams& p, int head_dim, hipStream_t stream) {
         else hipLaunchKernelGGL((attn_fwd_kernel<128, true, false>), grid, block, 0, stream, p);
         else if (pr) hipLaunchKernelGGL((attn_fwd_kernel<128, false, true>), grid, block, 0, stream, p);
         else hipLaunchKernelGGL((attn_fwd_kernel<128, false, false>), grid, block, 0, stream, p);
+    } else if (head_dim == 32) {
+        if (p.softclamp) if (pr) hipLaunchKernelGGL((attn_fwd_kernel<32, true, true>), grid, block, 0, stream, p);
+        else hipLaunchKernelGGL((attn_fwd_kernel<32, true, false>), grid, block, 0, stream, p);
+        else if (pr) hipLaunchKernelGGL((attn_fwd_kernel<32, false, true>), grid, block, 0, stream, p);
+        else hipLaunchKernelGGL((attn_fwd_kernel<32, false, false>), grid, block, 0, stream, p);
     } else {
         // unsupported head dim is a host-side error (checked in bindings)
         __builtin_trap();

@@ -471,11 +471,14 @@ __global__ __launch_bounds__(NTHREADS, 1) void attn_fwd_v2_kernel(FwdParams p) {
                     }
                     smax = fmaxf(smax, cross_half(smax));
 
-                    float m_new = fmaxf(m_run[qb], smax);
+                    float m_new0 = fmaxf(m_run[qb], smax);
                     growth[qb] = !__all(smax <= m_run[qb]);
                     alpha[qb] = growth[qb]
-                        ? __builtin_amdgcn_exp2f(m_run[qb] - m_new) : 1.f;
-                    m_run[qb] = m_new;
+                        ? __builtin_amdgcn_exp2f(m_run[qb] - m_new0) : 1.f;
+                    m_run[qb] = m_new0;
+                    // all-masked rows: clamp the exp-domain max so
+                    // exp2(MASK - MASK) cannot become 1 (see attn_fwd.hip)
+                    const float m_new = fmaxf(m_new0, -1.7e38f);
                     constexpr bool RAW = FULL && !SOFTCLAMP;
                     float part[4] = {0.f, 0.f, 0.f, 0.f};
                     uint32_t pk[NKV32 * 8];
@@ -598,7 +601,7 @@ __global__ __launch_bounds__(NTHREADS, 1) void attn_fwd_v2_kernel(FwdParams p) {
 
 // returns true if v2 handled this launch
 bool launch_attn_fwd_v2(const FwdParams& p, int head_dim, hipStream_t stream) {
-    if (p.ablate) return false;   // ticks ARE supported (7 stamps/tile)
+    if (p.ablate || p.bias) return false;   // ticks ARE supported (7 stamps/tile)
     long qtiles = (p.nq + v2::QROWS_WG - 1) / v2::QROWS_WG;
     dim3 grid(qtiles * p.b * p.h, 1, p.kv_split > 1 ? p.kv_split : 1);
     dim3 block(v2::NTHREADS);

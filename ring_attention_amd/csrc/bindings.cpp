@@ -30,12 +30,15 @@ void attn_fwd(
     int64_t win, bool has_win,
     bool softclamp, double softclamp_value,
     bool is_first, bool is_last, int64_t kv_split, int64_t ablate,
-    std::optional<at::Tensor> ticks) {
+    std::optional<at::Tensor> ticks,
+    std::optional<at::Tensor> bias, bool bias_mat) {
     CHECK_BF16_CONTIG(q); CHECK_BF16_CONTIG(k); CHECK_BF16_CONTIG(v);
     TORCH_CHECK(q.dim() == 4 && k.dim() == 4 && v.dim() == 4, "q/k/v must be (B,N,H,D)");
     const int64_t B = q.size(0), Nq = q.size(1), H = q.size(2), D = q.size(3);
     const int64_t Nk = k.size(1), HK = k.size(2);
-    TORCH_CHECK(D == 64 || D == 128, "head dim must be 64 or 128 (got ", D, ")");
+    TORCH_CHECK(D == 32 || D == 64 || D == 128,
+                "head dim must be 32/64/128 (got ", D, "); arbitrary d <= 128 is "
+                "padded at the python layer");
     TORCH_CHECK(H % HK == 0, "q heads must be a multiple of kv heads");
     TORCH_CHECK(k.size(0) == B && v.size(0) == B && v.size(1) == Nk && v.size(2) == HK && v.size(3) == D);
 
@@ -82,6 +85,13 @@ void attn_fwd(
     p.ablate = (int)ablate;
     p.ticks = nullptr;
     if (ticks.has_value()) p.ticks = (unsigned long long*)ticks->data_ptr();
+    p.bias = nullptr; p.bias_mat = bias_mat ? 1 : 0;
+    if (bias.has_value()) {
+        CHECK_F32_CONTIG((*bias));
+        TORCH_CHECK(bias->numel() == (bias_mat ? B * H * Nq * Nk : B * H * Nk),
+                    "bias must be (B,H,Nk) or (B,H,Nq,Nk) fp32");
+        p.bias = bias->data_ptr<float>();
+    }
     // causal pairing (uniform per-WG work) whenever the diagonal cuts this
     // kv range and nothing incompatible is on (splits balance differently;
     // windows are already uniform; ticks index per tile)
@@ -92,7 +102,7 @@ void attn_fwd(
         // any grid.z split) still fills the 256 CUs (measured: engaging
         // below that idles half the chip and loses)
         const long z = kv_split > 1 ? kv_split : 1;
-        if (causal && !has_win && diag < Nk
+        if (causal && !has_win && diag < Nk && !bias.has_value()
             && ((T + 1) / 2) * B * H * z >= 256 && !ticks.has_value()
             && ablate == 0 && !std::getenv("RING_ATTN_NO_PAIR"))
             p.paired = (int)T;
@@ -144,13 +154,14 @@ void attn_bwd(
     bool softclamp, double softclamp_value, bool accumulate, int64_t split,
     int64_t which,     // 0 = both, 1 = dq only, 2 = dk/dv only
     std::optional<at::Tensor> desc_dq,    // int32 (U,3): tile, t_lo, t_hi
-    std::optional<at::Tensor> desc_dkv) {
+    std::optional<at::Tensor> desc_dkv,
+    std::optional<at::Tensor> bias, bool bias_mat) {
     CHECK_BF16_CONTIG(q); CHECK_BF16_CONTIG(k); CHECK_BF16_CONTIG(v); CHECK_BF16_CONTIG(dout);
     CHECK_F32_CONTIG(lse); CHECK_F32_CONTIG(delta);
     CHECK_F32_CONTIG(dq); CHECK_F32_CONTIG(dk); CHECK_F32_CONTIG(dv);
     const int64_t B = q.size(0), Nq = q.size(1), H = q.size(2), D = q.size(3);
     const int64_t Nk = k.size(1), HK = k.size(2);
-    TORCH_CHECK(D == 64 || D == 128, "head dim must be 64 or 128");
+    TORCH_CHECK(D == 32 || D == 64 || D == 128, "head dim must be 32/64/128");
     TORCH_CHECK(dq.numel() == q.numel() && dk.numel() == k.numel() && dv.numel() == v.numel());
     TORCH_CHECK(lse.numel() == B * H * Nq && delta.numel() == B * H * Nq);
 
@@ -171,11 +182,18 @@ void attn_bwd(
     p.win = win; p.has_win = has_win;
     p.accumulate = accumulate;
     p.split = (int)split;
+    p.bias = nullptr; p.bias_mat = bias_mat ? 1 : 0;
+    if (bias.has_value()) {
+        CHECK_F32_CONTIG((*bias));
+        TORCH_CHECK(bias->numel() == (bias_mat ? B * H * Nq * Nk : B * H * Nk));
+        p.bias = bias->data_ptr<float>();
+    }
 
     // causal pairing: uniform per-WG work when the diagonal cuts this range.
     // dq pairs over Q tiles, dkv over KV tiles (different counts when the
     // kv range is gathered), so set per launch.
     const bool pair_ok = causal && !has_win && diag < Nk
+                         && !bias.has_value()
                          && !std::getenv("RING_ATTN_NO_PAIR");
     const long z = split > 1 ? split : 1;
     auto set_desc = [&](const std::optional<at::Tensor>& dsc) {
@@ -210,7 +228,7 @@ at::Tensor rotary_apply(at::Tensor x, at::Tensor cos_t, at::Tensor sin_t, double
     CHECK_BF16_CONTIG(x);
     CHECK_F32_CONTIG(cos_t); CHECK_F32_CONTIG(sin_t);
     const int64_t B = x.size(0), N = x.size(1), H = x.size(2), D = x.size(3);
-    TORCH_CHECK(D == 64 || D == 128, "head dim must be 64 or 128");
+    TORCH_CHECK(D == 32 || D == 64 || D == 128, "head dim must be 32/64/128");
     TORCH_CHECK(cos_t.size(0) == N && cos_t.size(1) == D / 2);
     auto out = at::empty_like(x);
     RotaryParams p{};
@@ -226,7 +244,8 @@ at::Tensor rotary_apply(at::Tensor x, at::Tensor cos_t, at::Tensor sin_t, double
     return out;
 }
 
-std::vector<at::Tensor> decode_partial(at::Tensor q, at::Tensor k, at::Tensor v) {
+std::vector<at::Tensor> decode_partial(at::Tensor q, at::Tensor k, at::Tensor v,
+                                       double sm_scale) {
     // q (B,H,1,D); k,v (B,H,N,D) bf16
     // -> (out fp32 (S,B,H,1,D), lse fp32 (S,B,H,1,1)): S kv-chunk partials,
     //    merged by the caller with the standard logsumexp combine
@@ -242,7 +261,7 @@ std::vector<at::Tensor> decode_partial(at::Tensor q, at::Tensor k, at::Tensor v)
     p.q = q.data_ptr(); p.k = k.data_ptr(); p.v = v.data_ptr();
     p.out = out.data_ptr<float>(); p.lse = lse.data_ptr<float>();
     p.b = (int)B; p.h = (int)H; p.n = N;
-    p.scale = (float)(1.0 / std::sqrt((double)D));
+    p.scale = sm_scale > 0 ? (float)sm_scale : (float)(1.0 / std::sqrt((double)D));
     p.chunks = chunks;
     launch_decode_partial(p, (int)D, at::hip::getCurrentHIPStream());
     TORCH_CHECK(hipGetLastError() == hipSuccess, "decode launch failed");
@@ -253,7 +272,7 @@ at::Tensor attn_delta(at::Tensor dout, at::Tensor out) {
     // dout, out (B,N,H,D) bf16 -> delta fp32 (B,H,N) = rowsum(dout*out)
     CHECK_BF16_CONTIG(dout); CHECK_BF16_CONTIG(out);
     const int64_t B = dout.size(0), N = dout.size(1), H = dout.size(2), D = dout.size(3);
-    TORCH_CHECK(D == 64 || D == 128, "head dim must be 64 or 128");
+    TORCH_CHECK(D == 32 || D == 64 || D == 128, "head dim must be 32/64/128");
     auto delta = at::empty({B, H, N}, dout.options().dtype(at::kFloat));
     DeltaParams p{};
     p.dout = dout.data_ptr(); p.out = out.data_ptr();
@@ -267,10 +286,27 @@ at::Tensor attn_delta(at::Tensor dout, at::Tensor out) {
 }  // namespace ring_attn
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
-    mod.def("attn_fwd", &ring_attn::attn_fwd, "CDNA4 flash attention forward (resumable)");
+    mod.def("attn_fwd", &ring_attn::attn_fwd, "CDNA4 flash attention forward (resumable)",
+            py::arg("q"), py::arg("k"), py::arg("v"), py::arg("kmask"),
+            py::arg("o_acc"), py::arg("m"), py::arg("l"), py::arg("out"),
+            py::arg("lse"), py::arg("scale"), py::arg("causal"), py::arg("diag"),
+            py::arg("q_stride"), py::arg("win"), py::arg("has_win"),
+            py::arg("softclamp"), py::arg("softclamp_value"), py::arg("is_first"),
+            py::arg("is_last"), py::arg("kv_split"), py::arg("ablate"),
+            py::arg("ticks"), py::arg("bias") = std::nullopt,
+            py::arg("bias_mat") = false);
     mod.def("attn_fwd_merge", &ring_attn::attn_fwd_merge, "merge kv-split partials");
-    mod.def("attn_bwd", &ring_attn::attn_bwd, "CDNA4 flash attention backward");
-    mod.def("decode_partial", &ring_attn::decode_partial, "CDNA4 single-query decode partial");
+    mod.def("attn_bwd", &ring_attn::attn_bwd, "CDNA4 flash attention backward",
+            py::arg("q"), py::arg("k"), py::arg("v"), py::arg("dout"),
+            py::arg("kmask"), py::arg("lse"), py::arg("delta"), py::arg("dq"),
+            py::arg("dk"), py::arg("dv"), py::arg("scale"), py::arg("causal"),
+            py::arg("diag"), py::arg("q_stride"), py::arg("win"),
+            py::arg("has_win"), py::arg("softclamp"), py::arg("softclamp_value"),
+            py::arg("accumulate"), py::arg("split"), py::arg("which"),
+            py::arg("desc_dq"), py::arg("desc_dkv"),
+            py::arg("bias") = std::nullopt, py::arg("bias_mat") = false);
+    mod.def("decode_partial", &ring_attn::decode_partial, "CDNA4 single-query decode partial",
+            py::arg("q"), py::arg("k"), py::arg("v"), py::arg("sm_scale") = -1.0);
     mod.def("attn_delta", &ring_attn::attn_delta, "fused delta = rowsum(dO*O) preprocess");
     mod.def("rotary_apply", &ring_attn::rotary_apply, "fused rotary embedding (table-driven)");
 }

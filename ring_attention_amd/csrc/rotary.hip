@@ -50,6 +50,8 @@ void launch_rotary(const RotaryParams& p, int head_dim, hipStream_t stream) {
     dim3 block(256);
     if (head_dim == 64) {
         hipLaunchKernelGGL(rotary_kernel<64>, grid, block, 0, stream, p);
+    } else if (head_dim == 32) {
+        hipLaunchKernelGGL(rotary_kernel<32>, grid, block, 0, stream, p);
     } else if (head_dim == 128) {
         hipLaunchKernelGGL(rotary_kernel<128>, grid, block, 0, stream, p);
     } else {

@@ -41,6 +41,6 @@ def require():
     return ext
 
 
-def decode_partial(q, k, v):
+def decode_partial(q, k, v, sm_scale=-1.0):
     """Single-query decode partial via the HIP kernel: (out fp32, lse fp32)."""
-    return require().decode_partial(q, k, v)
+    return require().decode_partial(q, k, v, sm_scale)

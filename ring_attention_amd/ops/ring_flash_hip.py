@@ -36,6 +36,7 @@ with its §2.5 bugs fixed.
 from __future__ import annotations
 
 import torch
+import torch.nn.functional as F
 from torch import Tensor
 from torch.autograd import Function
 
@@ -212,15 +213,44 @@ def _use_desc(causal, lookback, diag, nk, grid_wgs):
             and not os.environ.get("RING_ATTN_NO_DESC"))
 
 
+# ---------------------------------------------------------------------------
+# arbitrary head dims: the gfx950 kernels are instantiated for d in
+# {32, 64, 128}; any other d <= 128 is zero-padded to the next instantiated
+# size (mathematically exact: zero q/k pad columns contribute nothing to
+# QK^T, zero v pad columns are sliced off, and autograd differentiates the
+# pad/slice).  Counterpart of the reference's next-pow2 BLOCK_HEADDIM padding
+# (triton_flash_attn.py:359,380-421).
+# ---------------------------------------------------------------------------
+_KERNEL_DIMS = (32, 64, 128)
+
+
+def kernel_head_dim(d: int) -> int:
+    for kd in _KERNEL_DIMS:
+        if d <= kd:
+            return kd
+    raise ValueError(f"head dim {d} > 128 not supported by the gfx950 kernels")
+
+
+def _pad_head_dim(q, k, v):
+    """Zero-pad head dims to the kernel size (the callers pass the TRUE
+    d**-0.5 softmax scale through sm_scale, so numerics are exact)."""
+    d = q.shape[-1]
+    kd = kernel_head_dim(d)
+    if kd == d:
+        return q, k, v
+    return F.pad(q, (0, kd - d)), F.pad(k, (0, kd - d)), F.pad(v, (0, kd - d))
+
+
 class RingFlashAttentionHIPFunction(Function):
     @staticmethod
     def forward(ctx, q, k, v, mask, causal, bucket_size, ring_reduce, striped,
-                max_lookback_seq_len, ring_size, softclamp_qk_sim, softclamp_value):
+                max_lookback_seq_len, ring_size, softclamp_qk_sim, softclamp_value,
+                sm_scale=None):
         assert q.is_cuda, "HIP path requires GPU tensors"
         b, n, h, d = q.shape
         hk = k.shape[2]
         in_dtype = q.dtype
-        scale = d ** -0.5
+        scale = sm_scale if sm_scale is not None else d ** -0.5
         lookback = max_lookback_seq_len
         if lookback is not None:
             assert causal, "lookback (sliding window) requires causal"
@@ -271,6 +301,7 @@ class RingFlashAttentionHIPFunction(Function):
                              softclamp_qk_sim, softclamp_value, True, True, 1, 0, None)
             ctx.save_for_backward(qb, kb, vb, out, lse,
                                   mask_u8 if mask_u8 is not None else torch.empty(0))
+            ctx.sm_scale = scale
             ctx.params = (causal, striped, lookback, hops, softclamp_qk_sim,
                           softclamp_value, use_ring, topo.ring_size, in_dtype,
                           "allgather")
@@ -343,6 +374,7 @@ class RingFlashAttentionHIPFunction(Function):
 
         ctx.save_for_backward(qb, kb, vb, out, lse,
                               mask_u8 if mask_u8 is not None else torch.empty(0))
+        ctx.sm_scale = scale
         ctx.params = (causal, striped, lookback, hops, softclamp_qk_sim,
                       softclamp_value, use_ring, topo.ring_size, in_dtype, "ring")
         return out.to(in_dtype), lse
@@ -355,7 +387,7 @@ class RingFlashAttentionHIPFunction(Function):
          use_ring, ring_size, in_dtype, strategy) = ctx.params
         b, n, h, d = qb.shape
         hk = kb.shape[2]
-        scale = d ** -0.5
+        scale = ctx.sm_scale
 
         topo = RingTopology(ring_size if use_ring else 1,
                             rank=None if use_ring else 0,
@@ -498,7 +530,7 @@ class RingFlashAttentionHIPFunction(Function):
         dv_home = dkv[1].view(b, hk, d, n).permute(0, 3, 1, 2).contiguous()
 
         return (dq.to(in_dtype), dk_home.to(in_dtype), dv_home.to(in_dtype),
-                None, None, None, None, None, None, None, None, None)
+                None, None, None, None, None, None, None, None, None, None)
 
 
 def ring_flash_attn_hip_(
@@ -513,9 +545,18 @@ def ring_flash_attn_hip_(
     softclamp_qk_sim: bool = False,
     softclamp_value: float = 50.0,
 ) -> tuple[Tensor, Tensor]:
-    return RingFlashAttentionHIPFunction.apply(
+    d = q.shape[-1]
+    sm_scale = None
+    if kernel_head_dim(d) != d:
+        q, k, v = _pad_head_dim(q, k, v)
+        sm_scale = d ** -0.5
+    out, lse = RingFlashAttentionHIPFunction.apply(
         q, k, v, mask, causal, bucket_size, ring_reduce_col, striped_ring_attn,
-        max_lookback_seq_len, ring_size, softclamp_qk_sim, softclamp_value)
+        max_lookback_seq_len, ring_size, softclamp_qk_sim, softclamp_value,
+        sm_scale)
+    if out.shape[-1] != d:
+        out = out[..., :d]
+    return out, lse
 
 
 def ring_flash_attn_hip(q, k, v, **kwargs) -> Tensor:
@@ -532,10 +573,10 @@ class FlashAttnOffsetFunction(Function):
     zig_zag_attention.py:123-139)."""
 
     @staticmethod
-    def forward(ctx, q, k, v, q_offset, causal):
+    def forward(ctx, q, k, v, q_offset, causal, sm_scale=None):
         assert q.is_cuda
         b, n, h, d = q.shape
-        scale = d ** -0.5
+        scale = sm_scale if sm_scale is not None else d ** -0.5
         ext = hip_ext.require()
         qb = q.to(torch.bfloat16).contiguous()
         kb = k.to(torch.bfloat16).contiguous()
@@ -546,17 +587,16 @@ class FlashAttnOffsetFunction(Function):
                      scale, causal, q_offset, 1, 0, False, False, 50.0,
                      True, True, 1, 0, None)
         ctx.save_for_backward(qb, kb, vb, out, lse)
-        ctx.meta = (q_offset, causal, q.dtype)
+        ctx.meta = (q_offset, causal, q.dtype, scale)
         return out.to(q.dtype)
 
     @staticmethod
     def backward(ctx, do):
         qb, kb, vb, out, lse = ctx.saved_tensors
-        q_offset, causal, in_dtype = ctx.meta
+        q_offset, causal, in_dtype, scale = ctx.meta
         b, n, h, d = qb.shape
         hk = kb.shape[2]
         nk = kb.shape[1]
-        scale = d ** -0.5
         ext = hip_ext.require()
         dob = do.to(torch.bfloat16).contiguous()
         delta = ext.attn_delta(dob, out)
@@ -573,9 +613,115 @@ class FlashAttnOffsetFunction(Function):
                      ddq, ddkv)
         dk = dk_n.permute(0, 2, 1, 3).contiguous()
         dv = dv_n.permute(0, 3, 1, 2).contiguous()
-        return (dq.to(in_dtype), dk.to(in_dtype), dv.to(in_dtype), None, None)
+        return (dq.to(in_dtype), dk.to(in_dtype), dv.to(in_dtype), None, None, None)
 
 
 def flash_attn_offset(q, k, v, q_offset=0, causal=True):
     """q (b, n, h, d) attends k/v (b, nk, hk, d) with attend(i,j) <=> j <= i + q_offset."""
-    return FlashAttnOffsetFunction.apply(q, k, v, q_offset, causal)
+    d = q.shape[-1]
+    sm_scale = None
+    if kernel_head_dim(d) != d:
+        q, k, v = _pad_head_dim(q, k, v)
+        sm_scale = d ** -0.5
+    out = FlashAttnOffsetFunction.apply(q, k, v, q_offset, causal, sm_scale)
+    return out[..., :d] if out.shape[-1] != d else out
+
+
+class FlashAttnFunction(Function):
+    """Single-device flash attention with the FULL L0 flag surface of the
+    reference kernel pair (triton_flash_attn.py:304-430, 988-1128): additive
+    bias (vector (b,h,nk) or matrix (b,h,n,nk)), causal with optional strict
+    diagonal (striped attention), key-pad mask, tanh softclamp and sliding
+    window.  bias gradients are not produced (parity: the reference backward
+    has no db either)."""
+
+    @staticmethod
+    def forward(ctx, q, k, v, bias, bias_is_matrix, key_pad_mask, causal,
+                causal_mask_diagonal, softclamp, softclamp_value, window,
+                sm_scale=None):
+        assert q.is_cuda
+        b, n, h, d = q.shape
+        scale = sm_scale if sm_scale is not None else d ** -0.5
+        ext = hip_ext.require()
+        qb = q.to(torch.bfloat16).contiguous()
+        kb = k.to(torch.bfloat16).contiguous()
+        vb = v.to(torch.bfloat16).contiguous()
+        biasf = bias.float().contiguous() if bias is not None else None
+        mk = key_pad_mask.to(torch.uint8).contiguous() if key_pad_mask is not None else None
+        out = torch.empty_like(qb)
+        lse = torch.empty(b, h, n, device=q.device, dtype=torch.float32)
+        diag = -1 if (causal and causal_mask_diagonal) else 0
+        ext.attn_fwd(qb, kb, vb, mk, None, None, None, out, lse,
+                     scale, causal, diag, 1,
+                     window if window is not None else 0, window is not None,
+                     softclamp, softclamp_value, True, True, 1, 0, None,
+                     bias=biasf, bias_mat=bool(bias_is_matrix))
+        ctx.save_for_backward(qb, kb, vb, out, lse,
+                              *( (biasf,) if biasf is not None else () ),
+                              *( (mk,) if mk is not None else () ))
+        ctx.meta = (bias is not None, bias_is_matrix, key_pad_mask is not None,
+                    causal, diag, softclamp, softclamp_value, window, q.dtype,
+                    scale)
+        return out.to(q.dtype)
+
+    @staticmethod
+    def backward(ctx, do):
+        (has_bias, bias_mat, has_mask, causal, diag, softclamp,
+         softclamp_value, window, in_dtype, scale) = ctx.meta
+        saved = list(ctx.saved_tensors)
+        qb, kb, vb, out, lse = saved[:5]
+        rest = saved[5:]
+        biasf = rest.pop(0) if has_bias else None
+        mk = rest.pop(0) if has_mask else None
+        b, n, h, d = qb.shape
+        hk = kb.shape[2]
+        nk = kb.shape[1]
+        ext = hip_ext.require()
+        dob = do.to(torch.bfloat16).contiguous()
+        delta = ext.attn_delta(dob, out)
+        dq = torch.empty(b, n, h, d, device=qb.device, dtype=torch.float32)
+        dk_n = torch.empty(b, hk, nk, d, device=qb.device, dtype=torch.float32)
+        dv_n = torch.empty(b, hk, d, nk, device=qb.device, dtype=torch.float32)
+        ext.attn_bwd(qb, kb, vb, dob, mk, lse, delta, dq, dk_n, dv_n,
+                     scale, causal, diag, 1,
+                     window if window is not None else 0, window is not None,
+                     softclamp, softclamp_value, False, 1, 0, None, None,
+                     bias=biasf, bias_mat=bool(bias_mat))
+        dk = dk_n.permute(0, 2, 1, 3).contiguous()
+        dv = dv_n.permute(0, 3, 1, 2).contiguous()
+        return (dq.to(in_dtype), dk.to(in_dtype), dv.to(in_dtype),
+                None, None, None, None, None, None, None, None, None)
+
+
+def flash_attn(
+    q: Tensor, k: Tensor, v: Tensor,
+    bias: Tensor | None = None,
+    key_pad_mask: Tensor | None = None,
+    causal: bool = False,
+    causal_mask_diagonal: bool = False,
+    softclamp_qk_sim: bool = False,
+    softclamp_value: float = 50.0,
+    window: int | None = None,
+) -> Tensor:
+    """Single-device flash attention, (b, n, h, d) layout, any d <= 128.
+
+    ``bias``: additive attention bias in the natural-log domain, shaped
+    (b, h, nk) (broadcast over query rows) or (b, h, n, nk).  GQA: q heads h
+    pair kv heads via qh % hk (reference tile convention)."""
+    d = q.shape[-1]
+    bias_is_matrix = bias is not None and bias.dim() == 4
+    if bias is not None:
+        bh = q.shape[2]
+        if bias_is_matrix:
+            bias = bias.expand(q.shape[0], bh, q.shape[1], k.shape[1])
+        else:
+            bias = bias.expand(q.shape[0], bh, k.shape[1])
+    sm_scale = None
+    if kernel_head_dim(d) != d:
+        q, k, v = _pad_head_dim(q, k, v)
+        sm_scale = d ** -0.5
+    out = FlashAttnFunction.apply(q, k, v, bias, bias_is_matrix, key_pad_mask,
+                                  causal, causal_mask_diagonal,
+                                  softclamp_qk_sim, softclamp_value, window,
+                                  sm_scale)
+    return out[..., :d] if out.shape[-1] != d else out

@@ -25,10 +25,20 @@ def _local_decode_partial(q: Tensor, k: Tensor, v: Tensor) -> tuple[Tensor, Tens
     if q.is_cuda:
         from .ops import hip_ext
         if hip_ext.available():
+            d = q.shape[-1]
+            dv = v.shape[-1]
+            kd = 64 if max(d, dv) <= 64 else 128
+            import torch.nn.functional as _F
+            qb = _F.pad(q, (0, kd - d)) if d != kd else q
+            kb = _F.pad(k, (0, kd - d)) if d != kd else k
+            vb = _F.pad(v, (0, kd - dv)) if dv != kd else v
             outs, lses = hip_ext.decode_partial(
-                q.to(torch.bfloat16).contiguous(),
-                k.to(torch.bfloat16).contiguous(),
-                v.to(torch.bfloat16).contiguous())
+                qb.to(torch.bfloat16).contiguous(),
+                kb.to(torch.bfloat16).contiguous(),
+                vb.to(torch.bfloat16).contiguous(),
+                sm_scale=d ** -0.5)
+            if dv != kd:
+                outs = outs[..., :dv]
             # merge the S kv-chunk partials (same math as the cross-rank merge)
             m = lses.max(dim=0).values                       # (b,h,1,1)
             w = (lses - m[None]).exp()                       # (S,b,h,1,1)
