@@ -40,7 +40,23 @@ def main():
     ap.add_argument("--striped", action="store_true")
     ap.add_argument("--fwd-only", action="store_true",
                     help="measure forward only (diagnostics; not the headline metric)")
+    ap.add_argument("--config", type=int, default=None, choices=[2, 3, 4, 5],
+                    help="BASELINE.json config preset: 2=non-causal ring 64k-class, "
+                         "3=causal striped 256k-class, 4=GQA 32q/4kv zig-zag "
+                         "causal 1M-class, 5=tree-decode step 128k KV")
     args = ap.parse_args()
+
+    # BASELINE.json config presets (per-GPU shards of the 8-GPU configs, so
+    # the same command measures any world size the driver launches)
+    if args.config == 2:
+        pass                                    # = the defaults
+    elif args.config == 3:
+        args.causal = True; args.striped = True; args.seq_per_gpu = 32768
+    elif args.config == 4:
+        args.causal = True
+        args.heads, args.kv_heads, args.seq_per_gpu = 32, 4, 131072
+    elif args.config == 5:
+        args.seq_per_gpu = 131072
 
     world = int(os.environ.get("WORLD_SIZE", "1"))
     rank = int(os.environ.get("RANK", "0"))
@@ -66,18 +82,55 @@ def main():
 
     b, n, h, d = args.batch, args.seq_per_gpu, args.heads, args.d_head
     hk = args.kv_heads if args.kv_heads is not None else h
+    if not on_gpu and args.config in (4, 5):
+        n = args.seq_per_gpu = min(args.seq_per_gpu, 512)
     torch.manual_seed(1234 + rank)
-    q = torch.randn(b, n, h, d, device=device, dtype=dtype, requires_grad=True)
-    k = torch.randn(b, n, hk, d, device=device, dtype=dtype, requires_grad=True)
-    v = torch.randn(b, n, hk, d, device=device, dtype=dtype, requires_grad=True)
 
-    def step():
-        out, _ = attn(q, k, v, causal=args.causal, ring_reduce_col=True,
-                      striped_ring_attn=args.striped, ring_size=world,
-                      bucket_size=min(n, 1024))
-        if not args.fwd_only:
-            out.backward(out.detach())  # fwd + full bwd (dq, dk, dv incl. ring)
-            q.grad = None; k.grad = None; v.grad = None
+    if args.config == 5:
+        # tree-attention decode: single-query step over world-sharded KV
+        from ring_attention_amd.tree_decode import tree_attn_decode
+        q5 = torch.randn(b, h, 1, d, device=device, dtype=dtype)
+        k5 = torch.randn(b, h, n, d, device=device, dtype=dtype)
+        v5 = torch.randn(b, h, n, d, device=device, dtype=dtype)
+
+        def step():
+            tree_attn_decode(q5, k5, v5, shard_kv_seq=False)
+    elif args.config == 4:
+        # zig-zag causal GQA + fused rotary (per-GPU slice of the 1M config)
+        from ring_attention_amd.zigzag import zig_zag_attn
+        from ring_attention_amd.models.rotary import apply_rotary_pos_emb
+        q = torch.randn(b, n, h, d, device=device, dtype=dtype, requires_grad=True)
+        k = torch.randn(b, n, hk, d, device=device, dtype=dtype, requires_grad=True)
+        v = torch.randn(b, n, hk, d, device=device, dtype=dtype, requires_grad=True)
+        half = n // 2
+        starts = (rank * half, (2 * world - 1 - rank) * half)
+        pos = torch.cat([starts[0] + torch.arange(half, device=device),
+                         starts[1] + torch.arange(n - half, device=device)])
+        inv_freq = 1.0 / (10000.0 ** (torch.arange(0, d, 2, device=device).float() / d))
+        f_ = torch.einsum("i,j->ij", pos.float(), inv_freq)
+        freqs = torch.cat((f_, f_), dim=-1)
+
+        def step():
+            qr = apply_rotary_pos_emb(freqs, q)
+            kr = apply_rotary_pos_emb(freqs, k)
+            out = zig_zag_attn(qr.permute(0, 2, 1, 3), kr.permute(0, 2, 1, 3),
+                               v.permute(0, 2, 1, 3), causal=True,
+                               q_chunk_starts=starts)
+            if not args.fwd_only:
+                out.backward(out.detach())
+                q.grad = None; k.grad = None; v.grad = None
+    else:
+        q = torch.randn(b, n, h, d, device=device, dtype=dtype, requires_grad=True)
+        k = torch.randn(b, n, hk, d, device=device, dtype=dtype, requires_grad=True)
+        v = torch.randn(b, n, hk, d, device=device, dtype=dtype, requires_grad=True)
+
+        def step():
+            out, _ = attn(q, k, v, causal=args.causal, ring_reduce_col=True,
+                          striped_ring_attn=args.striped, ring_size=world,
+                          bucket_size=min(n, 1024))
+            if not args.fwd_only:
+                out.backward(out.detach())  # fwd + full bwd (dq, dk, dv incl. ring)
+                q.grad = None; k.grad = None; v.grad = None
 
     for _ in range(args.warmup):
         step()
@@ -131,8 +184,36 @@ def main():
         fwd_flops_per_gpu /= 2
     total_flops = (1.0 if args.fwd_only else 2.5) * fwd_flops_per_gpu * world * args.steps
     tflops_aggregate = total_flops / secs / 1e12
+    if args.config == 4:
+        # zig-zag attends the GLOBAL gathered kv per GPU: n * n_total * ... is
+        # already that; convention unchanged
+        pass
 
-    if rank == 0:
+    if rank == 0 and args.config == 5:
+        print(json.dumps({
+            "metric": "tree_decode_step_latency",
+            "value": round(ms_per_step * 1e3, 1),
+            "unit": "us/step",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(ms_per_step, 4),
+            "higher_is_better": False,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "bf16" if on_gpu else "fp32-cpu-fallback",
+            "data": "synthetic",
+            "config": {
+                "model": "tree_attn_decode (single-query, sharded KV)",
+                "global_batch": b,
+                "seq_len": n * world, "seq_per_gpu": n,
+                "heads": h, "d_head": d,
+                "parallelism": f"tree{world}",
+                "hipgraph": graph is not None,
+                "kv_stream_TBps": round(b * h * n * d * 2 * 2 / (ms_per_step * 1e-3) / 1e12, 2),
+            },
+        }))
+    elif rank == 0:
         print(json.dumps({
             "metric": "attn_tflops",
             "value": round(tflops_aggregate, 2),
@@ -147,8 +228,9 @@ def main():
             "dtype": "bf16" if on_gpu else "fp32-cpu-fallback",
             "data": "synthetic",
             "config": {
-                "model": "ring_flash_attn (non-causal, d_head 64)" if not args.causal
-                         else "ring_flash_attn (causal)",
+                "model": ("zig_zag_attn (causal GQA + rotary)" if args.config == 4
+                          else "ring_flash_attn (non-causal, d_head 64)" if not args.causal
+                          else "ring_flash_attn (causal%s)" % (" striped" if args.striped else "")),
                 "global_batch": b,
                 "seq_len": n_total,
                 "seq_per_gpu": n,

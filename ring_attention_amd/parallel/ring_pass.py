@@ -59,13 +59,20 @@ def ring_pass(topo: RingTopology, *tensors: Tensor, num_hops: int = 1) -> tuple[
     """
     if topo.ring_size == 1 or not is_distributed():
         return tuple(tensors)
+    from ..utils.tracing import GLOBAL_RING_STATS, trace_range
     cur = [t.contiguous() for t in tensors]
     bufs = [torch.empty_like(t) for t in cur]
-    for _ in range(num_hops % topo.ring_size):  # ring_size hops is the identity
-        reqs = _exchange(topo, cur, bufs)
-        for r in reqs:
-            r.wait()
-        cur, bufs = bufs, cur
+    nbytes = sum(t.element_size() * t.numel() for t in cur)
+    GLOBAL_RING_STATS.start()
+    hops_done = 0
+    with trace_range("ring_pass.multi_hop"):
+        for _ in range(num_hops % topo.ring_size):  # ring_size hops = identity
+            reqs = _exchange(topo, cur, bufs)
+            for r in reqs:
+                r.wait()
+            cur, bufs = bufs, cur
+            hops_done += 1
+    GLOBAL_RING_STATS.stop(hops_done, hops_done * nbytes)
     return tuple(cur)
 
 
@@ -89,21 +96,27 @@ def all_ring_pass(
         # even with a single compute hop on a real ring, nothing needs sending
         return
 
+    from ..utils.tracing import GLOBAL_RING_STATS, trace_range
     cur = [t.contiguous() for t in tensors]
     bufs = [torch.empty_like(t) for t in cur]
     reqs = None
+    nbytes = sum(t.element_size() * t.numel() for t in cur)
 
     for hop in range(max_hops):
         is_last = hop == max_hops - 1
         if not is_last:
             # post the exchange for the NEXT hop now; compute on `cur` overlaps it
-            reqs = _exchange(topo, cur, bufs)
+            with trace_range(f"ring_pass.post_hop{hop + 1}"):
+                reqs = _exchange(topo, cur, bufs)
+            GLOBAL_RING_STATS.start()
 
         yield RingInfo(hop, topo.source_of_hop(hop), hop == 0, is_last), tuple(cur)
 
         if not is_last:
-            for r in reqs:
-                r.wait()
+            with trace_range(f"ring_pass.wait_hop{hop + 1}"):
+                for r in reqs:
+                    r.wait()
+            GLOBAL_RING_STATS.stop(1, nbytes)
             cur, bufs = bufs, cur
 
 
@@ -147,15 +160,20 @@ class RingAccumulator:
             return
 
         if self._reqs is not None:
+            from ..utils.tracing import GLOBAL_RING_STATS
             for r in self._reqs:
                 r.wait()
             self._reqs = None
+            GLOBAL_RING_STATS.stop(
+                1, self._recv.element_size() * self._recv.numel())
             contribution = contribution + self._recv
 
         self._acc = contribution.contiguous()
         if not is_last:
+            from ..utils.tracing import GLOBAL_RING_STATS
             self._recv = torch.empty_like(self._acc)
             self._reqs = _exchange(self.topo, [self._acc], [self._recv])
+            GLOBAL_RING_STATS.start()
 
     def finish(self, total_hops: int) -> Tensor:
         """Route the final accumulator to its home rank and return it.

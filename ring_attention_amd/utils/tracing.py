@@ -4,7 +4,9 @@
   rocprofv3 --marker-trace timelines) plus a torch.profiler record_function
   scope, when those facilities exist; free no-ops otherwise.
 - ``RingStats``: per-process counters for ring communication (hops, bytes,
-  wall time) that the attention functions update; readable for logging.
+  comm wall time) updated by the ring engine itself (``all_ring_pass``,
+  ``ring_pass`` multi-hop routing and ``RingAccumulator``), so every ring
+  attention strategy feeds them; read/reset via GLOBAL_RING_STATS.
 """
 
 from __future__ import annotations
