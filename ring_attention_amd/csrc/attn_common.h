@@ -131,12 +131,15 @@ struct DeltaParams {
 void launch_attn_delta(const DeltaParams& p, int head_dim, hipStream_t stream);
 
 struct DecodeParams {
-    const void* q;          // bf16 (B, H, 1, D)
-    const void* k;          // bf16 (B, H, N, D)
-    const void* v;          // bf16 (B, H, N, D)
-    float* out;             // fp32 (S, B, H, 1, D) per-chunk partials
-    float* lse;             // fp32 (S, B, H, 1, 1)
-    int b, h;
+    const void* q;          // bf16 (B, HQ, NQ, D): NQ query tokens per head
+                            // (speculative / tree-decode heads)
+    const void* k;          // bf16 (B, HK, N, D)
+    const void* v;          // bf16 (B, HK, N, D)
+    float* out;             // fp32 (S, B, HQ, NQ, D) per-chunk partials
+    float* lse;             // fp32 (S, B, HQ, NQ, 1)
+    int b, h;               // h = HQ (query heads)
+    int hk;                 // kv heads; q head qh reads kv head qh % hk
+    int nq;                 // query tokens per head (>= 1)
     long n;
     float scale;
     long chunks;            // kv-split S (0 = auto)

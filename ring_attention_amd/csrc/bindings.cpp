@@ -246,21 +246,24 @@ at::Tensor rotary_apply(at::Tensor x, at::Tensor cos_t, at::Tensor sin_t, double
 
 std::vector<at::Tensor> decode_partial(at::Tensor q, at::Tensor k, at::Tensor v,
                                        double sm_scale) {
-    // q (B,H,1,D); k,v (B,H,N,D) bf16
-    // -> (out fp32 (S,B,H,1,D), lse fp32 (S,B,H,1,1)): S kv-chunk partials,
-    //    merged by the caller with the standard logsumexp combine
+    // q (B,HQ,NQ,D); k,v (B,HK,N,D) bf16 — NQ query tokens (speculative /
+    // tree heads), HQ % HK == 0 (GQA, tile pairing qh % hk)
+    // -> (out fp32 (S,B,HQ,NQ,D), lse fp32 (S,B,HQ,NQ,1)): S kv-chunk
+    //    partials, merged by the caller with the logsumexp combine
     CHECK_BF16_CONTIG(q); CHECK_BF16_CONTIG(k); CHECK_BF16_CONTIG(v);
-    const int64_t B = q.size(0), H = q.size(1), D = q.size(3), N = k.size(2);
+    const int64_t B = q.size(0), H = q.size(1), NQ = q.size(2), D = q.size(3);
+    const int64_t HK = k.size(1), N = k.size(2);
     TORCH_CHECK(D == 64 || D == 128, "head dim must be 64 or 128");
-    int64_t waves = B * H;
+    TORCH_CHECK(H % HK == 0, "q heads must be a multiple of kv heads");
+    int64_t waves = B * H * NQ;
     int64_t chunks = std::max<int64_t>(
         1, std::min<int64_t>(N / 1024 + 1, 1024 / std::max<int64_t>(waves / 4, 1)));
-    auto out = at::empty({chunks, B, H, 1, D}, q.options().dtype(at::kFloat));
-    auto lse = at::empty({chunks, B, H, 1, 1}, q.options().dtype(at::kFloat));
+    auto out = at::empty({chunks, B, H, NQ, D}, q.options().dtype(at::kFloat));
+    auto lse = at::empty({chunks, B, H, NQ, 1}, q.options().dtype(at::kFloat));
     DecodeParams p{};
     p.q = q.data_ptr(); p.k = k.data_ptr(); p.v = v.data_ptr();
     p.out = out.data_ptr<float>(); p.lse = lse.data_ptr<float>();
-    p.b = (int)B; p.h = (int)H; p.n = N;
+    p.b = (int)B; p.h = (int)H; p.hk = (int)HK; p.nq = (int)NQ; p.n = N;
     p.scale = sm_scale > 0 ? (float)sm_scale : (float)(1.0 / std::sqrt((double)D));
     p.chunks = chunks;
     launch_decode_partial(p, (int)D, at::hip::getCurrentHIPStream());

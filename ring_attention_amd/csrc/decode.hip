@@ -15,23 +15,30 @@ namespace ring_attn {
 
 template <int D>
 __global__ __launch_bounds__(256) void decode_partial_kernel(DecodeParams p) {
-    // one wave per (b, h); 4 waves per block; blockIdx.y splits the KV range
-    // into `chunks` partials (merged on the host) so long sequences use the
-    // whole chip instead of b*h waves
+    // one wave per (b, q-head, query token); 4 waves per block; blockIdx.y
+    // splits the KV range into `chunks` partials (merged on the host) so
+    // long sequences use the whole chip.  Multi-query (speculative / tree
+    // heads) and GQA share the kv stream through L2/L3: the extra waves of
+    // one kv head re-read lines the first wave just fetched, so >= 2
+    // queries/step cost far less than 2x one query.
+    const int rows = p.b * p.h * p.nq;
     const int wave_global = (blockIdx.x * 4) + (threadIdx.x >> 6);
-    if (wave_global >= p.b * p.h) return;
-    const int b = wave_global / p.h;
-    const int h = wave_global % p.h;
+    if (wave_global >= rows) return;
+    const int iq = wave_global % p.nq;
+    const int bh = wave_global / p.nq;
+    const int b = bh / p.h;
+    const int h = bh % p.h;
+    const int hk = h % p.hk;           // reference tile GQA pairing
     const int lane = threadIdx.x & 63;
     const int chunk = blockIdx.y;
     const long per = (p.n + gridDim.y - 1) / gridDim.y;
     const long j_lo = chunk * per;
     const long j_hi = min(p.n, j_lo + per);
-    const long part_off = (long)chunk * p.b * p.h;
+    const long part_off = (long)chunk * rows;
 
-    const __bf16* qp = (const __bf16*)p.q + ((long)b * p.h + h) * D;
-    const __bf16* kp = (const __bf16*)p.k + ((long)b * p.h + h) * p.n * D;
-    const __bf16* vp = (const __bf16*)p.v + ((long)b * p.h + h) * p.n * D;
+    const __bf16* qp = (const __bf16*)p.q + (((long)b * p.h + h) * p.nq + iq) * D;
+    const __bf16* kp = (const __bf16*)p.k + ((long)b * p.hk + hk) * p.n * D;
+    const __bf16* vp = (const __bf16*)p.v + ((long)b * p.hk + hk) * p.n * D;
 
     // q in registers (fp32), replicated per lane as needed
     float qreg[D];
@@ -87,15 +94,15 @@ __global__ __launch_bounds__(256) void decode_partial_kernel(DecodeParams p) {
     float l_safe = fmaxf(l, 1e-38f);
     if (lane == 0) {
         float inv = 1.f / l_safe;
-        float* op = p.out + (part_off + (long)b * p.h + h) * D;
+        float* op = p.out + (part_off + (long)bh * p.nq + iq) * D;
         #pragma unroll
         for (int d = 0; d < D; ++d) op[d] = acc[d] * inv;
-        p.lse[part_off + (long)b * p.h + h] = __logf(l_safe) + m;
+        p.lse[part_off + (long)bh * p.nq + iq] = __logf(l_safe) + m;
     }
 }
 
 void launch_decode_partial(const DecodeParams& p, int head_dim, hipStream_t stream) {
-    int waves = p.b * p.h;
+    int waves = p.b * p.h * p.nq;
     // enough chunks to put ~1 wave per CU at this (b*h): each wave streams
     // its KV slice; host merges the per-chunk partials
     long target = 1024;

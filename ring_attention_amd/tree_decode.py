@@ -9,6 +9,11 @@ packed into ONE RCCL all-reduce on a fused [den | num] buffer — 2 collective
 rounds per decoded token instead of 3.
 
 Layout parity with the reference: q (b, h, 1, d); k, v (b, h, n, dv).
+Generalized beyond the reference (tree_attn_decoding.py:54-79 handles one
+query per head, no GQA): q may carry NQ query tokens per head (speculative /
+tree-decode heads) as (b, h, nq, d), and k/v may have fewer heads
+(b, hk, n, dv) with the tile GQA pairing qh % hk — the HIP kernel shares the
+kv stream across the extra queries through L2/L3.
 """
 
 from __future__ import annotations
@@ -21,7 +26,7 @@ from .parallel import get_rank, get_world_size, is_distributed
 
 
 def _local_decode_partial(q: Tensor, k: Tensor, v: Tensor) -> tuple[Tensor, Tensor]:
-    """Local flash-decode partial: returns (out fp32 (b,h,1,dv), lse fp32 (b,h,1,1))."""
+    """Local flash-decode partial: (out fp32 (b,h,nq,dv), lse fp32 (b,h,nq,1))."""
     if q.is_cuda:
         from .ops import hip_ext
         if hip_ext.available():
@@ -48,10 +53,15 @@ def _local_decode_partial(q: Tensor, k: Tensor, v: Tensor) -> tuple[Tensor, Tens
             lse = den.log() + m
             return out, lse
     scale = q.shape[-1] ** -0.5
-    sim = torch.einsum("bhid,bhjd->bhij", q.float(), k.float()) * scale
+    kf, vf = k.float(), v.float()
+    groups = q.shape[1] // k.shape[1]
+    if groups > 1:                      # tile GQA pairing (qh % hk)
+        kf = kf.repeat(1, groups, 1, 1)
+        vf = vf.repeat(1, groups, 1, 1)
+    sim = torch.einsum("bhid,bhjd->bhij", q.float(), kf) * scale
     lse = sim.logsumexp(dim=-1, keepdim=True)
     attn = torch.softmax(sim, dim=-1)
-    out = torch.einsum("bhij,bhjd->bhid", attn, v.float())
+    out = torch.einsum("bhij,bhjd->bhid", attn, vf)
     return out, lse
 
 
@@ -66,7 +76,7 @@ def tree_attn_decode(
 ) -> Tensor:
     assert (k is None) == (v is None)
     dtype = q.dtype
-    b, h, one, d = q.shape
+    b, h, nq, d = q.shape
 
     # capture the value dim from the FULL v before chunking: a rank whose
     # shard is empty must still pack a (b,h,1,1+dv)-shaped all-reduce buffer,
@@ -84,8 +94,8 @@ def tree_attn_decode(
         local_out, lse = _local_decode_partial(q, k, v)
     else:
         # seq shorter than world: this rank holds nothing
-        local_out = q.new_zeros((b, h, one, dim_v), dtype=torch.float32)
-        lse = torch.full((b, h, one, 1), -torch.finfo(torch.float32).max,
+        local_out = q.new_zeros((b, h, nq, dim_v), dtype=torch.float32)
+        lse = torch.full((b, h, nq, 1), -torch.finfo(torch.float32).max,
                          device=q.device, dtype=torch.float32)
 
     if not is_distributed():
@@ -96,8 +106,8 @@ def tree_attn_decode(
     dist.all_reduce(max_lse, dist.ReduceOp.MAX)
 
     # round 2: ONE summed all-reduce over the packed [den | num] buffer
-    den = (lse - max_lse).exp()                         # (b,h,1,1)
-    packed = torch.cat((den, local_out * den), dim=-1)  # (b,h,1,1+dv)
+    den = (lse - max_lse).exp()                         # (b,h,nq,1)
+    packed = torch.cat((den, local_out * den), dim=-1)  # (b,h,nq,1+dv)
     dist.all_reduce(packed)
     den_sum, num_sum = packed[..., :1], packed[..., 1:]
 

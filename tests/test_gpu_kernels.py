@@ -528,3 +528,25 @@ def test_flash_attn_strict_diagonal():
     err = (out.float().cpu()[:, 1:] - ref[:, 1:]).abs().max().item()
     assert err < 2e-2
     assert out[:, 0].abs().max().item() < 1e-6
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("nq,groups", [(1, 1), (4, 1), (2, 4), (4, 2)])
+def test_decode_partial_multiquery_gqa(nq, groups):
+    # HIP decode kernel: NQ query tokens per head + GQA (kv stream shared
+    # via cache), exact vs eager
+    from ring_attention_amd.tree_decode import tree_attn_decode
+    b, h, n, d = 2, 8, 2048, 64
+    hk = h // groups
+    torch.manual_seed(41)
+    q = torch.randn(b, h, nq, d, device="cuda", dtype=torch.bfloat16)
+    k = torch.randn(b, hk, n, d, device="cuda", dtype=torch.bfloat16)
+    v = torch.randn(b, hk, n, d, device="cuda", dtype=torch.bfloat16)
+    out = tree_attn_decode(q, k, v, shard_kv_seq=False)
+    qc, kc, vc = q.float().cpu(), k.float().cpu(), v.float().cpu()
+    kc = kc.repeat(1, groups, 1, 1)
+    vc = vc.repeat(1, groups, 1, 1)
+    sim = torch.einsum("bhid,bhjd->bhij", qc, kc) * d ** -0.5
+    ref = torch.einsum("bhij,bhjd->bhid", sim.softmax(-1), vc)
+    err = (out.float().cpu() - ref).abs().max().item()
+    assert err < 3e-3, f"decode err {err}"

@@ -203,3 +203,29 @@ def _tree_case_dv(rank, world, seq_len, dv):
 def test_tree_decode_dv_mismatch_world4_short_seq():
     # empty-shard ranks must size the packed buffer from v's dv
     run_distributed(4, _tree_case_dv, 3, 24)
+
+
+def _tree_case_multiq_gqa(rank, world, seq_len, nq, groups):
+    # generalized decode: nq query tokens per head + GQA kv heads
+    torch.manual_seed(23)
+    b, h, d = 1, 4, 16
+    hk = h // groups
+    q = torch.randn(b, h, nq, d)
+    k = torch.randn(b, hk, seq_len, d)
+    v = torch.randn(b, hk, seq_len, d)
+    out = tree_attn_decode(q, k, v, shard_kv_seq=True)
+    kk = k.repeat(1, groups, 1, 1)
+    vv = v.repeat(1, groups, 1, 1)
+    sim = torch.einsum("bhid,bhjd->bhij", q, kk) * d ** -0.5
+    ref = torch.einsum("bhij,bhjd->bhid", sim.softmax(-1), vv)
+    err = (out - ref).abs().max().item()
+    assert err < 1e-5, f"tree decode multiq/gqa err {err}"
+    return err
+
+
+def test_tree_decode_multiquery_world2():
+    run_distributed(2, _tree_case_multiq_gqa, 64, 4, 1)
+
+
+def test_tree_decode_gqa_world2():
+    run_distributed(2, _tree_case_multiq_gqa, 64, 2, 2)
