@@ -98,14 +98,19 @@ __device__ __forceinline__ void stage_transposed(
 // ---------------------------------------------------------------------------
 static constexpr int DQ_WAVES = 8;
 static constexpr int DQ_QROWS_WG = DQ_WAVES * 32;     // 256
-template <int D> constexpr int dq_kvblk() { return D == 64 ? 128 : 64; }  // LDS budget
+// The kt image is gone (K^T fragments come from the row-major k image via
+// ds_read_b64_tr_b16 — the v2 forward's V recipe, hardware-verified in
+// tools/hw_probe.hip), which frees a third of the LDS and the transpose
+// staging.  d128 stays at KVB=64: the KVB=128 variant its freed LDS allows
+// spills 280-336 B/lane into the hot loop and measured SLOWER (174 vs 204
+// TF headline) — the round-1 "fwd KVBLK=128 at d128" trap again.
+template <int D> constexpr int dq_kvblk() { return D == 64 ? 128 : 64; }
 
 template <int D>
 struct DqLds {
     static constexpr int KVB = dq_kvblk<D>();
     __align__(16) __bf16 k[2][KVB * D];    // [kv][d] swizzled
     __align__(16) __bf16 v[2][KVB * D];    // [kv][d] swizzled
-    __align__(16) __bf16 kt[2][D * KVB];   // [d][kv] swizzled (pair-staged)
     unsigned char kmask[2][KVB];
 };
 
@@ -194,8 +199,6 @@ void attn_bwd_dq_kernel(BwdParams p) { // cap 256 VGPR, stop the 220 B/thread
     constexpr int CH = D * 2 / 16;
     constexpr int KCHUNKS = DQ_KVBLK * CH;
     constexpr int KREGS = (KCHUNKS + 511) / 512;
-    constexpr int TPAIRS = (DQ_KVBLK / 2) * (D / 8);
-    constexpr int TREGS = (TPAIRS + 511) / 512;
     const __bf16* kbase = (const __bf16*)p.k + ((long)b * p.nk) * p.hk * D + (long)hk * D;
     const __bf16* vbase = (const __bf16*)p.v + ((long)b * p.nk) * p.hk * D + (long)hk * D;
     const unsigned char* mbase = p.kmask ? (const unsigned char*)p.kmask + (long)b * p.nk : nullptr;
@@ -206,12 +209,9 @@ void attn_bwd_dq_kernel(BwdParams p) { // cap 256 VGPR, stop the 220 B/thread
         + (tid / CH) * kv_row_stride + (tid % CH) * 8;
     const __bf16* vptr = vbase + (long)t_lo * tile_stride
         + (tid / CH) * kv_row_stride + (tid % CH) * 8;
-    const __bf16* ktpa = kbase + (long)t_lo * tile_stride
-        + ((tid % (DQ_KVBLK / 2)) * 2) * kv_row_stride + (tid / (DQ_KVBLK / 2)) * 8;
     long j0_next = (long)t_lo * DQ_KVBLK;
 
     uint4 kst[KREGS], vst[KREGS];
-    bf16x8 kta_st[TREGS], ktb_st[TREGS];
     unsigned char mst = 1;
 
     auto load_tile = [&]() {
@@ -228,19 +228,9 @@ void attn_bwd_dq_kernel(BwdParams p) { // cap 256 VGPR, stop the 220 B/thread
                 vst[r] = okr ? *(const uint4*)(vptr + off) : uint4{0, 0, 0, 0};
             }
         }
-        #pragma unroll
-        for (int r = 0; r < TREGS; ++r) {
-            int c = tid + r * 512;
-            if (c < TPAIRS) {
-                const __bf16* sa = ktpa + r * (512 / (DQ_KVBLK / 2)) * 8;
-                long ja = j0 + (c % (DQ_KVBLK / 2)) * 2;
-                kta_st[r] = (full || ja <= jmax) ? *(const bf16x8*)sa : bf16x8{};
-                ktb_st[r] = (full || ja + 1 <= jmax) ? *(const bf16x8*)(sa + kv_row_stride) : bf16x8{};
-            }
-        }
         if (mbase && tid < DQ_KVBLK)
             mst = (j0 + tid <= jmax) ? mbase[j0 + tid] : 0;
-        kptr += tile_stride; vptr += tile_stride; ktpa += tile_stride;
+        kptr += tile_stride; vptr += tile_stride;
         j0_next += DQ_KVBLK;
     };
 
@@ -252,21 +242,6 @@ void attn_bwd_dq_kernel(BwdParams p) { // cap 256 VGPR, stop the 220 B/thread
                 int row = c / CH, ch = c % CH;
                 *(uint4*)(lds.k[par] + row * D + bswz<D / 8>(row, ch) * 8) = kst[r];
                 *(uint4*)(lds.v[par] + row * D + bswz<D / 8>(row, ch) * 8) = vst[r];
-            }
-        }
-        #pragma unroll
-        for (int r = 0; r < TREGS; ++r) {
-            int c = tid + r * 512;
-            if (c < TPAIRS) {
-                int jp = c % (DQ_KVBLK / 2);
-                int d0 = (c / (DQ_KVBLK / 2)) * 8;
-                #pragma unroll
-                for (int e = 0; e < 8; ++e) {
-                    int dd = d0 + e;
-                    int byte_off = dd * DQ_KVBLK * 2 + ((jp * 4) ^ ((dd & 7) << 4));
-                    __bf16 pr[2] = {kta_st[r][e], ktb_st[r][e]};
-                    *(uint32_t*)((char*)lds.kt[par] + byte_off) = *(uint32_t*)pr;
-                }
             }
         }
         if (mbase && tid < DQ_KVBLK) lds.kmask[par][tid] = mst;
@@ -389,21 +364,54 @@ void attn_bwd_dq_kernel(BwdParams p) { // cap 256 VGPR, stop the 220 B/thread
                     frag[kb * 2 + half][c + 2] = r[1];
                 }
 
-        // dq^T[d][q] += K^T[d][kv] x ds^T[kv][q]
-        __builtin_amdgcn_s_setprio(1);
-        #pragma unroll
-        for (int db = 0; db < DBLK; ++db) {
-            int drow = db * 32 + l31;
+        // dq^T[d][q] += K^T[d][kv] x ds^T[kv][q].  K^T A-fragments come
+        // straight from the row-major swizzled k image via
+        // ds_read_b64_tr_b16: lane f=l&15 of each 16-lane group fetches
+        // K[kvq + (f>>2)][dg + 4*(f&3) ..+4] and the hardware
+        // redistribution delivers out[l][j] = K[kvq + j][dg + (l&15)] —
+        // i.e. lane l31 holds d column db*32 + l31, 4 kv per read.
+        {
+            const int f15 = lane & 15;
+            const int g16 = (lane >> 4) & 1;
+            const unsigned kb_lds = (unsigned)(uintptr_t)
+                (__attribute__((address_space(3))) __bf16*)lds.k[par];
+            __builtin_amdgcn_s_setprio(1);
+            // issue ALL K^T tr-reads up front (asm loads cannot be software-
+            // pipelined by the scheduler, so one batched wait beats a wait
+            // per d-block), then run the whole MFMA stream
+            unsigned long long ktr[DBLK][DQ_NBLK * 2][2];
             #pragma unroll
-            for (int ks = 0; ks < DQ_NBLK * 2; ++ks) {
-                int chunk = ks * 2 + lhi;
-                bf16x8 ktf = *(const bf16x8*)(lds.kt[par] + drow * DQ_KVBLK +
-                                              bswz<DQ_KVBLK / 8>(drow, chunk) * 8);
-                dq_acc[db] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
-                    ktf, *(const bf16x8*)frag[ks], dq_acc[db], 0, 0, 0);
+            for (int db = 0; db < DBLK; ++db) {
+                const int dg = db * 32 + g16 * 16 + 4 * (f15 & 3);
+                #pragma unroll
+                for (int ks = 0; ks < DQ_NBLK * 2; ++ks) {
+                    #pragma unroll
+                    for (int half = 0; half < 2; ++half) {
+                        const int kv = ks * 16 + 8 * lhi + 4 * half + (f15 >> 2);
+                        // swizzled element address of (kv, dg): 16B chunk
+                        // c = dg/8 XORed with the row's swizzle key
+                        const int c = (dg / 8) ^ (kv & (CH < 8 ? CH - 1 : 7));
+                        const unsigned a = kb_lds
+                            + (unsigned)(kv * (D * 2) + c * 16 + (dg % 8) * 2);
+                        asm volatile("ds_read_b64_tr_b16 %0, %1"
+                                     : "=&v"(ktr[db][ks][half]) : "v"(a) : "memory");
+                    }
+                }
             }
+            asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+            __builtin_amdgcn_sched_barrier(0);
+            #pragma unroll
+            for (int db = 0; db < DBLK; ++db)
+                #pragma unroll
+                for (int ks = 0; ks < DQ_NBLK * 2; ++ks) {
+                    bf16x8 ktf;
+                    *(unsigned long long*)&ktf = ktr[db][ks][0];
+                    *((unsigned long long*)&ktf + 1) = ktr[db][ks][1];
+                    dq_acc[db] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+                        ktf, *(const bf16x8*)frag[ks], dq_acc[db], 0, 0, 0);
+                }
+            __builtin_amdgcn_s_setprio(0);
         }
-        __builtin_amdgcn_s_setprio(0);
     }
 
     if (!row_valid) continue;
