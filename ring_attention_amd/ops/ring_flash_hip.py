@@ -128,7 +128,7 @@ def _hop_geometry(rq: int, rk: int, n: int, ring_size: int, striped: bool,
 
 
 def _causal_balance_split(causal, lookback, diag_cuts, grid_wgs):
-    """Causal load-balance hook (now a no-op).
+    """Causal load-balance grid.z factor (1 except one measured case).
 
     Round-1 history: a 2-way grid.z split here gave +19%/+8% at 16k/32k
     causal, but it is SUPERSEDED by in-kernel paired-tile scheduling — the
