@@ -452,7 +452,7 @@ class RingFlashAttentionHIPFunction(Function):
             dk_home = own[:half].view(b, hk, n, d).permute(0, 2, 1, 3).contiguous()
             dv_home = own[half:].view(b, hk, d, n).permute(0, 3, 1, 2).contiguous()
             return (dq.to(in_dtype), dk_home.to(in_dtype), dv_home.to(in_dtype),
-                    None, None, None, None, None, None, None, None, None)
+                    None, None, None, None, None, None, None, None, None, None)
         # independent grid.z splits: dq's grid is (q-tiles x b*h), dkv's is
         # (kv-tiles x b*hk) — GQA shrinks the latter (e.g. hk=2 -> 64 WGs)
         qtiles = (n + 255) // 256
