@@ -9,20 +9,26 @@ path, never a silent eager fallback.
 from __future__ import annotations
 
 
+import threading
+
 _ext = None
 _tried = False
+_lock = threading.Lock()
 
 
 def _load():
     global _ext, _tried
     if _tried:
         return _ext
-    _tried = True
-    try:
-        import importlib
-        _ext = importlib.import_module("ring_attention_amd._ring_attn_hip")
-    except ImportError:
-        _ext = None
+    with _lock:        # threaded first use (loopback tests): import once
+        if _tried:
+            return _ext
+        try:
+            import importlib
+            _ext = importlib.import_module("ring_attention_amd._ring_attn_hip")
+        except ImportError:
+            _ext = None
+        _tried = True
     return _ext
 
 

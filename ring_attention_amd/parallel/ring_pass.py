@@ -60,17 +60,25 @@ def ring_pass(topo: RingTopology, *tensors: Tensor, num_hops: int = 1) -> tuple[
     if topo.ring_size == 1 or not is_distributed():
         return tuple(tensors)
     from ..utils.tracing import GLOBAL_RING_STATS, trace_range
+    # NEVER recv into the caller's tensors: ping-pong between two INTERNAL
+    # buffers (reusing the inputs as receive targets silently mutated them
+    # from hop 2 on — the world>=3 saved-KV corruption found in round 2)
     cur = [t.contiguous() for t in tensors]
-    bufs = [torch.empty_like(t) for t in cur]
+    recv = [torch.empty_like(t) for t in cur]
+    spare = None
     nbytes = sum(t.element_size() * t.numel() for t in cur)
     GLOBAL_RING_STATS.start()
     hops_done = 0
     with trace_range("ring_pass.multi_hop"):
-        for _ in range(num_hops % topo.ring_size):  # ring_size hops = identity
-            reqs = _exchange(topo, cur, bufs)
+        for hop in range(num_hops % topo.ring_size):  # ring_size hops = identity
+            reqs = _exchange(topo, cur, recv)
             for r in reqs:
                 r.wait()
-            cur, bufs = bufs, cur
+            if hop == 0:
+                spare = [torch.empty_like(t) for t in cur]  # caller's cur retired
+            else:
+                spare = cur
+            cur, recv = recv, spare
             hops_done += 1
     GLOBAL_RING_STATS.stop(hops_done, hops_done * nbytes)
     return tuple(cur)
@@ -97,8 +105,15 @@ def all_ring_pass(
         return
 
     from ..utils.tracing import GLOBAL_RING_STATS, trace_range
+    # Buffer discipline: the caller's tensors are SEND-only.  Receives go to
+    # internal buffers that ping-pong between two sets allocated here; the
+    # input tensors are never written (reusing them as receive targets from
+    # hop 2 on silently mutated the caller's K/V — the world>=3 corruption
+    # of saved forward tensors found in round 2; world 2 never re-used them,
+    # which is why every 2-rank test passed).
     cur = [t.contiguous() for t in tensors]
-    bufs = [torch.empty_like(t) for t in cur]
+    recv = [torch.empty_like(t) for t in cur]
+    spare = None
     reqs = None
     nbytes = sum(t.element_size() * t.numel() for t in cur)
 
@@ -107,7 +122,7 @@ def all_ring_pass(
         if not is_last:
             # post the exchange for the NEXT hop now; compute on `cur` overlaps it
             with trace_range(f"ring_pass.post_hop{hop + 1}"):
-                reqs = _exchange(topo, cur, bufs)
+                reqs = _exchange(topo, cur, recv)
             GLOBAL_RING_STATS.start()
 
         yield RingInfo(hop, topo.source_of_hop(hop), hop == 0, is_last), tuple(cur)
@@ -117,7 +132,11 @@ def all_ring_pass(
                 for r in reqs:
                     r.wait()
             GLOBAL_RING_STATS.stop(1, nbytes)
-            cur, bufs = bufs, cur
+            if hop == 0:
+                spare = [torch.empty_like(t) for t in cur]  # caller's cur retired
+            else:
+                spare = cur
+            cur, recv = recv, spare
 
 
 def null_ring_pass(*tensors: Tensor) -> Iterator[tuple[RingInfo, tuple[Tensor, ...]]]:

@@ -422,3 +422,46 @@ def test_loopback_bench_shaped_step():
             r = outs["ring"][rank][i].float()
             e = (a - r).abs().max().item() / (r.abs().max().item() + 1e-6)
             assert e < 2e-2, f"rank {rank} {name} strategy mismatch {e}"
+
+
+def _ring_immutability_case(rank, world):
+    """all_ring_pass/ring_pass must never write the caller's tensors.
+
+    Regression for the round-2 world>=3 corruption: receive buffers used to
+    ping-pong with the caller's own tensors, so from hop 2 on the ring WROTE
+    INTO the inputs — the HIP forward's saved K/V then held another rank's
+    shard and every backward gradient followed the wrong shard (world 2
+    never re-used the input as a buffer, masking the bug in 2-rank tests).
+    """
+    from ring_attention_amd.parallel import (RingAccumulator, RingTopology,
+                                             all_ring_pass, ring_pass)
+    topo = RingTopology(world)
+    kb = torch.full((8,), float(rank))
+    vb = torch.full((8,), 100.0 + rank)
+    for info, tensors in all_ring_pass(topo, kb, vb, max_hops=world):
+        assert tensors[0][0].item() == topo.source_of_hop(info.hop)
+        assert tensors[1][0].item() == 100 + topo.source_of_hop(info.hop)
+    assert kb[0].item() == rank, f"all_ring_pass mutated its input: {kb[0]}"
+    assert vb[0].item() == 100 + rank
+    # a SECOND walk over the same tensors must see the same shards
+    acc = RingAccumulator(topo)
+    for info, tensors in all_ring_pass(topo, kb, vb, max_hops=world):
+        assert tensors[0][0].item() == topo.source_of_hop(info.hop)
+        c = torch.zeros(world, world)
+        c[rank, int(tensors[0][0].item())] = 1
+        acc.step(c, info.is_last)
+    home = acc.finish(world)
+    exp = torch.zeros(world, world)
+    exp[:, rank] = 1
+    assert torch.equal(home, exp), f"homecoming routed wrong: {home.sum(0)}"
+    t = torch.full((4,), float(rank))
+    (moved,) = ring_pass(topo, t, num_hops=3)
+    assert t[0].item() == rank, "ring_pass mutated its input"
+    assert moved[0].item() == (rank - 3) % world
+    return True
+
+
+@pytest.mark.parametrize("world", [3, 4, 8])
+def test_ring_engine_input_immutability(world):
+    from .loopback_dist import loopback_world
+    assert all(loopback_world(world, lambda r: _ring_immutability_case(r, world)))
