@@ -291,13 +291,17 @@ class RingFlashAttentionHIPFunction(Function):
                 m_part = torch.empty(kv_split, b, h, n, device=q.device, dtype=torch.float32)
                 l_part = torch.empty(kv_split, b, h, n, device=q.device, dtype=torch.float32)
                 ext.attn_fwd(qb, k_full, v_full, m_full, o_part, m_part, l_part,
-                             None, None, scale, causal, diag, q_stride, 0, False,
+                             None, None, scale, causal, diag, q_stride,
+                             lookback if lookback is not None else 0,
+                             lookback is not None,
                              softclamp_qk_sim, softclamp_value, True, True, kv_split, 0, None)
                 ext.attn_fwd_merge(o_part, m_part, l_part, None, None, None,
                                    out, lse, kv_split, b, h, d, n, True, True)
             else:
                 ext.attn_fwd(qb, k_full, v_full, m_full, None, None, None, out, lse,
-                             scale, causal, diag, q_stride, 0, False,
+                             scale, causal, diag, q_stride,
+                             lookback if lookback is not None else 0,
+                             lookback is not None,
                              softclamp_qk_sim, softclamp_value, True, True, 1, 0, None)
             ctx.save_for_backward(qb, kb, vb, out, lse,
                                   mask_u8 if mask_u8 is not None else torch.empty(0))
@@ -437,11 +441,15 @@ class RingFlashAttentionHIPFunction(Function):
             dv_full = fac(b, hk, d, n_total, device=qb.device, dtype=torch.float32)
             ext.attn_bwd(qb, k_full, v_full, dob, m_full, lse, delta,
                          dq, dk_full, dv_full, scale, causal, diag, q_stride,
-                         0, False, softclamp_qk_sim, softclamp_value, False, split_dq, 1,
+                         lookback if lookback is not None else 0,
+                         lookback is not None,
+                         softclamp_qk_sim, softclamp_value, False, split_dq, 1,
                          ddq, None)
             ext.attn_bwd(qb, k_full, v_full, dob, m_full, lse, delta,
                          dq, dk_full, dv_full, scale, causal, diag, q_stride,
-                         0, False, softclamp_qk_sim, softclamp_value, False, split_dkv, 2,
+                         lookback if lookback is not None else 0,
+                         lookback is not None,
+                         softclamp_qk_sim, softclamp_value, False, split_dkv, 2,
                          None, ddkv)
             # ONE reduce-scatter returns each rank's dk/dv shard (summed)
             dk_chunks = _scatter_chunks_of_global(dk_full, R, striped, dim=2)
