@@ -158,3 +158,34 @@ def test_walk_descriptor_coverage():
                         got[tile] = (lo, hi)
                 assert got == expect, (kind, d, nq, nk, diag, stride)
     _DESC_CACHE.clear()
+
+
+def test_head_dim_padding_helpers():
+    """kernel_head_dim / _pad_head_dim algebra (the GPU parity tests cover
+    the kernels; this pins the pure-python mapping)."""
+    import torch
+    import torch.nn.functional as F
+    from ring_attention_amd.ops.ring_flash_hip import kernel_head_dim, _pad_head_dim
+    from ring_attention_amd.ops.reference import default_attention
+
+    assert [kernel_head_dim(d) for d in (8, 32, 33, 64, 65, 96, 128)] == \
+        [32, 32, 64, 64, 128, 128, 128]
+    import pytest
+    with pytest.raises(ValueError):
+        kernel_head_dim(129)
+
+    torch.manual_seed(0)
+    b, n, h, d = 1, 64, 2, 40
+    q = torch.randn(b, n, h, d)
+    k = torch.randn_like(q)
+    v = torch.randn_like(q)
+    qp, kp, vp = _pad_head_dim(q, k, v)
+    assert qp.shape[-1] == 64 and torch.equal(qp[..., :d], q)
+    assert torch.equal(kp[..., d:], torch.zeros_like(kp[..., d:]))
+    # padded attention with the true scale == unpadded attention
+    ref = default_attention(q, k, v, causal=True)
+    sim = torch.einsum("bihd,bjhd->bhij", qp.float(), kp.float()) * d ** -0.5
+    pos = torch.arange(n)
+    sim = sim.masked_fill((pos[None, :] > pos[:, None])[None, None], float("-inf"))
+    out = torch.einsum("bhij,bjhd->bihd", sim.softmax(-1), vp.float())[..., :d]
+    assert (out - ref).abs().max().item() < 1e-5
