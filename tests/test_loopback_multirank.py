@@ -228,11 +228,13 @@ def test_loopback_tree_decode():
 
 
 @pytest.mark.gpu
-def test_loopback_transformer_gpu():
-    """Full RingTransformer (HIP kernels, rotary, sharded CE) at world 2 on
-    one GPU via the loopback layer, against the replicated non-ring twin."""
+@pytest.mark.parametrize("world", [2, 4])
+def test_loopback_transformer_gpu(world):
+    """Full RingTransformer (HIP kernels, rotary, sharded CE) at world 2/4
+    on one GPU via the loopback layer, against the replicated non-ring twin
+    (world 4 exercises the multi-hop transport the round-2 fix guards)."""
     from ring_attention_amd import RingTransformer
-    world, seq = 2, 1024
+    seq = 512 * world
     model_kwargs = dict(
         num_tokens=256, dim=256, depth=2, causal=True, dim_head=64, heads=4,
         ff_mult=2, num_grouped_query_heads=2, bucket_size=256,
