@@ -24,11 +24,22 @@ __global__ __launch_bounds__(256) void decode_partial_kernel(DecodeParams p) {
     const int rows = p.b * p.h * p.nq;
     const int wave_global = (blockIdx.x * 4) + (threadIdx.x >> 6);
     if (wave_global >= rows) return;
+    // group-major wave->row mapping: consecutive waves (the 4 of a block,
+    // which land on ONE CU and one XCD L2) cover the query tokens and then
+    // the GQA group-mates of the SAME kv head, so their shared kv stream is
+    // fetched once per L2.  The natural (b,h,iq) order instead put 4
+    // DIFFERENT kv heads in each block and scattered a group's readers
+    // across the 8 XCDs' private L2s — each re-fetched the stream from HBM
+    // (measured: 32q/4kv nq=1 decode at 0.27 TB/s effective).  Output
+    // layout is unchanged; only the wave->row assignment differs.
+    const int G = p.h / p.hk;
     const int iq = wave_global % p.nq;
-    const int bh = wave_global / p.nq;
-    const int b = bh / p.h;
-    const int h = bh % p.h;
-    const int hk = h % p.hk;           // reference tile GQA pairing
+    int r = wave_global / p.nq;
+    const int g = r % G;  r /= G;
+    const int hk = r % p.hk;           // reference tile GQA pairing
+    const int b = r / p.hk;
+    const int h = hk + g * p.hk;       // group-mate g of kv head hk
+    const int bh = b * p.h + h;
     const int lane = threadIdx.x & 63;
     const int chunk = blockIdx.y;
     const long per = (p.n + gridDim.y - 1) / gridDim.y;
