@@ -550,3 +550,27 @@ def test_decode_partial_multiquery_gqa(nq, groups):
     ref = torch.einsum("bhij,bhjd->bhid", sim.softmax(-1), vc)
     err = (out.float().cpu() - ref).abs().max().item()
     assert err < 3e-3, f"decode err {err}"
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("h,hk,nq", [(6, 3, 3), (6, 2, 1), (12, 4, 5), (8, 8, 3)])
+def test_decode_partial_odd_shapes(h, hk, nq):
+    # guards the group-major wave->row decomposition (iq / group-mate / kv
+    # head / batch unpacking) at group sizes and wave counts that don't
+    # align to the 4-wave block: odd groups, nq*G < 4, partial last block,
+    # n not a multiple of the lane stride
+    from ring_attention_amd.tree_decode import tree_attn_decode
+    b, n, d = 2, 1000, 64
+    groups = h // hk
+    torch.manual_seed(43)
+    q = torch.randn(b, h, nq, d, device="cuda", dtype=torch.bfloat16)
+    k = torch.randn(b, hk, n, d, device="cuda", dtype=torch.bfloat16)
+    v = torch.randn(b, hk, n, d, device="cuda", dtype=torch.bfloat16)
+    out = tree_attn_decode(q, k, v, shard_kv_seq=False)
+    qc, kc, vc = q.float().cpu(), k.float().cpu(), v.float().cpu()
+    kc = kc.repeat(1, groups, 1, 1)
+    vc = vc.repeat(1, groups, 1, 1)
+    sim = torch.einsum("bhid,bhjd->bhij", qc, kc) * d ** -0.5
+    ref = torch.einsum("bhij,bhjd->bhid", sim.softmax(-1), vc)
+    err = (out.float().cpu() - ref).abs().max().item()
+    assert err < 3e-3, f"decode err {err}"
