@@ -90,6 +90,9 @@ struct FwdLds {
 static constexpr float LOG2E = 1.4426950408889634f;
 static constexpr float LN2 = 0.6931471805599453f;
 
+template <class F>
+__device__ __attribute__((noinline)) void fwd_noinline_call(F&& f) { f(); }
+
 // ---------------------------------------------------------------------------
 // forward kernel
 // ---------------------------------------------------------------------------
@@ -126,13 +129,17 @@ __global__ __launch_bounds__(NTHREADS, 1) void attn_fwd_kernel(FwdParams p) {
         __syncthreads();                       // LDS handoff between tiles
     }
 
+    auto fwd_body = [&]() {
+    const FwdParams P = p;   // register-local copy (see dkv_noinline_call):
+                             // the noinline frame would otherwise re-read
+                             // fields through scratch
     const long irow0 = (long)qtile * QROWS_WG + wid * QROWS_WAVE;  // this wave's first q row
     const long i = irow0 + l31;                                     // this lane's q row
-    const bool row_valid = i < p.nq;
+    const bool row_valid = i < P.nq;
     const long i_clamped = row_valid ? i : 0;
 
     // ---- load Q fragments (bf16x8 per k-step): q[b, i, h, ks*16 + lhi*8 .. +8]
-    const __bf16* qbase = (const __bf16*)p.q + ((long)b * p.nq + i_clamped) * p.h * D + (long)h * D;
+    const __bf16* qbase = (const __bf16*)P.q + ((long)b * P.nq + i_clamped) * P.h * D + (long)h * D;
     bf16x8 qf[KSTEPS];
     #pragma unroll
     for (int ks = 0; ks < KSTEPS; ++ks)
@@ -145,20 +152,20 @@ __global__ __launch_bounds__(NTHREADS, 1) void attn_fwd_kernel(FwdParams p) {
     #pragma unroll
     for (int db = 0; db < DBLK; ++db) o_acc[db] = f32x16{};
 
-    const bool split_mode = p.kv_split > 1;
+    const bool split_mode = P.kv_split > 1;
     const int zsplit = blockIdx.z;
-    if (!p.is_first && !split_mode) {   // resume from a previous ring hop
-        const float* mrow = p.m + ((long)b * p.h + h) * p.nq;
-        const float* lrow = p.l + ((long)b * p.h + h) * p.nq;
+    if (!P.is_first && !split_mode) {   // resume from a previous ring hop
+        const float* mrow = P.m + ((long)b * P.h + h) * P.nq;
+        const float* lrow = P.l + ((long)b * P.h + h) * P.nq;
         m_run = mrow[i_clamped] * LOG2E;   // external contract is natural log
         l_run = lrow[i_clamped];
-        const float* oa = p.o_acc + (((long)b * p.h + h) * D) * p.nq;
+        const float* oa = P.o_acc + (((long)b * P.h + h) * D) * P.nq;
         #pragma unroll
         for (int db = 0; db < DBLK; ++db)
             #pragma unroll
             for (int r = 0; r < 16; ++r) {
                 int d = db * 32 + (r & 3) + 8 * (r >> 2) + 4 * lhi;
-                o_acc[db][r] = oa[(long)d * p.nq + i_clamped];
+                o_acc[db][r] = oa[(long)d * P.nq + i_clamped];
             }
     }
 
@@ -167,19 +174,19 @@ __global__ __launch_bounds__(NTHREADS, 1) void attn_fwd_kernel(FwdParams p) {
     // staging pipeline branch-free)
     // q positions in kv-local coordinates: qpos(i) = i * q_stride + diag
     const long wg_i_min = (long)qtile * QROWS_WG;
-    const long wg_i_max = min((long)(qtile + 1) * QROWS_WG, p.nq) - 1;
-    const long wg_q_min = wg_i_min * p.q_stride + p.diag;
-    const long wg_q_max = wg_i_max * p.q_stride + p.diag;
-    const long qpos_i = i * p.q_stride + p.diag;      // this lane's q position
-    const int num_kv_tiles = (int)((p.nk + KVBLK - 1) / KVBLK);
+    const long wg_i_max = min((long)(qtile + 1) * QROWS_WG, P.nq) - 1;
+    const long wg_q_min = wg_i_min * P.q_stride + P.diag;
+    const long wg_q_max = wg_i_max * P.q_stride + P.diag;
+    const long qpos_i = i * P.q_stride + P.diag;      // this lane's q position
+    const int num_kv_tiles = (int)((P.nk + KVBLK - 1) / KVBLK);
 
     int t_lo = 0, t_hi = num_kv_tiles;
-    if (p.causal) {
+    if (P.causal) {
         t_hi = wg_q_max < 0 ? 0 : min((long)num_kv_tiles, wg_q_max / KVBLK + 1);
     }
-    if (p.has_win) {
+    if (P.has_win) {
         // tile t attends iff j0 + KVBLK - 1 >= wg_q_min - win
-        long x = wg_q_min - p.win - KVBLK + 1;
+        long x = wg_q_min - P.win - KVBLK + 1;
         t_lo = x <= 0 ? 0 : (int)((x + KVBLK - 1) / KVBLK);
         if (t_lo > t_hi) t_lo = t_hi;
     }
@@ -187,7 +194,7 @@ __global__ __launch_bounds__(NTHREADS, 1) void attn_fwd_kernel(FwdParams p) {
         // this z's share of THIS WG's valid tile range (fractional split:
         // a global-range split is skewed against the causal trapezoid)
         int valid = t_hi - t_lo;
-        int per_split = (valid + p.kv_split - 1) / p.kv_split;
+        int per_split = (valid + P.kv_split - 1) / P.kv_split;
         int base = t_lo;
         t_lo = base + min(valid, zsplit * per_split);
         t_hi = base + min(valid, (zsplit + 1) * per_split);
@@ -199,9 +206,9 @@ __global__ __launch_bounds__(NTHREADS, 1) void attn_fwd_kernel(FwdParams p) {
     constexpr int KREGS = (KCHUNKS + NTHREADS - 1) / NTHREADS;
     constexpr int VPAIRS = (KVBLK / 2) * (D / 8);
     constexpr int VREGS = (VPAIRS + NTHREADS - 1) / NTHREADS;
-    const __bf16* kbase = (const __bf16*)p.k + ((long)b * p.nk) * p.hk * D + (long)hk * D;
-    const __bf16* vbase = (const __bf16*)p.v + ((long)b * p.nk) * p.hk * D + (long)hk * D;
-    const unsigned char* mbase = p.kmask ? (const unsigned char*)p.kmask + (long)b * p.nk : nullptr;
+    const __bf16* kbase = (const __bf16*)P.k + ((long)b * P.nk) * P.hk * D + (long)hk * D;
+    const __bf16* vbase = (const __bf16*)P.v + ((long)b * P.nk) * P.hk * D + (long)hk * D;
+    const unsigned char* mbase = P.kmask ? (const unsigned char*)P.kmask + (long)b * P.nk : nullptr;
 
     uint4 kst[KREGS];
     bf16x8 vsta[VREGS], vstb[VREGS];
@@ -210,7 +217,7 @@ __global__ __launch_bounds__(NTHREADS, 1) void attn_fwd_kernel(FwdParams p) {
     // running per-thread source pointers: load_tile is always called for
     // consecutive tiles, so addresses advance by a constant — no per-tile
     // 64-bit multiplies in the hot loop
-    const long kv_row_stride = (long)p.hk * D;
+    const long kv_row_stride = (long)P.hk * D;
     const long tile_stride = KVBLK * kv_row_stride;
     const __bf16* kptr = kbase + (long)t_lo * tile_stride
         + (tid / CH_PER_ROW) * kv_row_stride + (tid % CH_PER_ROW) * 8;
@@ -221,7 +228,7 @@ __global__ __launch_bounds__(NTHREADS, 1) void attn_fwd_kernel(FwdParams p) {
 
     auto load_tile = [&]() {
         const long j0 = j0_next;
-        const long jmax = min(j0 + KVBLK, p.nk) - 1;
+        const long jmax = min(j0 + KVBLK, P.nk) - 1;
         const bool full = jmax - j0 == KVBLK - 1;
         #pragma unroll
         for (int r = 0; r < KREGS; ++r) {
@@ -281,28 +288,28 @@ __global__ __launch_bounds__(NTHREADS, 1) void attn_fwd_kernel(FwdParams p) {
 
     // 3-deep pipeline over DOUBLE-buffered LDS: one barrier per tile; LDS
     // writes and the next-next tile's HBM loads fully overlap the MFMAs
-    const float scale2 = p.scale * LOG2E;    // softmax runs in the exp2 domain
+    const float scale2 = P.scale * LOG2E;    // softmax runs in the exp2 domain
     if (t_lo < t_hi) {
         load_tile();
         write_tile(t_lo & 1);
-        if (p.ablate == 1) write_tile((t_lo & 1) ^ 1);   // both buffers valid
-        if (t_lo + 1 < t_hi && p.ablate != 1) load_tile();
+        if (P.ablate == 1) write_tile((t_lo & 1) ^ 1);   // both buffers valid
+        if (t_lo + 1 < t_hi && P.ablate != 1) load_tile();
     }
 
     for (int t = t_lo; t < t_hi; ++t) {
         const int par = t & 1;
         const long j0 = (long)t * KVBLK;
-        const long jmax = min(j0 + KVBLK, p.nk) - 1;
+        const long jmax = min(j0 + KVBLK, P.nk) - 1;
         const bool full_tile =
             (jmax - j0 == KVBLK - 1) &&
-            (!p.causal || jmax <= wg_q_min) &&
-            (!p.has_win || (wg_q_max - j0) <= p.win) &&
-            !p.kmask && !p.bias;
+            (!P.causal || jmax <= wg_q_min) &&
+            (!P.has_win || (wg_q_max - j0) <= P.win) &&
+            !P.kmask && !P.bias;
 
         __syncthreads();
-        const bool stamp = p.ticks && blockIdx.x == 0 && bh == 0 && tid == 0
+        const bool stamp = P.ticks && blockIdx.x == 0 && bh == 0 && tid == 0
                            && blockIdx.z == 0;
-        if (stamp) p.ticks[t * 6 + 0] = __builtin_amdgcn_s_memtime();
+        if (stamp) P.ticks[t * 6 + 0] = __builtin_amdgcn_s_memtime();
 
         // ---- QK^T: S^T[kv][q] for the NBLK 32-row kv blocks
         f32x16 s[NBLK];
@@ -320,16 +327,16 @@ __global__ __launch_bounds__(NTHREADS, 1) void attn_fwd_kernel(FwdParams p) {
             }
         }
         __builtin_amdgcn_s_setprio(0);
-        if (stamp) p.ticks[t * 6 + 1] = __builtin_amdgcn_s_memtime();
+        if (stamp) P.ticks[t * 6 + 1] = __builtin_amdgcn_s_memtime();
 
         // stage tile t+1 into the other buffer while the MFMAs above retire
-        if (p.ablate != 1) {
+        if (P.ablate != 1) {
             if (t + 1 < t_hi) write_tile(par ^ 1);
             if (t + 2 < t_hi) load_tile();
         }
-        if (stamp) p.ticks[t * 6 + 2] = __builtin_amdgcn_s_memtime();
+        if (stamp) P.ticks[t * 6 + 2] = __builtin_amdgcn_s_memtime();
 
-        if (p.ablate == 2) {
+        if (P.ablate == 2) {
             // diagnostics: skip softmax VALU, feed PV garbage fragments kept
             // alive via asm (rule 17: a skipped phase must not DCE upstream)
             uint32_t gfrag[4];
@@ -374,8 +381,8 @@ __global__ __launch_bounds__(NTHREADS, 1) void attn_fwd_kernel(FwdParams p) {
                 for (int kb = 0; kb < NBLK; ++kb)
                     #pragma unroll
                     for (int r = 0; r < 16; ++r) {
-                        float xs = s[kb][r] * (p.scale * __builtin_amdgcn_rcpf(p.softclamp_value));
-                        float x = p.softclamp_value * fast_tanhf(xs) * LOG2E;
+                        float xs = s[kb][r] * (P.scale * __builtin_amdgcn_rcpf(P.softclamp_value));
+                        float x = P.softclamp_value * fast_tanhf(xs) * LOG2E;
                         s[kb][r] = x;
                         smax = fmaxf(smax, x);
                     }
@@ -387,32 +394,32 @@ __global__ __launch_bounds__(NTHREADS, 1) void attn_fwd_kernel(FwdParams p) {
                 for (int r = 0; r < 16; ++r) {
                     float x;
                     if constexpr (SOFTCLAMP) {
-                        float xs = s[kb][r] * (p.scale * __builtin_amdgcn_rcpf(p.softclamp_value));
-                        x = p.softclamp_value * fast_tanhf(xs) * LOG2E;
+                        float xs = s[kb][r] * (P.scale * __builtin_amdgcn_rcpf(P.softclamp_value));
+                        x = P.softclamp_value * fast_tanhf(xs) * LOG2E;
                     } else {
                         x = s[kb][r] * scale2;
                     }
                     long j = j0 + kb * 32 + (r & 3) + 8 * (r >> 2) + 4 * lhi;
                     bool ok = j <= jmax;
-                    if (p.bias && ok) {
+                    if (P.bias && ok) {
                         // reference semantics: sim = qk*scale + bias
                         // (natural log), applied after softclamp; our
                         // softmax runs in the exp2 domain
-                        const long bi = p.bias_mat
-                            ? (((long)b * p.h + h) * p.nq + i_clamped) * p.nk + j
-                            : ((long)b * p.h + h) * p.nk + j;
-                        x += p.bias[bi] * LOG2E;
+                        const long bi = P.bias_mat
+                            ? (((long)b * P.h + h) * P.nq + i_clamped) * P.nk + j
+                            : ((long)b * P.h + h) * P.nk + j;
+                        x += P.bias[bi] * LOG2E;
                     }
-                    if (p.causal) ok = ok && (j <= qpos_i);
-                    if (p.has_win) ok = ok && (qpos_i - j <= p.win);
-                    if (p.kmask) ok = ok && lds.kmask[par][j - j0];
+                    if (P.causal) ok = ok && (j <= qpos_i);
+                    if (P.has_win) ok = ok && (qpos_i - j <= P.win);
+                    if (P.kmask) ok = ok && lds.kmask[par][j - j0];
                     if (!ok) x = MASK_VALUE_F;
                     s[kb][r] = x;
                     smax = fmaxf(smax, x);
                 }
         }
         smax = fmaxf(smax, cross_half(smax));
-        if (stamp) p.ticks[t * 6 + 3] = __builtin_amdgcn_s_memtime();
+        if (stamp) P.ticks[t * 6 + 3] = __builtin_amdgcn_s_memtime();
 
         // ---- online softmax update (defer-max THR=0: exact — skip the O
         // rescale whenever the running max did not grow on any lane)
@@ -478,7 +485,7 @@ __global__ __launch_bounds__(NTHREADS, 1) void attn_fwd_kernel(FwdParams p) {
             }
         }
 
-        if (stamp) p.ticks[t * 6 + 4] = __builtin_amdgcn_s_memtime();
+        if (stamp) P.ticks[t * 6 + 4] = __builtin_amdgcn_s_memtime();
         // ---- PV: O^T[d][q] += V^T[d][kv] P^T[kv][q]
         __builtin_amdgcn_s_setprio(1);
         #pragma unroll
@@ -493,33 +500,33 @@ __global__ __launch_bounds__(NTHREADS, 1) void attn_fwd_kernel(FwdParams p) {
             }
         }
         __builtin_amdgcn_s_setprio(0);
-        if (stamp) p.ticks[t * 6 + 5] = __builtin_amdgcn_s_memtime();
+        if (stamp) P.ticks[t * 6 + 5] = __builtin_amdgcn_s_memtime();
     }
 
     // ---- epilogue
-    if (!row_valid) continue;
+    if (!row_valid) return;
 
     if (split_mode) {
         // write this split's unnormalized partial (merged by attn_fwd_merge)
-        const long part = (long)zsplit * p.b * p.h;
-        float* mrow = p.m + (part + (long)b * p.h + h) * p.nq;
-        float* lrow = p.l + (part + (long)b * p.h + h) * p.nq;
+        const long part = (long)zsplit * P.b * P.h;
+        float* mrow = P.m + (part + (long)b * P.h + h) * P.nq;
+        float* lrow = P.l + (part + (long)b * P.h + h) * P.nq;
         if (lhi == 0) { mrow[i] = m_run * LN2; lrow[i] = l_run; }
-        float* oa = p.o_acc + (part + (long)b * p.h + h) * D * p.nq;
+        float* oa = P.o_acc + (part + (long)b * P.h + h) * D * P.nq;
         #pragma unroll
         for (int db = 0; db < DBLK; ++db)
             #pragma unroll
             for (int r = 0; r < 16; ++r) {
                 int d = db * 32 + (r & 3) + 8 * (r >> 2) + 4 * lhi;
-                oa[(long)d * p.nq + i] = o_acc[db][r];
+                oa[(long)d * P.nq + i] = o_acc[db][r];
             }
-        continue;
+        return;
     }
 
-    if (p.is_last) {
+    if (P.is_last) {
         float l_safe = fmaxf(l_run, 1e-38f);
         float inv_l = 1.f / l_safe;
-        __bf16* ob = (__bf16*)p.out + ((long)b * p.nq + i) * p.h * D + (long)h * D;
+        __bf16* ob = (__bf16*)P.out + ((long)b * P.nq + i) * P.h * D + (long)h * D;
         #pragma unroll
         for (int db = 0; db < DBLK; ++db)
             #pragma unroll
@@ -532,22 +539,29 @@ __global__ __launch_bounds__(NTHREADS, 1) void attn_fwd_kernel(FwdParams p) {
                 *(uint2*)(ob + d) = *(uint2*)four;
             }
         if (lhi == 0) {
-            float* lsep = p.lse + ((long)b * p.h + h) * p.nq;
+            float* lsep = P.lse + ((long)b * P.h + h) * P.nq;
             lsep[i] = __logf(l_safe) + m_run * LN2;
         }
     } else {
-        float* mrow = p.m + ((long)b * p.h + h) * p.nq;
-        float* lrow = p.l + ((long)b * p.h + h) * p.nq;
+        float* mrow = P.m + ((long)b * P.h + h) * P.nq;
+        float* lrow = P.l + ((long)b * P.h + h) * P.nq;
         if (lhi == 0) { mrow[i] = m_run * LN2; lrow[i] = l_run; }
-        float* oa = p.o_acc + (((long)b * p.h + h) * D) * p.nq;
+        float* oa = P.o_acc + (((long)b * P.h + h) * D) * P.nq;
         #pragma unroll
         for (int db = 0; db < DBLK; ++db)
             #pragma unroll
             for (int r = 0; r < 16; ++r) {
                 int d = db * 32 + (r & 3) + 8 * (r >> 2) + 4 * lhi;
-                oa[(long)d * p.nq + i] = o_acc[db][r];
+                oa[(long)d * P.nq + i] = o_acc[db][r];
             }
     }
+    };
+    // noinline frame confines register allocation to the body — the pair
+    // loop's liveness made the PAIRED instantiations spill (d64 532 B,
+    // d128 656 B/lane) while the plain forms are clean; same lever as
+    // dkv_noinline_call (attn_bwd.hip)
+    if constexpr (PAIRED) fwd_noinline_call(fwd_body);
+    else fwd_body();
     }  // pair loop
 }
 
