@@ -104,6 +104,9 @@ static constexpr int DQ_QROWS_WG = DQ_WAVES * 32;     // 256
 // staging.  d128 stays at KVB=64: the KVB=128 variant its freed LDS allows
 // spills 280-336 B/lane into the hot loop and measured SLOWER (174 vs 204
 // TF headline) — the round-1 "fwd KVBLK=128 at d128" trap again.
+template <class F>
+__device__ __attribute__((noinline)) void dkv_noinline_call(F&& f) { f(); }
+
 template <int D> constexpr int dq_kvblk() { return D == 64 ? 128 : 64; }
 
 template <int D>
@@ -147,63 +150,65 @@ void attn_bwd_dq_kernel(BwdParams p) { // cap 256 VGPR, stop the 220 B/thread
         __syncthreads();                       // LDS handoff between tiles
     }
 
+    auto dq_body = [&]() {
+    const BwdParams P = p;   // register-local copy (see dkv_noinline_call)
     const long i = (long)qtile * DQ_QROWS_WG + wid * 32 + l31;
-    const bool row_valid = i < p.nq;
+    const bool row_valid = i < P.nq;
     const long ic = row_valid ? i : 0;
 
     // Q^T and dO^T fragments in registers (B-operand layout, lane = q)
-    const __bf16* qbase = (const __bf16*)p.q + ((long)b * p.nq + ic) * p.h * D + (long)h * D;
-    const __bf16* dobase = (const __bf16*)p.dout + ((long)b * p.nq + ic) * p.h * D + (long)h * D;
+    const __bf16* qbase = (const __bf16*)P.q + ((long)b * P.nq + ic) * P.h * D + (long)h * D;
+    const __bf16* dobase = (const __bf16*)P.dout + ((long)b * P.nq + ic) * P.h * D + (long)h * D;
     bf16x8 qf[KSTEPS], dof[KSTEPS];
     #pragma unroll
     for (int ks = 0; ks < KSTEPS; ++ks) {
         qf[ks] = *(const bf16x8*)(qbase + ks * 16 + lhi * 8);
         dof[ks] = *(const bf16x8*)(dobase + ks * 16 + lhi * 8);
     }
-    const float lse_i = p.lse[((long)b * p.h + h) * p.nq + ic] * 1.4426950408889634f;
-    const float delta_i = p.delta[((long)b * p.h + h) * p.nq + ic];
+    const float lse_i = P.lse[((long)b * P.h + h) * P.nq + ic] * 1.4426950408889634f;
+    const float delta_i = P.delta[((long)b * P.h + h) * P.nq + ic];
 
     f32x16 dq_acc[DBLK];
     #pragma unroll
     for (int db = 0; db < DBLK; ++db) dq_acc[db] = f32x16{};
 
     const long wg_i_min = (long)qtile * DQ_QROWS_WG;
-    const long wg_i_max = min((long)(qtile + 1) * DQ_QROWS_WG, p.nq) - 1;
-    const long wg_q_min = wg_i_min * p.q_stride + p.diag;
-    const long wg_q_max = wg_i_max * p.q_stride + p.diag;
-    const long qpos_i = i * p.q_stride + p.diag;
-    const int num_kv_tiles = (int)((p.nk + DQ_KVBLK - 1) / DQ_KVBLK);
+    const long wg_i_max = min((long)(qtile + 1) * DQ_QROWS_WG, P.nq) - 1;
+    const long wg_q_min = wg_i_min * P.q_stride + P.diag;
+    const long wg_q_max = wg_i_max * P.q_stride + P.diag;
+    const long qpos_i = i * P.q_stride + P.diag;
+    const int num_kv_tiles = (int)((P.nk + DQ_KVBLK - 1) / DQ_KVBLK);
 
     int t_lo = 0, t_hi = num_kv_tiles;
-    if (p.causal)
+    if (P.causal)
         t_hi = wg_q_max < 0 ? 0 : min((long)num_kv_tiles, wg_q_max / DQ_KVBLK + 1);
-    if (p.has_win) {
-        long x = wg_q_min - p.win - DQ_KVBLK + 1;
+    if (P.has_win) {
+        long x = wg_q_min - P.win - DQ_KVBLK + 1;
         t_lo = x <= 0 ? 0 : (int)((x + DQ_KVBLK - 1) / DQ_KVBLK);
         if (t_lo > t_hi) t_lo = t_hi;
     }
-    if (p.split > 1) {
+    if (P.split > 1) {
         // fractional split of this WG's own valid range (see attn_fwd)
         int valid = t_hi - t_lo;
-        int per = (valid + p.split - 1) / p.split;
+        int per = (valid + P.split - 1) / P.split;
         int base = t_lo;
         t_lo = base + min(valid, (int)(blockIdx.z * per));
         t_hi = base + min(valid, (int)((blockIdx.z + 1) * per));
     }
-    if (!PAIRED && p.desc) {     // descriptor mode: exact unit bounds
-        t_lo = p.desc[(long)blockIdx.x * 3 + 1];
-        t_hi = p.desc[(long)blockIdx.x * 3 + 2];
+    if (!PAIRED && P.desc) {     // descriptor mode: exact unit bounds
+        t_lo = P.desc[(long)blockIdx.x * 3 + 1];
+        t_hi = P.desc[(long)blockIdx.x * 3 + 2];
     }
 
     // staging: K + V row chunks, K^T pairs, running pointers
     constexpr int CH = D * 2 / 16;
     constexpr int KCHUNKS = DQ_KVBLK * CH;
     constexpr int KREGS = (KCHUNKS + 511) / 512;
-    const __bf16* kbase = (const __bf16*)p.k + ((long)b * p.nk) * p.hk * D + (long)hk * D;
-    const __bf16* vbase = (const __bf16*)p.v + ((long)b * p.nk) * p.hk * D + (long)hk * D;
-    const unsigned char* mbase = p.kmask ? (const unsigned char*)p.kmask + (long)b * p.nk : nullptr;
+    const __bf16* kbase = (const __bf16*)P.k + ((long)b * P.nk) * P.hk * D + (long)hk * D;
+    const __bf16* vbase = (const __bf16*)P.v + ((long)b * P.nk) * P.hk * D + (long)hk * D;
+    const unsigned char* mbase = P.kmask ? (const unsigned char*)P.kmask + (long)b * P.nk : nullptr;
 
-    const long kv_row_stride = (long)p.hk * D;
+    const long kv_row_stride = (long)P.hk * D;
     const long tile_stride = DQ_KVBLK * kv_row_stride;
     const __bf16* kptr = kbase + (long)t_lo * tile_stride
         + (tid / CH) * kv_row_stride + (tid % CH) * 8;
@@ -216,7 +221,7 @@ void attn_bwd_dq_kernel(BwdParams p) { // cap 256 VGPR, stop the 220 B/thread
 
     auto load_tile = [&]() {
         const long j0 = j0_next;
-        const long jmax = min(j0 + DQ_KVBLK, p.nk) - 1;
+        const long jmax = min(j0 + DQ_KVBLK, P.nk) - 1;
         const bool full = jmax - j0 == DQ_KVBLK - 1;
         #pragma unroll
         for (int r = 0; r < KREGS; ++r) {
@@ -247,7 +252,7 @@ void attn_bwd_dq_kernel(BwdParams p) { // cap 256 VGPR, stop the 220 B/thread
         if (mbase && tid < DQ_KVBLK) lds.kmask[par][tid] = mst;
     };
 
-    const float scale2 = p.scale * 1.4426950408889634f;   // exp2 domain
+    const float scale2 = P.scale * 1.4426950408889634f;   // exp2 domain
     if (t_lo < t_hi) {
         load_tile();
         write_tile(t_lo & 1);
@@ -257,12 +262,12 @@ void attn_bwd_dq_kernel(BwdParams p) { // cap 256 VGPR, stop the 220 B/thread
     for (int t = t_lo; t < t_hi; ++t) {
         const int par = t & 1;
         const long j0 = (long)t * DQ_KVBLK;
-        const long jmax = min(j0 + DQ_KVBLK, p.nk) - 1;
+        const long jmax = min(j0 + DQ_KVBLK, P.nk) - 1;
         const bool full_tile = !BIAS &&
             (jmax - j0 == DQ_KVBLK - 1) &&
-            (!p.causal || jmax <= wg_q_min) &&
-            (!p.has_win || (wg_q_max - j0) <= p.win) &&
-            !p.kmask;
+            (!P.causal || jmax <= wg_q_min) &&
+            (!P.has_win || (wg_q_max - j0) <= P.win) &&
+            !P.kmask;
 
         __syncthreads();
 
@@ -292,15 +297,15 @@ void attn_bwd_dq_kernel(BwdParams p) { // cap 256 VGPR, stop the 220 B/thread
                         int r = 2 * x2 + e;
                         float x, dtanh = 1.f;
                         if constexpr (SOFTCLAMP) {
-                            float inv_v = __builtin_amdgcn_rcpf(p.softclamp_value);
-                            float th = bfast_tanhf(s[r] * p.scale * inv_v);
-                            x = p.softclamp_value * th * 1.4426950408889634f;
+                            float inv_v = __builtin_amdgcn_rcpf(P.softclamp_value);
+                            float th = bfast_tanhf(s[r] * P.scale * inv_v);
+                            x = P.softclamp_value * th * 1.4426950408889634f;
                             dtanh = 1.f - th * th;
                         } else {
                             x = __builtin_fmaf(s[r], scale2, -lse_i);  // fold
                         }
                         float pv = __builtin_amdgcn_exp2f(SOFTCLAMP ? x - lse_i : x);
-                        dse[e] = pv * (dp[r] - delta_i) * dtanh * p.scale;
+                        dse[e] = pv * (dp[r] - delta_i) * dtanh * P.scale;
                     }
                     union { __hip_bfloat162 h2; uint32_t u; } cvt;
                     cvt.h2 = __float22bfloat162_rn(float2{dse[0], dse[1]});
@@ -316,9 +321,9 @@ void attn_bwd_dq_kernel(BwdParams p) { // cap 256 VGPR, stop the 220 B/thread
                         long jj = j0 + kb * 32 + (r & 3) + 8 * (r >> 2) + 4 * lhi;
                         float x, dtanh = 1.f;
                         if constexpr (SOFTCLAMP) {
-                            float inv_v = __builtin_amdgcn_rcpf(p.softclamp_value);
-                            float th = bfast_tanhf(s[r] * p.scale * inv_v);
-                            x = p.softclamp_value * th * 1.4426950408889634f;
+                            float inv_v = __builtin_amdgcn_rcpf(P.softclamp_value);
+                            float th = bfast_tanhf(s[r] * P.scale * inv_v);
+                            x = P.softclamp_value * th * 1.4426950408889634f;
                             dtanh = 1.f - th * th;
                         } else {
                             x = s[r] * scale2;
@@ -326,18 +331,18 @@ void attn_bwd_dq_kernel(BwdParams p) { // cap 256 VGPR, stop the 220 B/thread
                         bool ok = row_valid && jj <= jmax;
                         if constexpr (BIAS) {
                             if (ok) {
-                                const long bi = p.bias_mat
-                                    ? (((long)b * p.h + h) * p.nq + ic) * p.nk + jj
-                                    : ((long)b * p.h + h) * p.nk + jj;
-                                x += p.bias[bi] * 1.4426950408889634f;
+                                const long bi = P.bias_mat
+                                    ? (((long)b * P.h + h) * P.nq + ic) * P.nk + jj
+                                    : ((long)b * P.h + h) * P.nk + jj;
+                                x += P.bias[bi] * 1.4426950408889634f;
                             }
                         }
-                        if (p.causal) ok = ok && (jj <= qpos_i);
-                        if (p.has_win) ok = ok && (qpos_i - jj <= p.win);
-                        if (p.kmask) ok = ok && lds.kmask[par][jj - j0];
+                        if (P.causal) ok = ok && (jj <= qpos_i);
+                        if (P.has_win) ok = ok && (qpos_i - jj <= P.win);
+                        if (P.kmask) ok = ok && lds.kmask[par][jj - j0];
                         float pv = ok ? __builtin_amdgcn_exp2f(
                             SOFTCLAMP ? x - lse_i : x - lse_i) : 0.f;
-                        dse[e] = pv * (dp[r] - delta_i) * dtanh * p.scale;
+                        dse[e] = pv * (dp[r] - delta_i) * dtanh * P.scale;
                     }
                     union { __hip_bfloat162 h2; uint32_t u; } cvt;
                     cvt.h2 = __float22bfloat162_rn(float2{dse[0], dse[1]});
@@ -414,19 +419,24 @@ void attn_bwd_dq_kernel(BwdParams p) { // cap 256 VGPR, stop the 220 B/thread
         }
     }
 
-    if (!row_valid) continue;
+    if (!row_valid) return;
     // epilogue: dq (B, Nq, H, D) fp32; unique writer per row unless the kv
     // walk is split across grid.z (then fp32 atomics, contention = split)
-    float* dqp = p.dq + ((long)b * p.nq + i) * p.h * D + (long)h * D;
+    float* dqp = P.dq + ((long)b * P.nq + i) * P.h * D + (long)h * D;
     #pragma unroll
     for (int db = 0; db < DBLK; ++db)
         #pragma unroll
         for (int r = 0; r < 16; ++r) {
             int d = db * 32 + (r & 3) + 8 * (r >> 2) + 4 * lhi;
-            if (p.split > 1 || p.desc) atomicAdd(dqp + d, dq_acc[db][r]);
-            else if (p.accumulate) dqp[d] += dq_acc[db][r];
+            if (P.split > 1 || P.desc) atomicAdd(dqp + d, dq_acc[db][r]);
+            else if (P.accumulate) dqp[d] += dq_acc[db][r];
             else dqp[d] = dq_acc[db][r];
         }
+    };
+    // measured: the noinline frame REGRESSES d128-PAIRED dq (105 vs 123 TF)
+    // unlike fwd/dkv, but the lambda restructure itself (register-local P
+    // copy) lifts d128-plain (200 -> 207) — keep the body direct-called.
+    dq_body();
     }  // pair loop
 }
 
@@ -449,8 +459,6 @@ struct DkvLds {
     __align__(16) float delta[2][QT];
 };
 
-template <class F>
-__device__ __attribute__((noinline)) void dkv_noinline_call(F&& f) { f(); }
 
 template <int D, int QT, bool SOFTCLAMP, bool PAIRED, bool BIAS = false>
 __global__ __launch_bounds__(512, 2)   // see dq kernel note (552 B spills)
