@@ -38,6 +38,8 @@ def main():
     ap.add_argument("--batch", type=int, default=1)
     ap.add_argument("--causal", action="store_true")
     ap.add_argument("--striped", action="store_true")
+    ap.add_argument("--softclamp", action="store_true",
+                    help="gemma-style tanh score cap (dedicated kernel instantiations)")
     ap.add_argument("--fwd-only", action="store_true",
                     help="measure forward only (diagnostics; not the headline metric)")
     ap.add_argument("--config", type=int, default=None, choices=[2, 3, 4, 5],
@@ -127,7 +129,8 @@ def main():
         def step():
             out, _ = attn(q, k, v, causal=args.causal, ring_reduce_col=True,
                           striped_ring_attn=args.striped, ring_size=world,
-                          bucket_size=min(n, 1024))
+                          bucket_size=min(n, 1024),
+                          softclamp_qk_sim=args.softclamp)
             if not args.fwd_only:
                 out.backward(out.detach())  # fwd + full bwd (dq, dk, dv incl. ring)
                 q.grad = None; k.grad = None; v.grad = None
