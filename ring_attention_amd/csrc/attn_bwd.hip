@@ -838,9 +838,10 @@ void attn_bwd_dkv_kernel(BwdParams p) {
     };
     // the noinline frame confines register allocation to the body — at d64
     // it cuts the hot-loop spills roughly in half for BOTH instantiations
-    // (measured headline +4%, causal +17-24%); d128's plain path is better
-    // inlined (direct call), so only d64 and the paired form take the call
-    if constexpr (D == 64 || PAIRED) dkv_noinline_call(dkv_body);
+    // (measured headline +4%, causal +17-24%); d128 is better DIRECT for
+    // both forms (paired: 125 -> 138 TF causal 8k — the same value-copy
+    // preference as the d128 fwd/dq bodies), so only d64 takes the call
+    if constexpr (D == 64) dkv_noinline_call(dkv_body);
     else dkv_body();
     }  // pair loop
 }
