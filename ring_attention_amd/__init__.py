@@ -32,6 +32,8 @@ __all__ = [
     "ring_flash_attn_hip_",
     "flash_attn",
     "flash_attn_offset",
+    "flash_attn_fp8",
+    "quantize_fp8",
     "RingAttention",
     "RingTransformer",
     "RingRotaryEmbedding",
@@ -51,4 +53,7 @@ def __getattr__(name):
                 "flash_attn_offset"):
         from .ops import ring_flash_hip
         return getattr(ring_flash_hip, name)
+    if name in ("flash_attn_fp8", "quantize_fp8"):
+        from .ops import fp8
+        return getattr(fp8, name)
     raise AttributeError(f"module {__name__!r} has no attribute {name!r}")

@@ -147,6 +147,23 @@ struct DecodeParams {
 
 void launch_decode_partial(const DecodeParams& p, int head_dim, hipStream_t stream);
 
+struct Fp8FwdParams {
+    const void* q;      // e4m3 bytes (B, Nq, H, D)
+    const void* k;      // e4m3 bytes (B, Nk, H, D)
+    const void* vt;     // e4m3 bytes (B, H, D, Nk)  — host-pre-transposed V
+    const void* qs;     // e8m0 bytes (B, Nq, H): per-row scale exponent+127
+    const void* ks;     // e8m0 bytes (B, Nk, H)
+    const void* vs;     // e8m0 bytes (B, H, D, Nvs): per (d row, 64-kv chunk)
+    void* out;          // bf16 (B, Nq, H, D)
+    float* lse;         // fp32 (B, H, Nq)
+    int b, h;
+    long nq, nk;
+    int nvs;            // Nk / 64
+    float scale;
+};
+
+void launch_attn_fwd_fp8(const Fp8FwdParams& p, hipStream_t stream);
+
 struct RotaryParams {
     const void* x;      // bf16 (B, N, H, D)
     const float* cos_t; // fp32 (N, D/2) host-precomputed table

@@ -1,0 +1,78 @@
+"""MX-FP8 serving forward — an MI355X-only capability beyond the reference.
+
+The reference's kernels are bf16/fp16 Triton
+(/root/reference/ring_attention_pytorch/triton_flash_attn.py); gfx950 adds
+block-scaled e4m3 MFMA at ~2x the bf16 matrix rate, which this module
+exposes as a quantize + fused-forward pair for inference/prefill:
+
+    out, lse = flash_attn_fp8(q, k, v)          # (b, n, h, d) bf16 in/out
+
+Numerics: Q/K use one e8m0 scale per row (2^ceil(log2(amax/448))), V one
+per (d row, 64-kv chunk), and the softmax matrix P needs no scale at all
+(exp2(x - m) <= 1).  Softmax and the output accumulator stay fp32 inside
+the kernel; only the MFMA operands are 8-bit.  Expected error vs a bf16
+forward is ~1-3% relative on out (P is quantized to e4m3's 3 mantissa
+bits); use the bf16 path when training.
+
+v0 scope (kernel asserts): non-causal, no mask/bias, hk == h, d == 64,
+nq % 256 == 0, nk % 128 == 0, single shard (no ring pass).
+"""
+
+from __future__ import annotations
+
+import torch
+from torch import Tensor
+
+from . import hip_ext
+
+
+def _e8m0(amax: Tensor) -> tuple[Tensor, Tensor]:
+    """amax (positive fp32) -> (e8m0 byte = exp+127, scale = 2^exp)."""
+    e = torch.ceil(torch.log2(amax.clamp(min=2.0 ** -126) / 448.0))
+    e = e.clamp(-127.0, 127.0)
+    return (e + 127.0).to(torch.uint8), torch.exp2(e)
+
+
+def quantize_fp8(q: Tensor, k: Tensor, v: Tensor):
+    """Quantize (b, n, h, d) q/k/v for attn_fwd_fp8.
+
+    Returns (q8, k8, v8t, qs, ks, vs) uint8 views:
+      q8/k8 (b, n, h, d) e4m3; v8t (b, h, d, n) e4m3 (pre-transposed);
+      qs/ks (b, n, h) e8m0 row scales; vs (b, h, d, n // 64) e8m0.
+    """
+    qf, kf = q.float(), k.float()
+    qs_b, qs_s = _e8m0(qf.abs().amax(dim=-1))
+    ks_b, ks_s = _e8m0(kf.abs().amax(dim=-1))
+    q8 = (qf / qs_s.unsqueeze(-1)).to(torch.float8_e4m3fn)
+    k8 = (kf / ks_s.unsqueeze(-1)).to(torch.float8_e4m3fn)
+
+    vt = v.permute(0, 2, 3, 1).float().contiguous()      # (b, h, d, n)
+    b, h, d, n = vt.shape
+    assert n % 64 == 0, "fp8 path: kv length must be a multiple of 64"
+    vs_b, vs_s = _e8m0(vt.view(b, h, d, n // 64, 64).abs().amax(dim=-1))
+    v8t = (vt / vs_s.repeat_interleave(64, dim=-1)).to(torch.float8_e4m3fn)
+    return (q8.view(torch.uint8), k8.view(torch.uint8),
+            v8t.view(torch.uint8), qs_b.contiguous(), ks_b.contiguous(),
+            vs_b.contiguous())
+
+
+@torch.no_grad()
+def flash_attn_fp8(
+    q: Tensor, k: Tensor, v: Tensor,
+    sm_scale: float | None = None,
+) -> tuple[Tensor, Tensor]:
+    """MX-FP8 non-causal attention forward on (b, n, h, d) tensors.
+
+    Returns (out bf16 (b, n, h, d), lse fp32 (b, h, n)).  Quantizes
+    internally; pass pre-quantized operands via flash_attn_fp8_quantized
+    to amortize quantization across decode steps.
+    """
+    q8, k8, v8t, qs, ks, vs = quantize_fp8(q, k, v)
+    return flash_attn_fp8_quantized(
+        q8, k8, v8t, qs, ks, vs,
+        sm_scale if sm_scale is not None else q.shape[-1] ** -0.5)
+
+
+def flash_attn_fp8_quantized(q8, k8, v8t, qs, ks, vs, sm_scale: float):
+    out, lse = hip_ext.require().attn_fwd_fp8(q8, k8, v8t, qs, ks, vs, sm_scale)
+    return out, lse

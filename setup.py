@@ -23,6 +23,7 @@ ext = CUDAExtension(
         os.path.join(CSRC, "bindings.cpp"),
         os.path.join(CSRC, "attn_fwd.hip"),
         os.path.join(CSRC, "attn_fwd_v2.hip"),
+        os.path.join(CSRC, "attn_fwd_fp8.hip"),
         os.path.join(CSRC, "attn_bwd.hip"),
         os.path.join(CSRC, "decode.hip"),
         os.path.join(CSRC, "rotary.hip"),
