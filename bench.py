@@ -38,6 +38,9 @@ def main():
     ap.add_argument("--batch", type=int, default=1)
     ap.add_argument("--causal", action="store_true")
     ap.add_argument("--striped", action="store_true")
+    ap.add_argument("--fp8", action="store_true",
+                    help="MX-FP8 serving forward (e4m3 block-scaled MFMA; "
+                         "implies --fwd-only, non-causal, 1 GPU)")
     ap.add_argument("--softclamp", action="store_true",
                     help="gemma-style tanh score cap (dedicated kernel instantiations)")
     ap.add_argument("--fwd-only", action="store_true",
@@ -121,6 +124,19 @@ def main():
             if not args.fwd_only:
                 out.backward(out.detach())
                 q.grad = None; k.grad = None; v.grad = None
+    elif getattr(args, "fp8", False) and on_gpu:
+        # MX-FP8 serving forward: quantization is per-prefill (outside the
+        # step, like a KV-cache write); the step is the fused e4m3 forward
+        from ring_attention_amd.ops.fp8 import quantize_fp8, flash_attn_fp8_quantized
+        args.fwd_only = True
+        q = torch.randn(b, n, h, d, device=device, dtype=dtype)
+        k = torch.randn(b, n, h, d, device=device, dtype=dtype)
+        v = torch.randn(b, n, h, d, device=device, dtype=dtype)
+        fp8_args = quantize_fp8(q, k, v)
+        fp8_scale = d ** -0.5
+
+        def step():
+            flash_attn_fp8_quantized(*fp8_args, fp8_scale)
     else:
         q = torch.randn(b, n, h, d, device=device, dtype=dtype, requires_grad=True)
         k = torch.randn(b, n, hk, d, device=device, dtype=dtype, requires_grad=True)
@@ -228,10 +244,13 @@ def main():
             "higher_is_better": True,
             "scaling": "weak",
             "vs_baseline": None,
-            "dtype": "bf16" if on_gpu else "fp32-cpu-fallback",
+            "dtype": ("fp8_e4m3" if getattr(args, "fp8", False) and on_gpu
+                      else "bf16" if on_gpu else "fp32-cpu-fallback"),
             "data": "synthetic",
             "config": {
-                "model": ("zig_zag_attn (causal GQA + rotary)" if args.config == 4
+                "model": ("flash_attn_fp8 (MX-FP8 serving forward)"
+                          if getattr(args, "fp8", False) and on_gpu
+                          else "zig_zag_attn (causal GQA + rotary)" if args.config == 4
                           else "ring_flash_attn (non-causal, d_head 64)" if not args.causal
                           else "ring_flash_attn (causal%s)" % (" striped" if args.striped else "")),
                 "global_batch": b,
