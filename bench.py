@@ -40,7 +40,7 @@ def main():
     ap.add_argument("--striped", action="store_true")
     ap.add_argument("--fp8", action="store_true",
                     help="MX-FP8 serving forward (e4m3 block-scaled MFMA; "
-                         "implies --fwd-only, non-causal, 1 GPU)")
+                         "implies --fwd-only, 1 GPU; combine with --causal)")
     ap.add_argument("--softclamp", action="store_true",
                     help="gemma-style tanh score cap (dedicated kernel instantiations)")
     ap.add_argument("--fwd-only", action="store_true",
@@ -136,7 +136,7 @@ def main():
         fp8_scale = d ** -0.5
 
         def step():
-            flash_attn_fp8_quantized(*fp8_args, fp8_scale)
+            flash_attn_fp8_quantized(*fp8_args, fp8_scale, causal=args.causal)
     else:
         q = torch.randn(b, n, h, d, device=device, dtype=dtype, requires_grad=True)
         k = torch.randn(b, n, hk, d, device=device, dtype=dtype, requires_grad=True)

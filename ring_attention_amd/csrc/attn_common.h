@@ -156,10 +156,12 @@ struct Fp8FwdParams {
     const void* vs;     // e8m0 bytes (B, H, D, Nvs): per (d row, 64-kv chunk)
     void* out;          // bf16 (B, Nq, H, D)
     float* lse;         // fp32 (B, H, Nq)
-    int b, h;
+    int b, h, hk;       // GQA: kv heads (qh pairs qh % hk)
     long nq, nk;
     int nvs;            // Nk / 64
     float scale;
+    int causal;         // standard causal (qpos(i) = i), paired-tile grid
+    int paired;         // total q tiles T (causal launcher)
 };
 
 void launch_attn_fwd_fp8(const Fp8FwdParams& p, hipStream_t stream);

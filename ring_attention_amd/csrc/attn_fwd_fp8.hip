@@ -19,13 +19,17 @@
 //     hardware's k-block interleave entirely.  Granularity used:
 //       Q, K: one e8m0 per row (per 64-d chunk) = 2^ceil(log2(amax/448))
 //       V:    one e8m0 per (d row, 64-kv chunk)
-//       P:    exactly 1.0 (exp2(x - m) <= 1 by construction) -> byte 127
+//       P:    fixed 2^-8 (exp2(x - m) <= 1; the shift moves the e4m3
+//             subnormal cutoff from 2^-9 to 2^-17)
 //   * e4m3 is OCP e4m3fn (torch.float8_e4m3fn matches bit-for-bit).
 //
-// v0 scope (asserted in the binding): non-causal, no mask/bias/window/
-// softclamp, single shot (no ring resume), hk == h, nq % 256 == 0,
-// nk % 128 == 0, D = 64.  The ring path keeps bf16; this is the
-// single-shard serving prefill path.
+// Causal: the PAIRED instantiation uses the bf16 kernels' mirrored-tile
+// load balance (WG x runs q-tiles (x, T-1-x) — uniform T+1 kv-tile walks
+// per WG instead of the 2:1 triangle imbalance).  GQA pairs q head qh with
+// kv head qh % hk (reference tile convention).
+//
+// Scope (asserted in the binding): no mask/bias/window/softclamp, single
+// shot (no ring resume), D = 64, nq % 256 == 0, nk % 128 == 0.
 
 #include <hip/hip_runtime.h>
 #include <hip/hip_bf16.h>
@@ -41,8 +45,8 @@ static constexpr int FP8_WAVES = 8;
 static constexpr int FP8_QROWS_WAVE = 32;
 static constexpr int FP8_QROWS_WG = FP8_WAVES * FP8_QROWS_WAVE;   // 256
 static constexpr int FP8_NTHREADS = FP8_WAVES * 64;               // 512
-static constexpr int FP8_KVBLK = 128;
-static constexpr int FP8_NBLK = FP8_KVBLK / 32;                   // 4
+static constexpr int FP8_KVBLK = 128;                             // 256 measured
+static constexpr int FP8_NBLK = FP8_KVBLK / 32;                   // negative (profiles/README.md)
 
 // 16-byte-chunk XOR swizzle within a row (CH chunks per row), same
 // both-sides rule as the bf16 kernels
@@ -64,6 +68,10 @@ struct Fp8Lds {
     unsigned char ks[2][FP8_KVBLK];                      // k row e8m0
 };
 
+template <class F>
+__device__ __attribute__((noinline)) void fp8_noinline_call(F&& f) { f(); }
+
+template <bool PAIRED>
 __global__ __launch_bounds__(FP8_NTHREADS, 1)
 void attn_fwd_fp8_kernel(Fp8FwdParams p) {
     constexpr int D = 64;
@@ -82,16 +90,29 @@ void attn_fwd_fp8_kernel(Fp8FwdParams p) {
     const int bh = blockIdx.y;
     const int b = bh / p.h;
     const int h = bh % p.h;
+    const int hk = h % p.hk;              // reference tile GQA pairing
 
-    const long i = (long)blockIdx.x * FP8_QROWS_WG + wid * FP8_QROWS_WAVE + l31;
+    const int n_pit = PAIRED
+        ? (p.paired - 1 - (int)blockIdx.x == (int)blockIdx.x ? 1 : 2) : 1;
+    for (int pit = 0; pit < n_pit; ++pit) {
+    const int qtile = PAIRED
+        ? (pit == 0 ? (int)blockIdx.x : p.paired - 1 - (int)blockIdx.x)
+        : (int)blockIdx.x;
+    if (PAIRED && pit == 1) {
+        __syncthreads();                  // LDS handoff between tiles
+    }
+
+    auto fp8_body = [&]() {
+    const Fp8FwdParams P = p;             // register-local (noinline frame)
+    const long i = (long)qtile * FP8_QROWS_WG + wid * FP8_QROWS_WAVE + l31;
 
     // ---- Q fragment: 32 e4m3 bytes (d = 32*lhi .. +31) + row scale
-    const unsigned char* qrow = (const unsigned char*)p.q
-        + ((long)b * p.nq + i) * p.h * D + (long)h * D + 32 * lhi;
+    const unsigned char* qrow = (const unsigned char*)P.q
+        + ((long)b * P.nq + i) * P.h * D + (long)h * D + 32 * lhi;
     union { i32x8_ v; uint4 u4[2]; } qf;
     qf.u4[0] = *(const uint4*)qrow;
     qf.u4[1] = *(const uint4*)(qrow + 16);
-    const int qs = ((const unsigned char*)p.qs)[((long)b * p.nq + i) * p.h + h];
+    const int qs = ((const unsigned char*)P.qs)[((long)b * P.nq + i) * P.h + h];
 
     // ---- accumulators
     float m_run = MASK_VALUE_F, l_run = 0.f;
@@ -99,21 +120,25 @@ void attn_fwd_fp8_kernel(Fp8FwdParams p) {
     #pragma unroll
     for (int db = 0; db < DBLK; ++db) o_acc[db] = f32x16{};
 
-    const int num_kv_tiles = (int)(p.nk / FP8_KVBLK);
+    // ---- causal tile range for this workgroup
+    const long wg_q_min = (long)qtile * FP8_QROWS_WG;
+    const long wg_q_max = wg_q_min + FP8_QROWS_WG - 1;
+    const int num_kv_tiles = (int)(P.nk / FP8_KVBLK);
+    const int t_hi = P.causal
+        ? min(num_kv_tiles, (int)(wg_q_max / FP8_KVBLK) + 1) : num_kv_tiles;
 
     // ---- staging (one uint4 per thread per image per tile)
-    const unsigned char* kbase = (const unsigned char*)p.k
-        + ((long)b * p.nk) * p.h * D + (long)h * D;
-    const unsigned char* vtbase = (const unsigned char*)p.vt
-        + (((long)b * p.h + h) * D) * p.nk;
-    const unsigned char* ksbase = (const unsigned char*)p.ks
-        + ((long)b * p.nk) * p.h + h;
+    const unsigned char* kbase = (const unsigned char*)P.k
+        + ((long)b * P.nk) * P.hk * D + (long)hk * D;
+    const unsigned char* vtbase = (const unsigned char*)P.vt
+        + (((long)b * P.hk + hk) * D) * P.nk;
+    const unsigned char* ksbase = (const unsigned char*)P.ks
+        + ((long)b * P.nk) * P.hk + hk;
 
-    const long k_row_stride = (long)p.h * D;
+    const long k_row_stride = (long)P.hk * D;
     const unsigned char* kptr = kbase + (tid / 4) * k_row_stride + (tid % 4) * 16;
-    const unsigned char* vptr = vtbase + (long)(tid / 8) * p.nk + (tid % 8) * 16;
-    const unsigned char* ksptr = ksbase + (long)tid * p.h;
-    long j0_next = 0;
+    const unsigned char* vptr = vtbase + (long)(tid / 8) * P.nk + (tid % 8) * 16;
+    const unsigned char* ksptr = ksbase + (long)tid * P.hk;
 
     uint4 kst, vst;
     unsigned char ksst = 0;
@@ -124,8 +149,7 @@ void attn_fwd_fp8_kernel(Fp8FwdParams p) {
         if (tid < FP8_KVBLK) ksst = *ksptr;
         kptr += (long)FP8_KVBLK * k_row_stride;
         vptr += FP8_KVBLK;
-        ksptr += (long)FP8_KVBLK * p.h;
-        j0_next += FP8_KVBLK;
+        ksptr += (long)FP8_KVBLK * P.hk;
     };
     auto write_tile = [&](int par) {
         {   // K: row = tid/4 (kv), chunk = tid%4 of 4 (64 B rows)
@@ -139,18 +163,18 @@ void attn_fwd_fp8_kernel(Fp8FwdParams p) {
         if (tid < FP8_KVBLK) lds.ks[par][tid] = ksst;
     };
 
-    const float scale2 = p.scale * LOG2E_;
+    const float scale2 = P.scale * LOG2E_;
 
-    if (num_kv_tiles > 0) {
+    if (t_hi > 0) {
         load_tile();
         write_tile(0);
-        if (num_kv_tiles > 1) load_tile();
+        if (t_hi > 1) load_tile();
     }
 
-    for (int t = 0; t < num_kv_tiles; ++t) {
+    for (int t = 0; t < t_hi; ++t) {
         const int par = t & 1;
         const long j0 = (long)t * FP8_KVBLK;
-        (void)j0;
+        const bool full_tile = !P.causal || (j0 + FP8_KVBLK - 1 <= wg_q_min);
 
         __syncthreads();
 
@@ -171,24 +195,39 @@ void attn_fwd_fp8_kernel(Fp8FwdParams p) {
         __builtin_amdgcn_s_setprio(0);
 
         // stage tile t+1 while the MFMAs retire
-        if (t + 1 < num_kv_tiles) write_tile(par ^ 1);
-        if (t + 2 < num_kv_tiles) load_tile();
+        if (t + 1 < t_hi) write_tile(par ^ 1);
+        if (t + 2 < t_hi) load_tile();
 
         // ---- softmax (exp2 domain; scores are true-scale fp32 — the MFMA
-        // applied the e8m0 dequant in hardware)
+        // applied the e8m0 dequant in hardware).  Masked (diagonal) tiles
+        // scale + mask in place and switch the exp fold to escale = 1.
         float smax = MASK_VALUE_F;
-        #pragma unroll
-        for (int kb = 0; kb < FP8_NBLK; ++kb)
+        if (full_tile) {
             #pragma unroll
-            for (int r = 0; r < 16; ++r) smax = fmaxf(smax, s[kb][r]);
-        smax *= scale2;
+            for (int kb = 0; kb < FP8_NBLK; ++kb)
+                #pragma unroll
+                for (int r = 0; r < 16; ++r) smax = fmaxf(smax, s[kb][r]);
+            smax *= scale2;
+        } else {
+            #pragma unroll
+            for (int kb = 0; kb < FP8_NBLK; ++kb)
+                #pragma unroll
+                for (int r = 0; r < 16; ++r) {
+                    long j = j0 + kb * 32 + (r & 3) + 8 * (r >> 2) + 4 * lhi;
+                    float x = s[kb][r] * scale2;
+                    if (j > i) x = MASK_VALUE_F;   // qpos(i) = i (v0)
+                    s[kb][r] = x;
+                    smax = fmaxf(smax, x);
+                }
+        }
         smax = fmaxf(smax, fp8_cross_half(smax));
 
         float m_new = fmaxf(m_run, smax);
         const bool any_growth = !__all(smax <= m_run);
         const float m_exp = fmaxf(m_new, -1.7e38f);
+        const float escale = full_tile ? scale2 : 1.f;
 
-        // exp2 + pack to e4m3 (P scale = 1.0: values in (0,1])
+        // exp2 + pack to e4m3 at 2^8 (PV dequants via scale byte 119)
         // own_dw[kb][g] = bytes of kv rows (8g + 4*lhi .. +3) of block kb
         uint32_t own_dw[FP8_NBLK][4];
         float partial[FP8_NBLK * 8];
@@ -196,16 +235,12 @@ void attn_fwd_fp8_kernel(Fp8FwdParams p) {
         for (int kb = 0; kb < FP8_NBLK; ++kb) {
             #pragma unroll
             for (int g = 0; g < 4; ++g) {
-                float e0 = __builtin_amdgcn_exp2f(__builtin_fmaf(s[kb][4 * g + 0], scale2, -m_exp));
-                float e1 = __builtin_amdgcn_exp2f(__builtin_fmaf(s[kb][4 * g + 1], scale2, -m_exp));
-                float e2 = __builtin_amdgcn_exp2f(__builtin_fmaf(s[kb][4 * g + 2], scale2, -m_exp));
-                float e3 = __builtin_amdgcn_exp2f(__builtin_fmaf(s[kb][4 * g + 3], scale2, -m_exp));
+                float e0 = __builtin_amdgcn_exp2f(__builtin_fmaf(s[kb][4 * g + 0], escale, -m_exp));
+                float e1 = __builtin_amdgcn_exp2f(__builtin_fmaf(s[kb][4 * g + 1], escale, -m_exp));
+                float e2 = __builtin_amdgcn_exp2f(__builtin_fmaf(s[kb][4 * g + 2], escale, -m_exp));
+                float e3 = __builtin_amdgcn_exp2f(__builtin_fmaf(s[kb][4 * g + 3], escale, -m_exp));
                 partial[kb * 8 + 2 * g] = e0 + e1;
                 partial[kb * 8 + 2 * g + 1] = e2 + e3;
-                // quantize P * 2^8 (dequant via scale byte 119 = 2^-8): P is
-                // in (0,1], so the shift moves the e4m3 subnormal cutoff from
-                // 2^-9 to 2^-17 — tail probabilities survive; relative
-                // precision is unchanged (e4m3 mantissa is 3 bits regardless)
                 int u = __builtin_amdgcn_cvt_pk_fp8_f32(e0 * 256.f, e1 * 256.f, 0, false);
                 u = __builtin_amdgcn_cvt_pk_fp8_f32(e2 * 256.f, e3 * 256.f, u, true);
                 own_dw[kb][g] = (uint32_t)u;
@@ -234,9 +269,9 @@ void attn_fwd_fp8_kernel(Fp8FwdParams p) {
         //         slot 2g+1 <- block (2c + lhi) rows 8g+4..7
         // swap(own_dw[2c][g], own_dw[2c+1][g]):
         //   lanes<32: r0 = own [2c][g] (rows 8g+0-3), r1 = partner [2c][g]
-        //             (rows 8g+4-7)            -> block 2c  = 2c+lhi ✓
+        //             (rows 8g+4-7)            -> block 2c  = 2c+lhi
         //   lanes>=32: r0 = partner [2c+1][g] (rows 8g+0-3), r1 = own
-        //             [2c+1][g] (rows 8g+4-7)  -> block 2c+1 = 2c+lhi ✓
+        //             [2c+1][g] (rows 8g+4-7)  -> block 2c+1 = 2c+lhi
         #pragma unroll
         for (int c = 0; c < 2; ++c) {
             union { i32x8_ v; uint32_t dw[8]; } pf;
@@ -258,8 +293,8 @@ void attn_fwd_fp8_kernel(Fp8FwdParams p) {
                                            + fswz<8>(drow, 4 * c + 2 * lhi) * 16);
                 vf.u4[1] = *(const uint4*)(lds.vt[par] + drow * FP8_KVBLK
                                            + fswz<8>(drow, 4 * c + 2 * lhi + 1) * 16);
-                int sv = ((const unsigned char*)p.vs)[
-                    (((long)b * p.h + h) * D + drow) * p.nvs + (t * 2 + c)];
+                int sv = ((const unsigned char*)P.vs)[
+                    (((long)b * P.hk + hk) * D + drow) * P.nvs + (t * 2 + c)];
                 o_acc[db] = __builtin_amdgcn_mfma_scale_f32_32x32x64_f8f6f4(
                     vf.v, pf.v, o_acc[db], 0, 0, 0, sv, 0, 119);
             }
@@ -270,7 +305,7 @@ void attn_fwd_fp8_kernel(Fp8FwdParams p) {
     // ---- epilogue: normalize, emit bf16 out (B,Nq,H,D) + lse (B,H,Nq)
     float l_safe = fmaxf(l_run, 1e-38f);
     float inv_l = 1.f / l_safe;
-    __bf16* ob = (__bf16*)p.out + ((long)b * p.nq + i) * p.h * D + (long)h * D;
+    __bf16* ob = (__bf16*)P.out + ((long)b * P.nq + i) * P.h * D + (long)h * D;
     #pragma unroll
     for (int db = 0; db < DBLK; ++db)
         #pragma unroll
@@ -283,15 +318,27 @@ void attn_fwd_fp8_kernel(Fp8FwdParams p) {
             *(uint2*)(ob + d) = *(uint2*)four;
         }
     if (lhi == 0) {
-        float* lsep = p.lse + ((long)b * p.h + h) * p.nq;
+        float* lsep = P.lse + ((long)b * P.h + h) * P.nq;
         lsep[i] = __logf(l_safe) + m_run * LN2_;
     }
+    };
+    if constexpr (PAIRED) fp8_noinline_call(fp8_body);
+    else fp8_body();
+    }  // pair loop
 }
 
 void launch_attn_fwd_fp8(const Fp8FwdParams& p, hipStream_t stream) {
-    dim3 grid((unsigned)(p.nq / FP8_QROWS_WG), (unsigned)(p.b * p.h));
+    const long T = p.nq / FP8_QROWS_WG;
     dim3 block(FP8_NTHREADS);
-    hipLaunchKernelGGL(attn_fwd_fp8_kernel, grid, block, 0, stream, p);
+    if (p.causal) {
+        Fp8FwdParams pc = p;
+        pc.paired = (int)T;
+        dim3 grid((unsigned)((T + 1) / 2), (unsigned)(p.b * p.h));
+        hipLaunchKernelGGL(attn_fwd_fp8_kernel<true>, grid, block, 0, stream, pc);
+    } else {
+        dim3 grid((unsigned)T, (unsigned)(p.b * p.h));
+        hipLaunchKernelGGL(attn_fwd_fp8_kernel<false>, grid, block, 0, stream, p);
+    }
 }
 
 }  // namespace ring_attn

@@ -273,7 +273,7 @@ std::vector<at::Tensor> decode_partial(at::Tensor q, at::Tensor k, at::Tensor v,
 
 std::vector<at::Tensor> attn_fwd_fp8(at::Tensor q8, at::Tensor k8, at::Tensor v8t,
                                       at::Tensor qs, at::Tensor ks, at::Tensor vs,
-                                      double sm_scale) {
+                                      double sm_scale, bool causal) {
     // MX-FP8 serving forward (see attn_fwd_fp8.hip header for scope):
     //   q8 (B,Nq,H,D) u8 e4m3; k8 (B,Nk,H,D) u8; v8t (B,H,D,Nk) u8
     //   qs (B,Nq,H) u8 e8m0(+127); ks (B,Nk,H) u8; vs (B,H,D,Nk/64) u8
@@ -283,20 +283,23 @@ std::vector<at::Tensor> attn_fwd_fp8(at::Tensor q8, at::Tensor k8, at::Tensor v8
                     "fp8 operands must be contiguous uint8 views");
     }
     const int64_t B = q8.size(0), NQ = q8.size(1), H = q8.size(2), D = q8.size(3);
-    const int64_t NK = k8.size(1);
+    const int64_t NK = k8.size(1), HK = k8.size(2);
     TORCH_CHECK(D == 64, "fp8 path: head dim 64 only (v0)");
-    TORCH_CHECK(k8.size(2) == H, "fp8 path: hk == h required (v0)");
+    TORCH_CHECK(H % HK == 0, "fp8 path: q heads must be a multiple of kv heads");
     TORCH_CHECK(NQ % 256 == 0, "fp8 path: nq must be a multiple of 256 (v0)");
     TORCH_CHECK(NK % 128 == 0, "fp8 path: nk must be a multiple of 128 (v0)");
-    TORCH_CHECK(v8t.size(3) == NK && v8t.size(2) == D && vs.size(3) == NK / 64);
+    TORCH_CHECK(!causal || NQ == NK, "fp8 causal: nq == nk (v0)");
+    TORCH_CHECK(v8t.size(1) == HK && v8t.size(3) == NK && v8t.size(2) == D
+                && vs.size(1) == HK && vs.size(3) == NK / 64);
     auto out = at::empty({B, NQ, H, D}, q8.options().dtype(at::kBFloat16));
     auto lse = at::empty({B, H, NQ}, q8.options().dtype(at::kFloat));
     Fp8FwdParams p{};
     p.q = q8.data_ptr(); p.k = k8.data_ptr(); p.vt = v8t.data_ptr();
     p.qs = qs.data_ptr(); p.ks = ks.data_ptr(); p.vs = vs.data_ptr();
     p.out = out.data_ptr(); p.lse = lse.data_ptr<float>();
-    p.b = (int)B; p.h = (int)H; p.nq = NQ; p.nk = NK;
+    p.b = (int)B; p.h = (int)H; p.hk = (int)HK; p.nq = NQ; p.nk = NK;
     p.nvs = (int)(NK / 64);
+    p.causal = causal ? 1 : 0;
     p.scale = sm_scale > 0 ? (float)sm_scale : (float)(1.0 / std::sqrt((double)D));
     launch_attn_fwd_fp8(p, at::hip::getCurrentHIPStream());
     TORCH_CHECK(hipGetLastError() == hipSuccess, "attn_fwd_fp8 launch failed");
@@ -345,7 +348,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
     mod.def("attn_fwd_fp8", &ring_attn::attn_fwd_fp8,
             "CDNA4 MX-FP8 serving forward (e4m3 + e8m0 row scales)",
             py::arg("q8"), py::arg("k8"), py::arg("v8t"), py::arg("qs"),
-            py::arg("ks"), py::arg("vs"), py::arg("sm_scale") = -1.0);
+            py::arg("ks"), py::arg("vs"), py::arg("sm_scale") = -1.0,
+            py::arg("causal") = false);
     mod.def("attn_delta", &ring_attn::attn_delta, "fused delta = rowsum(dO*O) preprocess");
     mod.def("rotary_apply", &ring_attn::rotary_apply, "fused rotary embedding (table-driven)");
 }

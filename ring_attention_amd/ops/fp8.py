@@ -14,8 +14,11 @@ the kernel; only the MFMA operands are 8-bit.  Expected error vs a bf16
 forward is ~1-3% relative on out (P is quantized to e4m3's 3 mantissa
 bits); use the bf16 path when training.
 
-v0 scope (kernel asserts): non-causal, no mask/bias, hk == h, d == 64,
-nq % 256 == 0, nk % 128 == 0, single shard (no ring pass).
+Causal (``causal=True``) uses the bf16 kernels' mirrored paired-tile
+load balance; GQA follows the framework-wide ``qh % hk`` pairing.
+
+v0 scope (kernel asserts): no mask/bias/window, d == 64, nq % 256 == 0,
+nk % 128 == 0 (causal: nq == nk), single shard (no ring pass).
 """
 
 from __future__ import annotations
@@ -60,6 +63,7 @@ def quantize_fp8(q: Tensor, k: Tensor, v: Tensor):
 def flash_attn_fp8(
     q: Tensor, k: Tensor, v: Tensor,
     sm_scale: float | None = None,
+    causal: bool = False,
 ) -> tuple[Tensor, Tensor]:
     """MX-FP8 non-causal attention forward on (b, n, h, d) tensors.
 
@@ -70,9 +74,12 @@ def flash_attn_fp8(
     q8, k8, v8t, qs, ks, vs = quantize_fp8(q, k, v)
     return flash_attn_fp8_quantized(
         q8, k8, v8t, qs, ks, vs,
-        sm_scale if sm_scale is not None else q.shape[-1] ** -0.5)
+        sm_scale if sm_scale is not None else q.shape[-1] ** -0.5,
+        causal=causal)
 
 
-def flash_attn_fp8_quantized(q8, k8, v8t, qs, ks, vs, sm_scale: float):
-    out, lse = hip_ext.require().attn_fwd_fp8(q8, k8, v8t, qs, ks, vs, sm_scale)
+def flash_attn_fp8_quantized(q8, k8, v8t, qs, ks, vs, sm_scale: float,
+                             causal: bool = False):
+    out, lse = hip_ext.require().attn_fwd_fp8(q8, k8, v8t, qs, ks, vs,
+                                              sm_scale, causal)
     return out, lse

@@ -604,3 +604,30 @@ def test_flash_attn_fp8():
     assert mx < 0.15, f"fp8 out max err {mx}"
     lse_err = (lse.cpu() - ref_lse).abs().max().item()
     assert lse_err < 0.06, f"fp8 lse err {lse_err}"
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("causal,groups", [(True, 1), (False, 4), (True, 2)])
+def test_flash_attn_fp8_causal_gqa(causal, groups):
+    from ring_attention_amd.ops.fp8 import flash_attn_fp8
+    b, n, h, d = 1, 512, 4, 64
+    hk = h // groups
+    torch.manual_seed(11)
+    q = torch.randn(b, n, h, d, device="cuda", dtype=torch.bfloat16)
+    k = torch.randn(b, n, hk, d, device="cuda", dtype=torch.bfloat16)
+    v = torch.randn(b, n, hk, d, device="cuda", dtype=torch.bfloat16)
+    out, lse = flash_attn_fp8(q, k, v, causal=causal)
+    qf = q.float().cpu()
+    kf = k.float().cpu().repeat(1, 1, groups, 1)
+    vf = v.float().cpu().repeat(1, 1, groups, 1)
+    sim = torch.einsum("bihd,bjhd->bhij", qf, kf) * d ** -0.5
+    if causal:
+        pos = torch.arange(n)
+        sim = sim.masked_fill((pos[None, :] > pos[:, None])[None, None], float("-inf"))
+    ref = torch.einsum("bhij,bjhd->bihd", sim.softmax(-1), vf)
+    ref_lse = sim.logsumexp(dim=-1)
+    o = out.float().cpu()
+    rel = (o - ref).abs().mean().item() / ref.abs().mean().item()
+    assert rel < 0.09, f"fp8 out mean rel err {rel}"
+    assert (o - ref).abs().max().item() < 0.2
+    assert (lse.cpu() - ref_lse).abs().max().item() < 0.06
