@@ -151,9 +151,9 @@ struct Fp8FwdParams {
     const void* q;      // e4m3 bytes (B, Nq, H, D)
     const void* k;      // e4m3 bytes (B, Nk, H, D)
     const void* vt;     // e4m3 bytes (B, H, D, Nk)  — host-pre-transposed V
-    const void* qs;     // e8m0 bytes (B, Nq, H): per-row scale exponent+127
-    const void* ks;     // e8m0 bytes (B, Nk, H)
-    const void* vs;     // e8m0 bytes (B, H, D, Nvs): per (d row, 64-kv chunk)
+    const void* qs;     // e8m0 bytes (B, Nq, H, D/64): per (row, 64-d chunk)
+    const void* ks;     // e8m0 bytes (B, Nk, HK, D/64)
+    const void* vs;     // e8m0 bytes (B, HK, D, Nvs): per (d row, 64-kv chunk)
     void* out;          // bf16 (B, Nq, H, D)
     float* lse;         // fp32 (B, H, Nq)
     int b, h, hk;       // GQA: kv heads (qh pairs qh % hk)
@@ -164,7 +164,7 @@ struct Fp8FwdParams {
     int paired;         // total q tiles T (causal launcher)
 };
 
-void launch_attn_fwd_fp8(const Fp8FwdParams& p, hipStream_t stream);
+void launch_attn_fwd_fp8(const Fp8FwdParams& p, int head_dim, hipStream_t stream);
 
 struct RotaryParams {
     const void* x;      // bf16 (B, N, H, D)

@@ -29,7 +29,7 @@
 // kv head qh % hk (reference tile convention).
 //
 // Scope (asserted in the binding): no mask/bias/window/softclamp, single
-// shot (no ring resume), D = 64, nq % 256 == 0, nk % 128 == 0.
+// shot (no ring resume), D in {64, 128}, nq % 256 == 0, nk % 128 == 0.
 
 #include <hip/hip_runtime.h>
 #include <hip/hip_bf16.h>
@@ -62,24 +62,26 @@ __device__ __forceinline__ float fp8_cross_half(float x) {
     return (threadIdx.x & 32) ? lo.f : hi.f;
 }
 
+template <int D>
 struct Fp8Lds {
-    __align__(16) unsigned char k[2][FP8_KVBLK * 64];    // [kv][d] bytes
-    __align__(16) unsigned char vt[2][64 * FP8_KVBLK];   // [d][kv] bytes
-    unsigned char ks[2][FP8_KVBLK];                      // k row e8m0
+    __align__(16) unsigned char k[2][FP8_KVBLK * D];     // [kv][d] bytes
+    __align__(16) unsigned char vt[2][D * FP8_KVBLK];    // [d][kv] bytes
+    unsigned char ks[2][FP8_KVBLK * (D / 64)];           // k (row, chunk) e8m0
 };
 
 template <class F>
 __device__ __attribute__((noinline)) void fp8_noinline_call(F&& f) { f(); }
 
-template <bool PAIRED>
+template <int D, bool PAIRED>
 __global__ __launch_bounds__(FP8_NTHREADS, 1)
 void attn_fwd_fp8_kernel(Fp8FwdParams p) {
-    constexpr int D = 64;
-    constexpr int DBLK = D / 32;          // 2
+    constexpr int DBLK = D / 32;
+    constexpr int NDCH = D / 64;          // 64-deep MFMA k-chunks per head dim
+    constexpr int KCH = D * 2 / 32;       // 16B LDS chunks per K row (D bytes)
     constexpr float LOG2E_ = 1.4426950408889634f;
     constexpr float LN2_ = 0.6931471805599453f;
 
-    __shared__ Fp8Lds lds;
+    __shared__ Fp8Lds<D> lds;
 
     const int tid = threadIdx.x;
     const int lane = tid & 63;
@@ -106,13 +108,19 @@ void attn_fwd_fp8_kernel(Fp8FwdParams p) {
     const Fp8FwdParams P = p;             // register-local (noinline frame)
     const long i = (long)qtile * FP8_QROWS_WG + wid * FP8_QROWS_WAVE + l31;
 
-    // ---- Q fragment: 32 e4m3 bytes (d = 32*lhi .. +31) + row scale
+    // ---- Q fragments: per 64-d chunk, 32 e4m3 bytes (d = 64*dc + 32*lhi
+    // .. +31) + per-chunk row scale
     const unsigned char* qrow = (const unsigned char*)P.q
         + ((long)b * P.nq + i) * P.h * D + (long)h * D + 32 * lhi;
-    union { i32x8_ v; uint4 u4[2]; } qf;
-    qf.u4[0] = *(const uint4*)qrow;
-    qf.u4[1] = *(const uint4*)(qrow + 16);
-    const int qs = ((const unsigned char*)P.qs)[((long)b * P.nq + i) * P.h + h];
+    union { i32x8_ v; uint4 u4[2]; } qf[NDCH];
+    int qs[NDCH];
+    #pragma unroll
+    for (int dc = 0; dc < NDCH; ++dc) {
+        qf[dc].u4[0] = *(const uint4*)(qrow + dc * 64);
+        qf[dc].u4[1] = *(const uint4*)(qrow + dc * 64 + 16);
+        qs[dc] = ((const unsigned char*)P.qs)[
+            (((long)b * P.nq + i) * P.h + h) * NDCH + dc];
+    }
 
     // ---- accumulators
     float m_run = MASK_VALUE_F, l_run = 0.f;
@@ -133,34 +141,46 @@ void attn_fwd_fp8_kernel(Fp8FwdParams p) {
     const unsigned char* vtbase = (const unsigned char*)P.vt
         + (((long)b * P.hk + hk) * D) * P.nk;
     const unsigned char* ksbase = (const unsigned char*)P.ks
-        + ((long)b * P.nk) * P.hk + hk;
+        + (((long)b * P.nk) * P.hk + hk) * NDCH;
 
+    // per-tile staging: K = KVBLK*D bytes, V^T = D*KVBLK bytes — KREGS
+    // uint4 per thread each; ks = KVBLK*NDCH bytes
+    constexpr int KREGS = FP8_KVBLK * D / 16 / FP8_NTHREADS;
+    constexpr int VCH = FP8_KVBLK / 16;   // 16B chunks per V^T row
     const long k_row_stride = (long)P.hk * D;
-    const unsigned char* kptr = kbase + (tid / 4) * k_row_stride + (tid % 4) * 16;
-    const unsigned char* vptr = vtbase + (long)(tid / 8) * P.nk + (tid % 8) * 16;
-    const unsigned char* ksptr = ksbase + (long)tid * P.hk;
+    const unsigned char* kptr = kbase + (tid / KCH) * k_row_stride + (tid % KCH) * 16;
+    const unsigned char* vptr = vtbase + (long)(tid / VCH) * P.nk + (tid % VCH) * 16;
+    const unsigned char* ksptr = ksbase + (long)(tid / NDCH) * P.hk * NDCH + (tid % NDCH);
 
-    uint4 kst, vst;
+    uint4 kst[KREGS], vst[KREGS];
     unsigned char ksst = 0;
 
     auto load_tile = [&]() {
-        kst = *(const uint4*)kptr;
-        vst = *(const uint4*)vptr;
-        if (tid < FP8_KVBLK) ksst = *ksptr;
+        #pragma unroll
+        for (int r = 0; r < KREGS; ++r) {
+            kst[r] = *(const uint4*)(kptr + (long)(r * (FP8_NTHREADS / KCH)) * k_row_stride);
+            vst[r] = *(const uint4*)(vptr + (long)(r * (FP8_NTHREADS / VCH)) * P.nk);
+        }
+        if (tid < FP8_KVBLK * NDCH) ksst = *ksptr;
         kptr += (long)FP8_KVBLK * k_row_stride;
         vptr += FP8_KVBLK;
-        ksptr += (long)FP8_KVBLK * P.hk;
+        ksptr += (long)FP8_KVBLK * P.hk * NDCH;
     };
     auto write_tile = [&](int par) {
-        {   // K: row = tid/4 (kv), chunk = tid%4 of 4 (64 B rows)
-            int row = tid / 4, ch = tid % 4;
-            *(uint4*)(lds.k[par] + row * 64 + fswz<4>(row, ch) * 16) = kst;
+        #pragma unroll
+        for (int r = 0; r < KREGS; ++r) {
+            {   // K: D-byte rows, KCH 16B chunks
+                int c = tid + r * FP8_NTHREADS;
+                int row = c / KCH, ch = c % KCH;
+                *(uint4*)(lds.k[par] + row * D + fswz<KCH>(row, ch) * 16) = kst[r];
+            }
+            {   // V^T: KVBLK-byte rows, VCH chunks
+                int c = tid + r * FP8_NTHREADS;
+                int row = c / VCH, ch = c % VCH;
+                *(uint4*)(lds.vt[par] + row * FP8_KVBLK + fswz<8>(row, ch) * 16) = vst[r];
+            }
         }
-        {   // V^T: row = tid/8 (d), chunk = tid%8 of 8 (128 B rows)
-            int row = tid / 8, ch = tid % 8;
-            *(uint4*)(lds.vt[par] + row * FP8_KVBLK + fswz<8>(row, ch) * 16) = vst;
-        }
-        if (tid < FP8_KVBLK) lds.ks[par][tid] = ksst;
+        if (tid < FP8_KVBLK * NDCH) lds.ks[par][tid] = ksst;
     };
 
     const float scale2 = P.scale * LOG2E_;
@@ -178,19 +198,26 @@ void attn_fwd_fp8_kernel(Fp8FwdParams p) {
 
         __syncthreads();
 
-        // ---- QK^T: ONE scaled MFMA per 32-kv block (K = 64 = whole d)
+        // ---- QK^T: one scaled MFMA per (32-kv block, 64-d chunk); chunks
+        // chain through the fp32 accumulator (each dequants its own scales)
         f32x16 s[FP8_NBLK];
         __builtin_amdgcn_s_setprio(1);
         #pragma unroll
         for (int kb = 0; kb < FP8_NBLK; ++kb) {
             int krow = kb * 32 + l31;
-            union { i32x8_ v; uint4 u4[2]; } kf;
-            kf.u4[0] = *(const uint4*)(lds.k[par] + krow * 64 + fswz<4>(krow, 2 * lhi) * 16);
-            kf.u4[1] = *(const uint4*)(lds.k[par] + krow * 64 + fswz<4>(krow, 2 * lhi + 1) * 16);
-            int sa = lds.ks[par][krow];
             f32x16 acc = {};
-            s[kb] = __builtin_amdgcn_mfma_scale_f32_32x32x64_f8f6f4(
-                kf.v, qf.v, acc, 0, 0, 0, sa, 0, qs);
+            #pragma unroll
+            for (int dc = 0; dc < NDCH; ++dc) {
+                union { i32x8_ v; uint4 u4[2]; } kf;
+                kf.u4[0] = *(const uint4*)(lds.k[par] + krow * D
+                               + fswz<KCH>(krow, 4 * dc + 2 * lhi) * 16);
+                kf.u4[1] = *(const uint4*)(lds.k[par] + krow * D
+                               + fswz<KCH>(krow, 4 * dc + 2 * lhi + 1) * 16);
+                int sa = lds.ks[par][krow * NDCH + dc];
+                acc = __builtin_amdgcn_mfma_scale_f32_32x32x64_f8f6f4(
+                    kf.v, qf[dc].v, acc, 0, 0, 0, sa, 0, qs[dc]);
+            }
+            s[kb] = acc;
         }
         __builtin_amdgcn_s_setprio(0);
 
@@ -327,17 +354,23 @@ void attn_fwd_fp8_kernel(Fp8FwdParams p) {
     }  // pair loop
 }
 
-void launch_attn_fwd_fp8(const Fp8FwdParams& p, hipStream_t stream) {
+void launch_attn_fwd_fp8(const Fp8FwdParams& p, int head_dim, hipStream_t stream) {
     const long T = p.nq / FP8_QROWS_WG;
     dim3 block(FP8_NTHREADS);
     if (p.causal) {
         Fp8FwdParams pc = p;
         pc.paired = (int)T;
         dim3 grid((unsigned)((T + 1) / 2), (unsigned)(p.b * p.h));
-        hipLaunchKernelGGL(attn_fwd_fp8_kernel<true>, grid, block, 0, stream, pc);
+        if (head_dim == 64)
+            hipLaunchKernelGGL((attn_fwd_fp8_kernel<64, true>), grid, block, 0, stream, pc);
+        else
+            hipLaunchKernelGGL((attn_fwd_fp8_kernel<128, true>), grid, block, 0, stream, pc);
     } else {
         dim3 grid((unsigned)T, (unsigned)(p.b * p.h));
-        hipLaunchKernelGGL(attn_fwd_fp8_kernel<false>, grid, block, 0, stream, p);
+        if (head_dim == 64)
+            hipLaunchKernelGGL((attn_fwd_fp8_kernel<64, false>), grid, block, 0, stream, p);
+        else
+            hipLaunchKernelGGL((attn_fwd_fp8_kernel<128, false>), grid, block, 0, stream, p);
     }
 }
 

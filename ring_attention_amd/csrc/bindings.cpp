@@ -284,13 +284,16 @@ std::vector<at::Tensor> attn_fwd_fp8(at::Tensor q8, at::Tensor k8, at::Tensor v8
     }
     const int64_t B = q8.size(0), NQ = q8.size(1), H = q8.size(2), D = q8.size(3);
     const int64_t NK = k8.size(1), HK = k8.size(2);
-    TORCH_CHECK(D == 64, "fp8 path: head dim 64 only (v0)");
+    TORCH_CHECK(D == 64 || D == 128, "fp8 path: head dim 64 or 128");
     TORCH_CHECK(H % HK == 0, "fp8 path: q heads must be a multiple of kv heads");
     TORCH_CHECK(NQ % 256 == 0, "fp8 path: nq must be a multiple of 256 (v0)");
     TORCH_CHECK(NK % 128 == 0, "fp8 path: nk must be a multiple of 128 (v0)");
     TORCH_CHECK(!causal || NQ == NK, "fp8 causal: nq == nk (v0)");
     TORCH_CHECK(v8t.size(1) == HK && v8t.size(3) == NK && v8t.size(2) == D
                 && vs.size(1) == HK && vs.size(3) == NK / 64);
+    TORCH_CHECK(qs.numel() == B * NQ * H * (D / 64)
+                && ks.numel() == B * NK * HK * (D / 64),
+                "q/k scales must be per (row, 64-d chunk)");
     auto out = at::empty({B, NQ, H, D}, q8.options().dtype(at::kBFloat16));
     auto lse = at::empty({B, H, NQ}, q8.options().dtype(at::kFloat));
     Fp8FwdParams p{};
@@ -301,7 +304,7 @@ std::vector<at::Tensor> attn_fwd_fp8(at::Tensor q8, at::Tensor k8, at::Tensor v8
     p.nvs = (int)(NK / 64);
     p.causal = causal ? 1 : 0;
     p.scale = sm_scale > 0 ? (float)sm_scale : (float)(1.0 / std::sqrt((double)D));
-    launch_attn_fwd_fp8(p, at::hip::getCurrentHIPStream());
+    launch_attn_fwd_fp8(p, (int)D, at::hip::getCurrentHIPStream());
     TORCH_CHECK(hipGetLastError() == hipSuccess, "attn_fwd_fp8 launch failed");
     return {out, lse};
 }

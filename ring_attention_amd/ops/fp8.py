@@ -17,8 +17,9 @@ bits); use the bf16 path when training.
 Causal (``causal=True``) uses the bf16 kernels' mirrored paired-tile
 load balance; GQA follows the framework-wide ``qh % hk`` pairing.
 
-v0 scope (kernel asserts): no mask/bias/window, d == 64, nq % 256 == 0,
-nk % 128 == 0 (causal: nq == nk), single shard (no ring pass).
+v0 scope (kernel asserts): no mask/bias/window, d in {64, 128},
+nq % 256 == 0, nk % 128 == 0 (causal: nq == nk), single shard (no ring
+pass).
 """
 
 from __future__ import annotations
@@ -40,14 +41,19 @@ def quantize_fp8(q: Tensor, k: Tensor, v: Tensor):
     """Quantize (b, n, h, d) q/k/v for attn_fwd_fp8.
 
     Returns (q8, k8, v8t, qs, ks, vs) uint8 views:
-      q8/k8 (b, n, h, d) e4m3; v8t (b, h, d, n) e4m3 (pre-transposed);
-      qs/ks (b, n, h) e8m0 row scales; vs (b, h, d, n // 64) e8m0.
+      q8/k8 (b, n, h, d) e4m3; v8t (b, hk, d, n) e4m3 (pre-transposed);
+      qs/ks (b, n, h, d // 64) e8m0 row-chunk scales;
+      vs (b, hk, d, n // 64) e8m0.
     """
     qf, kf = q.float(), k.float()
-    qs_b, qs_s = _e8m0(qf.abs().amax(dim=-1))
-    ks_b, ks_s = _e8m0(kf.abs().amax(dim=-1))
-    q8 = (qf / qs_s.unsqueeze(-1)).to(torch.float8_e4m3fn)
-    k8 = (kf / ks_s.unsqueeze(-1)).to(torch.float8_e4m3fn)
+    d = q.shape[-1]
+    assert d % 64 == 0
+    nd = d // 64
+    # per (row, 64-d chunk) scales: shape (..., nd)
+    qs_b, qs_s = _e8m0(qf.view(*qf.shape[:-1], nd, 64).abs().amax(dim=-1))
+    ks_b, ks_s = _e8m0(kf.view(*kf.shape[:-1], nd, 64).abs().amax(dim=-1))
+    q8 = (qf / qs_s.repeat_interleave(64, dim=-1)).to(torch.float8_e4m3fn)
+    k8 = (kf / ks_s.repeat_interleave(64, dim=-1)).to(torch.float8_e4m3fn)
 
     vt = v.permute(0, 2, 3, 1).float().contiguous()      # (b, h, d, n)
     b, h, d, n = vt.shape

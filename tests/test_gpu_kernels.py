@@ -607,10 +607,10 @@ def test_flash_attn_fp8():
 
 
 @pytest.mark.gpu
-@pytest.mark.parametrize("causal,groups", [(True, 1), (False, 4), (True, 2)])
-def test_flash_attn_fp8_causal_gqa(causal, groups):
+@pytest.mark.parametrize("causal,groups,d", [(True, 1, 64), (False, 4, 64), (True, 2, 64), (False, 1, 128), (True, 2, 128)])
+def test_flash_attn_fp8_causal_gqa(causal, groups, d):
     from ring_attention_amd.ops.fp8 import flash_attn_fp8
-    b, n, h, d = 1, 512, 4, 64
+    b, n, h = 1, 512, 4
     hk = h // groups
     torch.manual_seed(11)
     q = torch.randn(b, n, h, d, device="cuda", dtype=torch.bfloat16)
