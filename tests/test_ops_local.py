@@ -189,3 +189,30 @@ def test_head_dim_padding_helpers():
     sim = sim.masked_fill((pos[None, :] > pos[:, None])[None, None], float("-inf"))
     out = torch.einsum("bhij,bjhd->bihd", sim.softmax(-1), vp.float())[..., :d]
     assert (out - ref).abs().max().item() < 1e-5
+
+
+def test_fp8_quantize_roundtrip_cpu():
+    # quantize_fp8 is pure torch — verify scale math and layouts on CPU:
+    # dequantized operands must match the originals within e4m3 resolution
+    import torch
+    from ring_attention_amd.ops.fp8 import quantize_fp8
+    torch.manual_seed(3)
+    b, n, h, hk, d = 2, 128, 4, 2, 128
+    q = torch.randn(b, n, h, d, dtype=torch.bfloat16) * 5
+    k = torch.randn(b, n, hk, d, dtype=torch.bfloat16)
+    v = torch.randn(b, n, hk, d, dtype=torch.bfloat16) * 0.1
+    q8, k8, v8t, qs, ks, vs = quantize_fp8(q, k, v)
+    assert q8.shape == (b, n, h, d) and qs.shape == (b, n, h, d // 64)
+    assert ks.shape == (b, n, hk, d // 64)
+    assert v8t.shape == (b, hk, d, n) and vs.shape == (b, hk, d, n // 64)
+    qdq = q8.view(torch.float8_e4m3fn).float() * torch.exp2(
+        qs.float() - 127).repeat_interleave(64, dim=-1)
+    rel = (qdq - q.float()).abs().max() / q.float().abs().max()
+    assert rel < 0.07, f"q roundtrip rel {rel}"
+    vdq = v8t.view(torch.float8_e4m3fn).float() * torch.exp2(
+        vs.float() - 127).repeat_interleave(64, dim=-1)
+    vref = v.permute(0, 2, 3, 1).float()
+    assert (vdq - vref).abs().max() / vref.abs().max() < 0.07
+    # e4m3 range respected: no inf/nan bytes (0x7f/0xff are nan in e4m3fn)
+    for t in (q8, k8, v8t):
+        assert not ((t == 0x7F) | (t == 0xFF)).any()
