@@ -33,6 +33,7 @@ __all__ = [
     "flash_attn",
     "flash_attn_offset",
     "flash_attn_fp8",
+    "ring_flash_attn_fp8",
     "quantize_fp8",
     "RingAttention",
     "RingTransformer",
@@ -53,7 +54,7 @@ def __getattr__(name):
                 "flash_attn_offset"):
         from .ops import ring_flash_hip
         return getattr(ring_flash_hip, name)
-    if name in ("flash_attn_fp8", "quantize_fp8"):
+    if name in ("flash_attn_fp8", "ring_flash_attn_fp8", "quantize_fp8"):
         from .ops import fp8
         return getattr(fp8, name)
     raise AttributeError(f"module {__name__!r} has no attribute {name!r}")

@@ -89,3 +89,49 @@ def flash_attn_fp8_quantized(q8, k8, v8t, qs, ks, vs, sm_scale: float,
     out, lse = hip_ext.require().attn_fwd_fp8(q8, k8, v8t, qs, ks, vs,
                                               sm_scale, causal)
     return out, lse
+
+
+@torch.no_grad()
+def ring_flash_attn_fp8(
+    q: Tensor, k: Tensor, v: Tensor,
+    sm_scale: float | None = None,
+    ring_size: int | None = None,
+) -> tuple[Tensor, Tensor]:
+    """Non-causal RING attention forward in MX-FP8 over sharded KV.
+
+    Each rank quantizes its LOCAL (b, n_shard, hk, d) k/v once; the 8-bit
+    shards then rotate the ring — half the bf16 wire bytes per hop (xGMI
+    ring hops are single-link bound, so the fp8 wire format directly halves
+    the transport time too) — and every hop's fused-fp8 partial merges by
+    logsumexp (exact online-softmax combine in fp32).
+
+    Returns (out bf16 (b, n, h, d), lse fp32 (b, h, n)).  Causal ring
+    geometry (per-hop diagonals) stays on the bf16 path for now.
+    """
+    from ..parallel import RingTopology, all_ring_pass, is_distributed
+    from ..parallel.ring_pass import null_ring_pass
+
+    sm = sm_scale if sm_scale is not None else q.shape[-1] ** -0.5
+    q8, k8, v8t, qs, ks, vs = quantize_fp8(q, k, v)
+
+    if is_distributed():
+        topo = RingTopology(ring_size)
+        hops = all_ring_pass(topo, k8, v8t, ks, vs)
+    else:
+        hops = null_ring_pass(k8, v8t, ks, vs)
+
+    out = lse = None
+    for _info, (k8h, v8th, ksh, vsh) in hops:
+        o_h, l_h = flash_attn_fp8_quantized(q8, k8h, v8th, qs, ksh, vsh, sm)
+        if out is None:
+            out, lse = o_h.float(), l_h
+        else:
+            m = torch.maximum(lse, l_h)
+            wa = (lse - m).exp()                      # (b, h, n)
+            wb = (l_h - m).exp()
+            den = wa + wb
+            wa4 = (wa / den).permute(0, 2, 1).unsqueeze(-1)   # (b, n, h, 1)
+            wb4 = (wb / den).permute(0, 2, 1).unsqueeze(-1)
+            out = out * wa4 + o_h.float() * wb4
+            lse = m + den.log()
+    return out.to(q.dtype), lse

@@ -467,3 +467,34 @@ def _ring_immutability_case(rank, world):
 def test_ring_engine_input_immutability(world):
     from .loopback_dist import loopback_world
     assert all(loopback_world(world, lambda r: _ring_immutability_case(r, world)))
+
+
+@pytest.mark.gpu
+def test_loopback_ring_fp8():
+    # MX-FP8 ring forward at world 2 (8-bit shards on the wire, per-hop
+    # logsumexp merge) vs the single-shard fp8 forward on the full KV
+    from ring_attention_amd.ops.fp8 import flash_attn_fp8, ring_flash_attn_fp8
+    world = 2
+    b, n, h, d = 1, 1024, 4, 64
+    torch.manual_seed(23)
+    q = torch.randn(b, n, h, d, device="cuda", dtype=torch.bfloat16)
+    k = torch.randn(b, n, h, d, device="cuda", dtype=torch.bfloat16)
+    v = torch.randn(b, n, h, d, device="cuda", dtype=torch.bfloat16)
+    ref, ref_lse = flash_attn_fp8(q, k, v)
+
+    def run(rank):
+        qs = q.chunk(world, dim=1)[rank]
+        ks = k.chunk(world, dim=1)[rank]
+        vs = v.chunk(world, dim=1)[rank]
+        return ring_flash_attn_fp8(qs, ks, vs)
+
+    results = loopback_world(world, run)
+    for rank, (out, lse) in enumerate(results):
+        want = ref.chunk(world, dim=1)[rank]
+        e = (out.float() - want.float()).abs().max().item()
+        s = want.float().abs().max().item() + 1e-6
+        # both sides quantize (shard-local scales differ slightly from
+        # full-tensor scales) — allow e4m3-level disagreement
+        assert e / s < 8e-2, f"rank {rank} fp8 ring rel err {e/s}"
+        want_lse = ref_lse.chunk(world, dim=2)[rank]
+        assert (lse - want_lse).abs().max().item() < 0.1
