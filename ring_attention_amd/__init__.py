@@ -19,7 +19,7 @@ from .models import (
     RingTransformer,
     apply_rotary_pos_emb,
 )
-from .tree_decode import tree_attn_decode
+from .tree_decode import tree_attn_decode, tree_attn_decode_fp8
 from .zigzag import zig_zag_attn, zig_zag_pad_seq, zig_zag_shard
 
 __version__ = "0.1.0"
@@ -42,6 +42,7 @@ __all__ = [
     "RMSNorm",
     "FeedForward",
     "tree_attn_decode",
+    "tree_attn_decode_fp8",
     "zig_zag_attn",
     "zig_zag_pad_seq",
     "zig_zag_shard",

@@ -143,9 +143,12 @@ struct DecodeParams {
     long n;
     float scale;
     long chunks;            // kv-split S (0 = auto)
+    const void* kscale;     // fp8 mode: e8m0 per kv row (B, HK, N)
+    const void* vscale;     // fp8 mode: e8m0 per kv row (B, HK, N)
 };
 
 void launch_decode_partial(const DecodeParams& p, int head_dim, hipStream_t stream);
+void launch_decode_partial_fp8(const DecodeParams& p, int head_dim, hipStream_t stream);
 
 struct Fp8FwdParams {
     const void* q;      // e4m3 bytes (B, Nq, H, D)

@@ -309,6 +309,38 @@ std::vector<at::Tensor> attn_fwd_fp8(at::Tensor q8, at::Tensor k8, at::Tensor v8
     return {out, lse};
 }
 
+std::vector<at::Tensor> decode_partial_fp8(at::Tensor q, at::Tensor k8, at::Tensor v8,
+                                           at::Tensor ks, at::Tensor vs,
+                                           double sm_scale) {
+    // FP8 KV-cache decode partial: q bf16 (B,HQ,NQ,D); k8/v8 e4m3 u8
+    // (B,HK,N,D) with per-row e8m0 scales ks/vs (B,HK,N).  Same output
+    // contract as decode_partial: (out fp32 (S,B,HQ,NQ,D), lse (S,B,HQ,NQ,1)).
+    CHECK_BF16_CONTIG(q);
+    for (auto* t : {&k8, &v8, &ks, &vs})
+        TORCH_CHECK(t->scalar_type() == at::kByte && t->is_contiguous(),
+                    "fp8 cache operands must be contiguous uint8");
+    const int64_t B = q.size(0), H = q.size(1), NQ = q.size(2), D = q.size(3);
+    const int64_t HK = k8.size(1), N = k8.size(2);
+    TORCH_CHECK(D == 64 || D == 128, "head dim must be 64 or 128");
+    TORCH_CHECK(H % HK == 0, "q heads must be a multiple of kv heads");
+    TORCH_CHECK(ks.numel() == B * HK * N && vs.numel() == B * HK * N);
+    int64_t waves = B * H * NQ;
+    int64_t chunks = std::max<int64_t>(
+        1, std::min<int64_t>(N / 1024 + 1, 1024 / std::max<int64_t>(waves / 4, 1)));
+    auto out = at::empty({chunks, B, H, NQ, D}, q.options().dtype(at::kFloat));
+    auto lse = at::empty({chunks, B, H, NQ, 1}, q.options().dtype(at::kFloat));
+    DecodeParams p{};
+    p.q = q.data_ptr(); p.k = k8.data_ptr(); p.v = v8.data_ptr();
+    p.kscale = ks.data_ptr(); p.vscale = vs.data_ptr();
+    p.out = out.data_ptr<float>(); p.lse = lse.data_ptr<float>();
+    p.b = (int)B; p.h = (int)H; p.hk = (int)HK; p.nq = (int)NQ; p.n = N;
+    p.scale = sm_scale > 0 ? (float)sm_scale : (float)(1.0 / std::sqrt((double)D));
+    p.chunks = chunks;
+    launch_decode_partial_fp8(p, (int)D, at::hip::getCurrentHIPStream());
+    TORCH_CHECK(hipGetLastError() == hipSuccess, "decode fp8 launch failed");
+    return {out, lse};
+}
+
 at::Tensor attn_delta(at::Tensor dout, at::Tensor out) {
     // dout, out (B,N,H,D) bf16 -> delta fp32 (B,H,N) = rowsum(dout*out)
     CHECK_BF16_CONTIG(dout); CHECK_BF16_CONTIG(out);
@@ -353,6 +385,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
             py::arg("q8"), py::arg("k8"), py::arg("v8t"), py::arg("qs"),
             py::arg("ks"), py::arg("vs"), py::arg("sm_scale") = -1.0,
             py::arg("causal") = false);
+    mod.def("decode_partial_fp8", &ring_attn::decode_partial_fp8,
+            "CDNA4 FP8 KV-cache decode partial",
+            py::arg("q"), py::arg("k8"), py::arg("v8"), py::arg("ks"),
+            py::arg("vs"), py::arg("sm_scale") = -1.0);
     mod.def("attn_delta", &ring_attn::attn_delta, "fused delta = rowsum(dO*O) preprocess");
     mod.def("rotary_apply", &ring_attn::rotary_apply, "fused rotary embedding (table-driven)");
 }

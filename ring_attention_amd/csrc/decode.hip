@@ -13,6 +13,8 @@
 
 namespace ring_attn {
 
+typedef float f32x2 __attribute__((ext_vector_type(2)));
+
 template <int D>
 __global__ __launch_bounds__(256) void decode_partial_kernel(DecodeParams p) {
     // one wave per (b, q-head, query token); 4 waves per block; blockIdx.y
@@ -125,6 +127,130 @@ void launch_decode_partial(const DecodeParams& p, int head_dim, hipStream_t stre
         hipLaunchKernelGGL(decode_partial_kernel<64>, grid, block, 0, stream, p);
     } else if (head_dim == 128) {
         hipLaunchKernelGGL(decode_partial_kernel<128>, grid, block, 0, stream, p);
+    } else {
+        __builtin_trap();
+    }
+}
+
+
+// ---------------------------------------------------------------------------
+// FP8 KV-cache decode: same structure, 8-bit KV stream (half the HBM bytes —
+// decode is bandwidth-bound, measured 1.5-1.6 TB/s on the bf16 stream).
+// k8/v8 are e4m3 rows with ONE e8m0 scale per (kv row): the k scale folds
+// into the score (s_true = 2^ek * dot(q, k8)) and the v scale folds into the
+// softmax weight (acc += (w * 2^ev) * v8) — the inner loop does the same
+// fp32 math as the bf16 kernel, only the loads halve.  q stays bf16.
+// ---------------------------------------------------------------------------
+template <int D>
+__global__ __launch_bounds__(256) void decode_partial_fp8_kernel(DecodeParams p) {
+    const int rows = p.b * p.h * p.nq;
+    const int wave_global = (blockIdx.x * 4) + (threadIdx.x >> 6);
+    if (wave_global >= rows) return;
+    const int G = p.h / p.hk;
+    const int iq = wave_global % p.nq;
+    int r = wave_global / p.nq;
+    const int g = r % G;  r /= G;
+    const int hk = r % p.hk;
+    const int b = r / p.hk;
+    const int h = hk + g * p.hk;
+    const int bh = b * p.h + h;
+    const int lane = threadIdx.x & 63;
+    const int chunk = blockIdx.y;
+    const long per = (p.n + gridDim.y - 1) / gridDim.y;
+    const long j_lo = chunk * per;
+    const long j_hi = min(p.n, j_lo + per);
+    const long part_off = (long)chunk * rows;
+
+    const __bf16* qp = (const __bf16*)p.q + (((long)b * p.h + h) * p.nq + iq) * D;
+    const unsigned char* kp = (const unsigned char*)p.k + ((long)b * p.hk + hk) * p.n * D;
+    const unsigned char* vp = (const unsigned char*)p.v + ((long)b * p.hk + hk) * p.n * D;
+    const unsigned char* ksp = (const unsigned char*)p.kscale + ((long)b * p.hk + hk) * p.n;
+    const unsigned char* vsp = (const unsigned char*)p.vscale + ((long)b * p.hk + hk) * p.n;
+
+    float qreg[D];
+    #pragma unroll
+    for (int d = 0; d < D; ++d) qreg[d] = (float)qp[d];
+
+    float m = MASK_VALUE_F, l = 0.f;
+    float acc[D];
+    #pragma unroll
+    for (int d = 0; d < D; ++d) acc[d] = 0.f;
+
+    for (long j = j_lo + lane; j < j_hi; j += 64) {
+        const unsigned char* krow = kp + j * D;
+        float s = 0.f;
+        #pragma unroll
+        for (int d = 0; d < D; d += 16) {
+            uint4 k16 = *(const uint4*)(krow + d);
+            const unsigned* kw = (const unsigned*)&k16;
+            #pragma unroll
+            for (int w4 = 0; w4 < 4; ++w4) {
+                f32x2 lo = __builtin_amdgcn_cvt_pk_f32_fp8(kw[w4], false);
+                f32x2 hi = __builtin_amdgcn_cvt_pk_f32_fp8(kw[w4], true);
+                s += qreg[d + 4 * w4 + 0] * lo[0] + qreg[d + 4 * w4 + 1] * lo[1]
+                   + qreg[d + 4 * w4 + 2] * hi[0] + qreg[d + 4 * w4 + 3] * hi[1];
+            }
+        }
+        s *= p.scale * exp2f((float)ksp[j] - 127.f);
+        float m_new = fmaxf(m, s);
+        float alpha = __expf(m - m_new);
+        float w = __expf(s - m_new) * exp2f((float)vsp[j] - 127.f);
+        l = l * alpha + __expf(s - m_new);
+        const unsigned char* vrow = vp + j * D;
+        #pragma unroll
+        for (int d = 0; d < D; d += 16) {
+            uint4 v16 = *(const uint4*)(vrow + d);
+            const unsigned* vw = (const unsigned*)&v16;
+            #pragma unroll
+            for (int w4 = 0; w4 < 4; ++w4) {
+                f32x2 lo = __builtin_amdgcn_cvt_pk_f32_fp8(vw[w4], false);
+                f32x2 hi = __builtin_amdgcn_cvt_pk_f32_fp8(vw[w4], true);
+                int base = d + 4 * w4;
+                acc[base + 0] = acc[base + 0] * alpha + w * lo[0];
+                acc[base + 1] = acc[base + 1] * alpha + w * lo[1];
+                acc[base + 2] = acc[base + 2] * alpha + w * hi[0];
+                acc[base + 3] = acc[base + 3] * alpha + w * hi[1];
+            }
+        }
+        m = m_new;
+    }
+
+    #pragma unroll
+    for (int off = 32; off > 0; off >>= 1) {
+        float m2 = __shfl_xor(m, off);
+        float l2 = __shfl_xor(l, off);
+        float m_new = fmaxf(m, m2);
+        float a1 = __expf(m - m_new), a2 = __expf(m2 - m_new);
+        l = l * a1 + l2 * a2;
+        #pragma unroll
+        for (int d = 0; d < D; ++d) {
+            float o2 = __shfl_xor(acc[d], off);
+            acc[d] = acc[d] * a1 + o2 * a2;
+        }
+        m = m_new;
+    }
+
+    float l_safe = fmaxf(l, 1e-38f);
+    if (lane == 0) {
+        float inv = 1.f / l_safe;
+        float* op = p.out + (part_off + (long)bh * p.nq + iq) * D;
+        #pragma unroll
+        for (int d = 0; d < D; ++d) op[d] = acc[d] * inv;
+        p.lse[part_off + (long)bh * p.nq + iq] = __logf(l_safe) + m;
+    }
+}
+
+void launch_decode_partial_fp8(const DecodeParams& p, int head_dim, hipStream_t stream) {
+    int waves = p.b * p.h * p.nq;
+    long target = 1024;
+    long chunks = p.chunks > 0 ? p.chunks
+                 : max(1L, min((long)(p.n / 1024 + 1), target / max(waves / 4, 1)));
+    dim3 grid((waves + 3) / 4, (unsigned)chunks);
+    dim3 block(256);
+    if (head_dim == 64) {
+        hipLaunchKernelGGL(decode_partial_fp8_kernel<64>, grid, block, 0, stream, p);
+    } else if (head_dim == 128) {
+        hipLaunchKernelGGL(decode_partial_fp8_kernel<128>, grid, block, 0, stream, p);
     } else {
         __builtin_trap();
     }

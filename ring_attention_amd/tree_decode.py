@@ -98,6 +98,10 @@ def tree_attn_decode(
         lse = torch.full((b, h, nq, 1), -torch.finfo(torch.float32).max,
                          device=q.device, dtype=torch.float32)
 
+    return _merge_across_ranks(local_out, lse, dtype, eps)
+
+
+def _merge_across_ranks(local_out: Tensor, lse: Tensor, dtype, eps: float) -> Tensor:
     if not is_distributed():
         return local_out.to(dtype)
 
@@ -113,3 +117,32 @@ def tree_attn_decode(
 
     out = num_sum / den_sum.clamp(min=eps)
     return out.to(dtype)
+
+
+@torch.no_grad()
+def tree_attn_decode_fp8(
+    q: Tensor,
+    k8: Tensor, v8: Tensor, ks: Tensor, vs: Tensor,
+    eps: float = 1e-8,
+) -> Tensor:
+    """Tree-attention decode over an FP8-quantized LOCAL KV-cache shard.
+
+    The cache (from ops.fp8.quantize_kv_cache, layouts (b, hk, n, d) e4m3 +
+    (b, hk, n) e8m0 row scales) streams at HALF the bf16 bytes — decode is
+    bandwidth-bound, so the step time roughly halves.  Cross-rank merge is
+    identical to tree_attn_decode (2 collective rounds).  Each rank passes
+    its own pre-quantized shard (there is no in-function sharding: a serving
+    cache lives pre-sharded next to its rank).
+    """
+    from .ops import hip_ext
+    dtype = q.dtype
+    d = q.shape[-1]
+    outs, lses = hip_ext.require().decode_partial_fp8(
+        q.to(torch.bfloat16).contiguous(), k8, v8, ks, vs, d ** -0.5)
+    m = lses.max(dim=0).values
+    w = (lses - m[None]).exp()
+    den = w.sum(dim=0)
+    num = (outs * w).sum(dim=0)
+    local_out = num / den.clamp(min=1e-38)
+    lse = den.log() + m
+    return _merge_across_ranks(local_out, lse, dtype, eps)

@@ -631,3 +631,37 @@ def test_flash_attn_fp8_causal_gqa(causal, groups, d):
     assert rel < 0.09, f"fp8 out mean rel err {rel}"
     assert (o - ref).abs().max().item() < 0.2
     assert (lse.cpu() - ref_lse).abs().max().item() < 0.06
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("h,hk,nq,d", [(4, 4, 1, 64), (8, 2, 2, 64), (4, 4, 1, 128)])
+def test_decode_fp8_cache(h, hk, nq, d):
+    # FP8 KV-cache decode vs the eager fp32 reference on the DEQUANTIZED
+    # cache (isolates kernel error from quantization error), plus a loose
+    # bound vs the unquantized reference
+    from ring_attention_amd.ops.fp8 import quantize_kv_cache
+    from ring_attention_amd.tree_decode import tree_attn_decode_fp8
+    b, n = 2, 2048
+    torch.manual_seed(13)
+    q = torch.randn(b, h, nq, d, device="cuda", dtype=torch.bfloat16)
+    k = torch.randn(b, hk, n, d, device="cuda", dtype=torch.bfloat16)
+    v = torch.randn(b, hk, n, d, device="cuda", dtype=torch.bfloat16)
+    k8, v8, ks, vs = quantize_kv_cache(k, v)
+    out = tree_attn_decode_fp8(q, k8, v8, ks, vs)
+    # dequantized-reference
+    kd = (k8.view(torch.float8_e4m3fn).float()
+          * torch.exp2(ks.float() - 127).unsqueeze(-1)).cpu()
+    vd = (v8.view(torch.float8_e4m3fn).float()
+          * torch.exp2(vs.float() - 127).unsqueeze(-1)).cpu()
+    groups = h // hk
+    kd = kd.repeat(1, groups, 1, 1); vd = vd.repeat(1, groups, 1, 1)
+    sim = torch.einsum("bhid,bhjd->bhij", q.float().cpu(), kd) * d ** -0.5
+    ref = torch.einsum("bhij,bhjd->bhid", sim.softmax(-1), vd)
+    err = (out.float().cpu() - ref).abs().max().item()
+    assert err < 5e-3, f"fp8 decode vs dequant ref err {err}"
+    # vs unquantized full-precision reference: quantization-floor bound
+    kf = k.float().cpu().repeat(1, groups, 1, 1)
+    vf = v.float().cpu().repeat(1, groups, 1, 1)
+    sim2 = torch.einsum("bhid,bhjd->bhij", q.float().cpu(), kf) * d ** -0.5
+    ref2 = torch.einsum("bhij,bhjd->bhid", sim2.softmax(-1), vf)
+    assert (out.float().cpu() - ref2).abs().max().item() < 0.1
