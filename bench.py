@@ -92,14 +92,23 @@ def main():
     torch.manual_seed(1234 + rank)
 
     if args.config == 5:
-        # tree-attention decode: single-query step over world-sharded KV
+        # tree-attention decode: single-query step over world-sharded KV.
+        # --fp8 streams the e4m3-quantized cache (half the HBM bytes)
         from ring_attention_amd.tree_decode import tree_attn_decode
         q5 = torch.randn(b, h, 1, d, device=device, dtype=dtype)
         k5 = torch.randn(b, h, n, d, device=device, dtype=dtype)
         v5 = torch.randn(b, h, n, d, device=device, dtype=dtype)
 
-        def step():
-            tree_attn_decode(q5, k5, v5, shard_kv_seq=False)
+        if getattr(args, "fp8", False) and on_gpu:
+            from ring_attention_amd.ops.fp8 import quantize_kv_cache
+            from ring_attention_amd.tree_decode import tree_attn_decode_fp8
+            c5 = quantize_kv_cache(k5, v5)
+
+            def step():
+                tree_attn_decode_fp8(q5, *c5)
+        else:
+            def step():
+                tree_attn_decode(q5, k5, v5, shard_kv_seq=False)
     elif args.config == 4:
         # zig-zag causal GQA + fused rotary (per-GPU slice of the 1M config)
         from ring_attention_amd.zigzag import zig_zag_attn
