@@ -245,7 +245,7 @@ at::Tensor rotary_apply(at::Tensor x, at::Tensor cos_t, at::Tensor sin_t, double
 }
 
 std::vector<at::Tensor> decode_partial(at::Tensor q, at::Tensor k, at::Tensor v,
-                                       double sm_scale) {
+                                       double sm_scale, int64_t chunks_in) {
     // q (B,HQ,NQ,D); k,v (B,HK,N,D) bf16 — NQ query tokens (speculative /
     // tree heads), HQ % HK == 0 (GQA, tile pairing qh % hk)
     // -> (out fp32 (S,B,HQ,NQ,D), lse fp32 (S,B,HQ,NQ,1)): S kv-chunk
@@ -256,8 +256,13 @@ std::vector<at::Tensor> decode_partial(at::Tensor q, at::Tensor k, at::Tensor v,
     TORCH_CHECK(D == 64 || D == 128, "head dim must be 64 or 128");
     TORCH_CHECK(H % HK == 0, "q heads must be a multiple of kv heads");
     int64_t waves = B * H * NQ;
-    int64_t chunks = std::max<int64_t>(
-        1, std::min<int64_t>(N / 1024 + 1, 1024 / std::max<int64_t>(waves / 4, 1)));
+    // kv-split heuristic (GPU-swept, tools/*decode_sweep*.py): ~512 keys
+    // per wave with total waves capped near 8192 and S capped at 256 (the
+    // host-side partial merge grows with S).  128k/1M bf16: 155/983 ->
+    // 112/607 us; fp8: 109/250 -> 60/247 us; GQA 32q/4kv: 229 -> 177 us.
+    int64_t chunks = chunks_in > 0 ? chunks_in : std::max<int64_t>(
+        1, std::min<int64_t>(std::min<int64_t>(N / 512 + 1, 256),
+                             8192 / std::max<int64_t>(waves, 1)));
     auto out = at::empty({chunks, B, H, NQ, D}, q.options().dtype(at::kFloat));
     auto lse = at::empty({chunks, B, H, NQ, 1}, q.options().dtype(at::kFloat));
     DecodeParams p{};
@@ -311,7 +316,7 @@ std::vector<at::Tensor> attn_fwd_fp8(at::Tensor q8, at::Tensor k8, at::Tensor v8
 
 std::vector<at::Tensor> decode_partial_fp8(at::Tensor q, at::Tensor k8, at::Tensor v8,
                                            at::Tensor ks, at::Tensor vs,
-                                           double sm_scale) {
+                                           double sm_scale, int64_t chunks_in) {
     // FP8 KV-cache decode partial: q bf16 (B,HQ,NQ,D); k8/v8 e4m3 u8
     // (B,HK,N,D) with per-row e8m0 scales ks/vs (B,HK,N).  Same output
     // contract as decode_partial: (out fp32 (S,B,HQ,NQ,D), lse (S,B,HQ,NQ,1)).
@@ -325,8 +330,13 @@ std::vector<at::Tensor> decode_partial_fp8(at::Tensor q, at::Tensor k8, at::Tens
     TORCH_CHECK(H % HK == 0, "q heads must be a multiple of kv heads");
     TORCH_CHECK(ks.numel() == B * HK * N && vs.numel() == B * HK * N);
     int64_t waves = B * H * NQ;
-    int64_t chunks = std::max<int64_t>(
-        1, std::min<int64_t>(N / 1024 + 1, 1024 / std::max<int64_t>(waves / 4, 1)));
+    // kv-split heuristic (GPU-swept, tools/*decode_sweep*.py): ~512 keys
+    // per wave with total waves capped near 8192 and S capped at 256 (the
+    // host-side partial merge grows with S).  128k/1M bf16: 155/983 ->
+    // 112/607 us; fp8: 109/250 -> 60/247 us; GQA 32q/4kv: 229 -> 177 us.
+    int64_t chunks = chunks_in > 0 ? chunks_in : std::max<int64_t>(
+        1, std::min<int64_t>(std::min<int64_t>(N / 512 + 1, 256),
+                             8192 / std::max<int64_t>(waves, 1)));
     auto out = at::empty({chunks, B, H, NQ, D}, q.options().dtype(at::kFloat));
     auto lse = at::empty({chunks, B, H, NQ, 1}, q.options().dtype(at::kFloat));
     DecodeParams p{};
@@ -379,7 +389,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
             py::arg("desc_dq"), py::arg("desc_dkv"),
             py::arg("bias") = std::nullopt, py::arg("bias_mat") = false);
     mod.def("decode_partial", &ring_attn::decode_partial, "CDNA4 single-query decode partial",
-            py::arg("q"), py::arg("k"), py::arg("v"), py::arg("sm_scale") = -1.0);
+            py::arg("q"), py::arg("k"), py::arg("v"), py::arg("sm_scale") = -1.0,
+            py::arg("chunks") = 0);
     mod.def("attn_fwd_fp8", &ring_attn::attn_fwd_fp8,
             "CDNA4 MX-FP8 serving forward (e4m3 + e8m0 row scales)",
             py::arg("q8"), py::arg("k8"), py::arg("v8t"), py::arg("qs"),
@@ -388,7 +399,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
     mod.def("decode_partial_fp8", &ring_attn::decode_partial_fp8,
             "CDNA4 FP8 KV-cache decode partial",
             py::arg("q"), py::arg("k8"), py::arg("v8"), py::arg("ks"),
-            py::arg("vs"), py::arg("sm_scale") = -1.0);
+            py::arg("vs"), py::arg("sm_scale") = -1.0, py::arg("chunks") = 0);
     mod.def("attn_delta", &ring_attn::attn_delta, "fused delta = rowsum(dO*O) preprocess");
     mod.def("rotary_apply", &ring_attn::rotary_apply, "fused rotary embedding (table-driven)");
 }
