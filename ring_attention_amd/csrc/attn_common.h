@@ -150,6 +150,17 @@ struct DecodeParams {
 void launch_decode_partial(const DecodeParams& p, int head_dim, hipStream_t stream);
 void launch_decode_partial_fp8(const DecodeParams& p, int head_dim, hipStream_t stream);
 
+struct DecodeMergeParams {
+    const float* outs;   // (S, rows, D)
+    const float* lses;   // (S, rows)
+    float* out;          // (rows, D)
+    float* lse;          // (rows)
+    long rows;           // B * H * NQ
+    int s;               // chunk count S
+};
+
+void launch_decode_merge(const DecodeMergeParams& p, int head_dim, hipStream_t stream);
+
 struct Fp8FwdParams {
     const void* q;      // e4m3 bytes (B, Nq, H, D)
     const void* k;      // e4m3 bytes (B, Nk, H, D)

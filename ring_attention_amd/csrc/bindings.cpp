@@ -351,6 +351,24 @@ std::vector<at::Tensor> decode_partial_fp8(at::Tensor q, at::Tensor k8, at::Tens
     return {out, lse};
 }
 
+std::vector<at::Tensor> decode_merge(at::Tensor outs, at::Tensor lses) {
+    // fuse the S kv-chunk partial merge: outs (S,B,H,NQ,D) fp32,
+    // lses (S,B,H,NQ,1) fp32 -> (out (B,H,NQ,D) fp32, lse (B,H,NQ,1) fp32)
+    CHECK_F32_CONTIG(outs); CHECK_F32_CONTIG(lses);
+    const int64_t S = outs.size(0), B = outs.size(1), H = outs.size(2),
+                  NQ = outs.size(3), D = outs.size(4);
+    TORCH_CHECK(D == 64 || D == 128, "head dim must be 64 or 128");
+    auto out = at::empty({B, H, NQ, D}, outs.options());
+    auto lse = at::empty({B, H, NQ, 1}, outs.options());
+    DecodeMergeParams p{};
+    p.outs = outs.data_ptr<float>(); p.lses = lses.data_ptr<float>();
+    p.out = out.data_ptr<float>(); p.lse = lse.data_ptr<float>();
+    p.rows = B * H * NQ; p.s = (int)S;
+    launch_decode_merge(p, (int)D, at::hip::getCurrentHIPStream());
+    TORCH_CHECK(hipGetLastError() == hipSuccess, "decode merge launch failed");
+    return {out, lse};
+}
+
 at::Tensor attn_delta(at::Tensor dout, at::Tensor out) {
     // dout, out (B,N,H,D) bf16 -> delta fp32 (B,H,N) = rowsum(dout*out)
     CHECK_BF16_CONTIG(dout); CHECK_BF16_CONTIG(out);
@@ -400,6 +418,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
             "CDNA4 FP8 KV-cache decode partial",
             py::arg("q"), py::arg("k8"), py::arg("v8"), py::arg("ks"),
             py::arg("vs"), py::arg("sm_scale") = -1.0, py::arg("chunks") = 0);
+    mod.def("decode_merge", &ring_attn::decode_merge,
+            "fused kv-chunk partial merge for decode");
     mod.def("attn_delta", &ring_attn::attn_delta, "fused delta = rowsum(dO*O) preprocess");
     mod.def("rotary_apply", &ring_attn::rotary_apply, "fused rotary embedding (table-driven)");
 }

@@ -256,4 +256,81 @@ void launch_decode_partial_fp8(const DecodeParams& p, int head_dim, hipStream_t 
     }
 }
 
+
+// ---------------------------------------------------------------------------
+// Fused kv-chunk merge: combine the S per-chunk partials (out, lse) into one
+// (out, lse) with the logsumexp weights.  The torch expression for this
+// (max + exp + weighted sums over the S axis) costs ~50 us of reduce/
+// elementwise dispatches per decode step at S = 256 — more than the decode
+// kernel itself (30 us).  One wave per (b, h, nq) row; lanes parallel over d.
+// ---------------------------------------------------------------------------
+template <int D>
+__global__ __launch_bounds__(256) void decode_merge_kernel(DecodeMergeParams p) {
+    // one BLOCK per row: the 4 waves split the S chunks (a single wave
+    // serially streaming S*D floats was latency-bound at decode's tiny
+    // row counts), lanes parallel over d; LDS combines the 4 partials
+    constexpr int DR = D / 64;
+    const long row = blockIdx.x;
+    const int wid = threadIdx.x >> 6;
+    const int lane = threadIdx.x & 63;
+
+    __shared__ float lm[4];
+    __shared__ float lden[4];
+    __shared__ float lacc[4][D];
+
+    // global max over all S (each thread strides the whole range)
+    float m = MASK_VALUE_F;
+    for (int sC = (int)threadIdx.x; sC < p.s; sC += 256)
+        m = fmaxf(m, p.lses[(long)sC * p.rows + row]);
+    #pragma unroll
+    for (int off = 32; off > 0; off >>= 1)
+        m = fmaxf(m, __shfl_xor(m, off));
+    if (lane == 0) lm[wid] = m;
+    __syncthreads();
+    m = fmaxf(fmaxf(lm[0], lm[1]), fmaxf(lm[2], lm[3]));
+
+    // wave w accumulates chunks s = w, w+4, ...
+    float den = 0.f;
+    float acc[DR];
+    #pragma unroll
+    for (int r = 0; r < DR; ++r) acc[r] = 0.f;
+    for (int sC = wid; sC < p.s; sC += 4) {
+        float w = __expf(p.lses[(long)sC * p.rows + row] - m);
+        den += w;
+        const float* op = p.outs + ((long)sC * p.rows + row) * D;
+        #pragma unroll
+        for (int r = 0; r < DR; ++r)
+            acc[r] += w * op[r * 64 + lane];
+    }
+    if (lane == 0) lden[wid] = den;
+    #pragma unroll
+    for (int r = 0; r < DR; ++r) lacc[wid][r * 64 + lane] = acc[r];
+    __syncthreads();
+
+    if (wid == 0) {
+        float den_t = lden[0] + lden[1] + lden[2] + lden[3];
+        float den_safe = fmaxf(den_t, 1e-38f);
+        float inv = 1.f / den_safe;
+        float* dst = p.out + row * D;
+        #pragma unroll
+        for (int r = 0; r < DR; ++r) {
+            int d = r * 64 + lane;
+            dst[d] = (lacc[0][d] + lacc[1][d] + lacc[2][d] + lacc[3][d]) * inv;
+        }
+        if (lane == 0) p.lse[row] = __logf(den_safe) + m;
+    }
+}
+
+void launch_decode_merge(const DecodeMergeParams& p, int head_dim, hipStream_t stream) {
+    dim3 grid((unsigned)p.rows);
+    dim3 block(256);
+    if (head_dim == 64) {
+        hipLaunchKernelGGL(decode_merge_kernel<64>, grid, block, 0, stream, p);
+    } else if (head_dim == 128) {
+        hipLaunchKernelGGL(decode_merge_kernel<128>, grid, block, 0, stream, p);
+    } else {
+        __builtin_trap();
+    }
+}
+
 }  // namespace ring_attn

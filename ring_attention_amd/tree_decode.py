@@ -42,15 +42,11 @@ def _local_decode_partial(q: Tensor, k: Tensor, v: Tensor) -> tuple[Tensor, Tens
                 kb.to(torch.bfloat16).contiguous(),
                 vb.to(torch.bfloat16).contiguous(),
                 sm_scale=d ** -0.5)
+            # fused S-chunk merge (the torch reduce expression costs more
+            # dispatches than the decode kernel itself at S = 256)
+            out, lse = hip_ext.require().decode_merge(outs, lses)
             if dv != kd:
-                outs = outs[..., :dv]
-            # merge the S kv-chunk partials (same math as the cross-rank merge)
-            m = lses.max(dim=0).values                       # (b,h,1,1)
-            w = (lses - m[None]).exp()                       # (S,b,h,1,1)
-            den = w.sum(dim=0)
-            num = (outs * w).sum(dim=0)
-            out = num / den.clamp(min=1e-38)
-            lse = den.log() + m
+                out = out[..., :dv]
             return out, lse
     scale = q.shape[-1] ** -0.5
     kf, vf = k.float(), v.float()
@@ -139,10 +135,5 @@ def tree_attn_decode_fp8(
     d = q.shape[-1]
     outs, lses = hip_ext.require().decode_partial_fp8(
         q.to(torch.bfloat16).contiguous(), k8, v8, ks, vs, d ** -0.5)
-    m = lses.max(dim=0).values
-    w = (lses - m[None]).exp()
-    den = w.sum(dim=0)
-    num = (outs * w).sum(dim=0)
-    local_out = num / den.clamp(min=1e-38)
-    lse = den.log() + m
+    local_out, lse = hip_ext.require().decode_merge(outs, lses)
     return _merge_across_ranks(local_out, lse, dtype, eps)
