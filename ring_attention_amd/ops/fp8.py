@@ -96,8 +96,9 @@ def ring_flash_attn_fp8(
     q: Tensor, k: Tensor, v: Tensor,
     sm_scale: float | None = None,
     ring_size: int | None = None,
+    causal: bool = False,
 ) -> tuple[Tensor, Tensor]:
-    """Non-causal RING attention forward in MX-FP8 over sharded KV.
+    """RING attention forward in MX-FP8 over sharded KV.
 
     Each rank quantizes its LOCAL (b, n_shard, hk, d) k/v once; the 8-bit
     shards then rotate the ring — half the bf16 wire bytes per hop (xGMI
@@ -105,8 +106,14 @@ def ring_flash_attn_fp8(
     the transport time too) — and every hop's fused-fp8 partial merges by
     logsumexp (exact online-softmax combine in fp32).
 
-    Returns (out bf16 (b, n, h, d), lse fp32 (b, h, n)).  Causal ring
-    geometry (per-hop diagonals) stays on the bf16 path for now.
+    Causal (rank-ordered shards, rank r owning global rows
+    [r*n_shard, (r+1)*n_shard)): a hop whose source ring-rank is ahead of
+    this rank contributes nothing and skips the kernel entirely; the
+    diagonal hop (source == self) runs the causal fp8 kernel; earlier
+    sources run the full non-causal kernel.  Striped/zig-zag layouts stay
+    on the bf16 ring path.
+
+    Returns (out bf16 (b, n, h, d), lse fp32 (b, h, n)).
     """
     from ..parallel import RingTopology, all_ring_pass, is_distributed
     from ..parallel.ring_pass import null_ring_pass
@@ -114,15 +121,22 @@ def ring_flash_attn_fp8(
     sm = sm_scale if sm_scale is not None else q.shape[-1] ** -0.5
     q8, k8, v8t, qs, ks, vs = quantize_fp8(q, k, v)
 
+    my_rank = 0
     if is_distributed():
         topo = RingTopology(ring_size)
+        my_rank = topo.ring_rank
         hops = all_ring_pass(topo, k8, v8t, ks, vs)
     else:
         hops = null_ring_pass(k8, v8t, ks, vs)
 
     out = lse = None
-    for _info, (k8h, v8th, ksh, vsh) in hops:
-        o_h, l_h = flash_attn_fp8_quantized(q8, k8h, v8th, qs, ksh, vsh, sm)
+    for info, (k8h, v8th, ksh, vsh) in hops:
+        src = info.source_ring_rank
+        if causal and src > my_rank:
+            continue                                  # strictly future rows
+        o_h, l_h = flash_attn_fp8_quantized(
+            q8, k8h, v8th, qs, ksh, vsh, sm,
+            causal=causal and src == my_rank)
         if out is None:
             out, lse = o_h.float(), l_h
         else:

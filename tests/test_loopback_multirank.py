@@ -498,3 +498,32 @@ def test_loopback_ring_fp8():
         assert e / s < 8e-2, f"rank {rank} fp8 ring rel err {e/s}"
         want_lse = ref_lse.chunk(world, dim=2)[rank]
         assert (lse - want_lse).abs().max().item() < 0.1
+
+
+@pytest.mark.gpu
+def test_loopback_ring_fp8_causal():
+    # causal fp8 ring at world 2: future-shard hops skip, the diagonal hop
+    # runs the causal fp8 kernel — vs the single-shard causal fp8 forward
+    from ring_attention_amd.ops.fp8 import flash_attn_fp8, ring_flash_attn_fp8
+    world = 2
+    b, n, h, d = 1, 1024, 4, 64
+    torch.manual_seed(31)
+    q = torch.randn(b, n, h, d, device="cuda", dtype=torch.bfloat16)
+    k = torch.randn(b, n, h, d, device="cuda", dtype=torch.bfloat16)
+    v = torch.randn(b, n, h, d, device="cuda", dtype=torch.bfloat16)
+    ref, ref_lse = flash_attn_fp8(q, k, v, causal=True)
+
+    def run(rank):
+        qs = q.chunk(world, dim=1)[rank]
+        ks = k.chunk(world, dim=1)[rank]
+        vs = v.chunk(world, dim=1)[rank]
+        return ring_flash_attn_fp8(qs, ks, vs, causal=True)
+
+    results = loopback_world(world, run)
+    for rank, (out, lse) in enumerate(results):
+        want = ref.chunk(world, dim=1)[rank]
+        e = (out.float() - want.float()).abs().max().item()
+        s = want.float().abs().max().item() + 1e-6
+        assert e / s < 8e-2, f"rank {rank} fp8 causal ring rel err {e/s}"
+        want_lse = ref_lse.chunk(world, dim=2)[rank]
+        assert (lse - want_lse).abs().max().item() < 0.12
