@@ -35,6 +35,9 @@ def main():
     ap.add_argument("--d-head", type=int, default=64)
     ap.add_argument("--batch", type=int, default=1)
     ap.add_argument("--steps", type=int, default=16, help="tokens to decode")
+    ap.add_argument("--fp8", action="store_true",
+                    help="serve from an e4m3-quantized KV cache (GPU only; "
+                         "half the HBM stream per step)")
     args = ap.parse_args()
 
     if "RANK" in os.environ and not is_distributed():
@@ -61,12 +64,25 @@ def main():
     torch.manual_seed(7)
     q = torch.randn(b, h, 1, d, device=device, dtype=dtype)
 
+    if args.fp8 and device == "cuda":
+        # quantize once at cache-write time; every decode step then streams
+        # 8-bit rows (scales fold into the softmax weights)
+        from ring_attention_amd.ops.fp8 import quantize_kv_cache
+        from ring_attention_amd.tree_decode import tree_attn_decode_fp8
+        cache8 = quantize_kv_cache(k_cache, v_cache)
+
+        def decode_step(q):
+            return tree_attn_decode_fp8(q, *cache8)
+    else:
+        def decode_step(q):
+            return tree_attn_decode(q, k_cache, v_cache, shard_kv_seq=False)
+
     outs = []
     t0 = time.perf_counter()
     for step in range(args.steps):
-        # one decode step over the sharded cache (shard_kv_seq=False: the
-        # cache is ALREADY sharded; each rank contributes its partial)
-        out = tree_attn_decode(q, k_cache, v_cache, shard_kv_seq=False)
+        # one decode step over the sharded cache (each rank contributes its
+        # local partial; the cache is ALREADY sharded)
+        out = decode_step(q)
         outs.append(out)
         # in a real server: out -> lm head -> next token -> append its K/V
         # to ONE rank's shard; here we just feed the output back as q
