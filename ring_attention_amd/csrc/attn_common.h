@@ -171,7 +171,9 @@ struct Fp8FwdParams {
     void* out;          // bf16 (B, Nq, H, D)
     float* lse;         // fp32 (B, H, Nq)
     int b, h, hk;       // GQA: kv heads (qh pairs qh % hk)
-    long nq, nk;
+    long nq, nk;        // PADDED sizes (nq % 256 == 0, nk % 128 == 0) — the
+                        // wrapper pads the quantized buffers; strides use these
+    long nk_true;       // true kv length: keys j >= nk_true are masked out
     int nvs;            // Nk / 64
     float scale;
     int causal;         // standard causal (qpos(i) = i), paired-tile grid

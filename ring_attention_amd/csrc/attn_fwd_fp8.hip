@@ -29,7 +29,11 @@
 // kv head qh % hk (reference tile convention).
 //
 // Scope (asserted in the binding): no mask/bias/window/softclamp, single
-// shot (no ring resume), D in {64, 128}, nq % 256 == 0, nk % 128 == 0.
+// shot (no ring resume), D in {64, 128}.  Ragged lengths are handled by the
+// PYTHON wrapper padding the quantized buffers to (nq%256, nk%128)
+// alignment (zero rows, scale 2^-127 — never NaN bytes) while the kernel
+// masks scores at the TRUE kv length; padded q rows are sliced off by the
+// wrapper.
 
 #include <hip/hip_runtime.h>
 #include <hip/hip_bf16.h>
@@ -194,7 +198,8 @@ void attn_fwd_fp8_kernel(Fp8FwdParams p) {
     for (int t = 0; t < t_hi; ++t) {
         const int par = t & 1;
         const long j0 = (long)t * FP8_KVBLK;
-        const bool full_tile = !P.causal || (j0 + FP8_KVBLK - 1 <= wg_q_min);
+        const bool full_tile = (j0 + FP8_KVBLK <= P.nk_true)
+            && (!P.causal || (j0 + FP8_KVBLK - 1 <= wg_q_min));
 
         __syncthreads();
 
@@ -242,7 +247,8 @@ void attn_fwd_fp8_kernel(Fp8FwdParams p) {
                 for (int r = 0; r < 16; ++r) {
                     long j = j0 + kb * 32 + (r & 3) + 8 * (r >> 2) + 4 * lhi;
                     float x = s[kb][r] * scale2;
-                    if (j > i) x = MASK_VALUE_F;   // qpos(i) = i (v0)
+                    if (j >= P.nk_true) x = MASK_VALUE_F;   // padded keys
+                    if (P.causal && j > i) x = MASK_VALUE_F; // qpos(i) = i
                     s[kb][r] = x;
                     smax = fmaxf(smax, x);
                 }

@@ -278,7 +278,7 @@ std::vector<at::Tensor> decode_partial(at::Tensor q, at::Tensor k, at::Tensor v,
 
 std::vector<at::Tensor> attn_fwd_fp8(at::Tensor q8, at::Tensor k8, at::Tensor v8t,
                                       at::Tensor qs, at::Tensor ks, at::Tensor vs,
-                                      double sm_scale, bool causal) {
+                                      double sm_scale, bool causal, int64_t nk_true) {
     // MX-FP8 serving forward (see attn_fwd_fp8.hip header for scope):
     //   q8 (B,Nq,H,D) u8 e4m3; k8 (B,Nk,H,D) u8; v8t (B,H,D,Nk) u8
     //   qs (B,Nq,H) u8 e8m0(+127); ks (B,Nk,H) u8; vs (B,H,D,Nk/64) u8
@@ -293,7 +293,7 @@ std::vector<at::Tensor> attn_fwd_fp8(at::Tensor q8, at::Tensor k8, at::Tensor v8
     TORCH_CHECK(H % HK == 0, "fp8 path: q heads must be a multiple of kv heads");
     TORCH_CHECK(NQ % 256 == 0, "fp8 path: nq must be a multiple of 256 (v0)");
     TORCH_CHECK(NK % 128 == 0, "fp8 path: nk must be a multiple of 128 (v0)");
-    TORCH_CHECK(!causal || NQ == NK, "fp8 causal: nq == nk (v0)");
+    if (nk_true <= 0) nk_true = NK;
     TORCH_CHECK(v8t.size(1) == HK && v8t.size(3) == NK && v8t.size(2) == D
                 && vs.size(1) == HK && vs.size(3) == NK / 64);
     TORCH_CHECK(qs.numel() == B * NQ * H * (D / 64)
@@ -306,6 +306,7 @@ std::vector<at::Tensor> attn_fwd_fp8(at::Tensor q8, at::Tensor k8, at::Tensor v8
     p.qs = qs.data_ptr(); p.ks = ks.data_ptr(); p.vs = vs.data_ptr();
     p.out = out.data_ptr(); p.lse = lse.data_ptr<float>();
     p.b = (int)B; p.h = (int)H; p.hk = (int)HK; p.nq = NQ; p.nk = NK;
+    p.nk_true = nk_true;
     p.nvs = (int)(NK / 64);
     p.causal = causal ? 1 : 0;
     p.scale = sm_scale > 0 ? (float)sm_scale : (float)(1.0 / std::sqrt((double)D));
@@ -413,7 +414,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
             "CDNA4 MX-FP8 serving forward (e4m3 + e8m0 row scales)",
             py::arg("q8"), py::arg("k8"), py::arg("v8t"), py::arg("qs"),
             py::arg("ks"), py::arg("vs"), py::arg("sm_scale") = -1.0,
-            py::arg("causal") = false);
+            py::arg("causal") = false, py::arg("nk_true") = 0);
     mod.def("decode_partial_fp8", &ring_attn::decode_partial_fp8,
             "CDNA4 FP8 KV-cache decode partial",
             py::arg("q"), py::arg("k8"), py::arg("v8"), py::arg("ks"),

@@ -37,13 +37,25 @@ def _e8m0(amax: Tensor) -> tuple[Tensor, Tensor]:
     return (e + 127.0).to(torch.uint8), torch.exp2(e)
 
 
+def _pad_dim(x: Tensor, dim: int, mult: int) -> Tensor:
+    n = x.shape[dim]
+    pad = (-n) % mult
+    if pad == 0:
+        return x.contiguous()
+    shape = list(x.shape)
+    shape[dim] = pad
+    return torch.cat((x, x.new_zeros(shape)), dim=dim).contiguous()
+
+
 def quantize_fp8(q: Tensor, k: Tensor, v: Tensor):
     """Quantize (b, n, h, d) q/k/v for attn_fwd_fp8.
 
-    Returns (q8, k8, v8t, qs, ks, vs) uint8 views:
-      q8/k8 (b, n, h, d) e4m3; v8t (b, hk, d, n) e4m3 (pre-transposed);
-      qs/ks (b, n, h, d // 64) e8m0 row-chunk scales;
-      vs (b, hk, d, n // 64) e8m0.
+    Returns (q8, k8, v8t, qs, ks, vs) uint8 views, PADDED to the kernel's
+    tile alignment (q rows to 256, kv rows to 128) with zero bytes and
+    scale 2^-127 — zero is a valid e4m3 (never NaN), and the kernel masks
+    scores at the true kv length (nk_true):
+      q8/k8 (b, nqـpad, h, d) e4m3; v8t (b, hk, d, nk_pad) e4m3;
+      qs/ks (b, n_pad, h, d // 64) e8m0; vs (b, hk, d, nk_pad // 64) e8m0.
     """
     qf, kf = q.float(), k.float()
     d = q.shape[-1]
@@ -55,13 +67,19 @@ def quantize_fp8(q: Tensor, k: Tensor, v: Tensor):
     q8 = (qf / qs_s.repeat_interleave(64, dim=-1)).to(torch.float8_e4m3fn)
     k8 = (kf / ks_s.repeat_interleave(64, dim=-1)).to(torch.float8_e4m3fn)
 
-    vt = v.permute(0, 2, 3, 1).float().contiguous()      # (b, h, d, n)
-    b, h, d, n = vt.shape
-    assert n % 64 == 0, "fp8 path: kv length must be a multiple of 64"
-    vs_b, vs_s = _e8m0(vt.view(b, h, d, n // 64, 64).abs().amax(dim=-1))
+    vt = v.permute(0, 2, 3, 1).float().contiguous()      # (b, hk, d, n)
+    b, hk, dv, n = vt.shape
+    n64 = n + ((-n) % 64)
+    vt = _pad_dim(vt, 3, 128)
+    vs_b, vs_s = _e8m0(vt[..., :n64].view(b, hk, dv, n64 // 64, 64)
+                       .abs().amax(dim=-1))
+    vs_b = _pad_dim(vs_b, 3, 2)          # to nk_pad // 64
+    vs_s = _pad_dim(vs_s, 3, 2).clamp(min=2.0 ** -127)
     v8t = (vt / vs_s.repeat_interleave(64, dim=-1)).to(torch.float8_e4m3fn)
-    return (q8.view(torch.uint8), k8.view(torch.uint8),
-            v8t.view(torch.uint8), qs_b.contiguous(), ks_b.contiguous(),
+    return (_pad_dim(q8.view(torch.uint8), 1, 256),
+            _pad_dim(k8.view(torch.uint8), 1, 128),
+            v8t.view(torch.uint8).contiguous(),
+            _pad_dim(qs_b, 1, 256), _pad_dim(ks_b, 1, 128),
             vs_b.contiguous())
 
 
@@ -78,16 +96,18 @@ def flash_attn_fp8(
     to amortize quantization across decode steps.
     """
     q8, k8, v8t, qs, ks, vs = quantize_fp8(q, k, v)
-    return flash_attn_fp8_quantized(
+    out, lse = flash_attn_fp8_quantized(
         q8, k8, v8t, qs, ks, vs,
         sm_scale if sm_scale is not None else q.shape[-1] ** -0.5,
-        causal=causal)
+        causal=causal, nk_true=k.shape[1])
+    nq = q.shape[1]
+    return out[:, :nq], lse[..., :nq]
 
 
 def flash_attn_fp8_quantized(q8, k8, v8t, qs, ks, vs, sm_scale: float,
-                             causal: bool = False):
+                             causal: bool = False, nk_true: int = 0):
     out, lse = hip_ext.require().attn_fwd_fp8(q8, k8, v8t, qs, ks, vs,
-                                              sm_scale, causal)
+                                              sm_scale, causal, nk_true)
     return out, lse
 
 
@@ -136,7 +156,7 @@ def ring_flash_attn_fp8(
             continue                                  # strictly future rows
         o_h, l_h = flash_attn_fp8_quantized(
             q8, k8h, v8th, qs, ksh, vsh, sm,
-            causal=causal and src == my_rank)
+            causal=causal and src == my_rank, nk_true=k.shape[1])
         if out is None:
             out, lse = o_h.float(), l_h
         else:
@@ -148,7 +168,8 @@ def ring_flash_attn_fp8(
             wb4 = (wb / den).permute(0, 2, 1).unsqueeze(-1)
             out = out * wa4 + o_h.float() * wb4
             lse = m + den.log()
-    return out.to(q.dtype), lse
+    nq = q.shape[1]
+    return out[:, :nq].to(q.dtype), lse[..., :nq]
 
 
 def quantize_kv_cache(k: Tensor, v: Tensor):

@@ -665,3 +665,29 @@ def test_decode_fp8_cache(h, hk, nq, d):
     sim2 = torch.einsum("bhid,bhjd->bhij", q.float().cpu(), kf) * d ** -0.5
     ref2 = torch.einsum("bhij,bhjd->bhid", sim2.softmax(-1), vf)
     assert (out.float().cpu() - ref2).abs().max().item() < 0.1
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("n,causal", [(1000, False), (777, True), (300, False)])
+def test_flash_attn_fp8_ragged(n, causal):
+    # ragged (non-tile-aligned) lengths: the wrapper pads the quantized
+    # buffers, the kernel masks scores at the true kv length
+    from ring_attention_amd.ops.fp8 import flash_attn_fp8
+    b, h, d = 1, 2, 64
+    torch.manual_seed(41)
+    q = torch.randn(b, n, h, d, device="cuda", dtype=torch.bfloat16)
+    k = torch.randn(b, n, h, d, device="cuda", dtype=torch.bfloat16)
+    v = torch.randn(b, n, h, d, device="cuda", dtype=torch.bfloat16)
+    out, lse = flash_attn_fp8(q, k, v, causal=causal)
+    assert out.shape == (b, n, h, d) and lse.shape == (b, h, n)
+    qf, kf, vf = q.float().cpu(), k.float().cpu(), v.float().cpu()
+    sim = torch.einsum("bihd,bjhd->bhij", qf, kf) * d ** -0.5
+    if causal:
+        pos = torch.arange(n)
+        sim = sim.masked_fill((pos[None, :] > pos[:, None])[None, None], float("-inf"))
+    ref = torch.einsum("bhij,bjhd->bihd", sim.softmax(-1), vf)
+    ref_lse = sim.logsumexp(dim=-1)
+    o = out.float().cpu()
+    rel = (o - ref).abs().mean().item() / ref.abs().mean().item()
+    assert rel < 0.09, f"fp8 ragged rel err {rel}"
+    assert (lse.cpu() - ref_lse).abs().max().item() < 0.06

@@ -202,15 +202,18 @@ def test_fp8_quantize_roundtrip_cpu():
     k = torch.randn(b, n, hk, d, dtype=torch.bfloat16)
     v = torch.randn(b, n, hk, d, dtype=torch.bfloat16) * 0.1
     q8, k8, v8t, qs, ks, vs = quantize_fp8(q, k, v)
-    assert q8.shape == (b, n, h, d) and qs.shape == (b, n, h, d // 64)
-    assert ks.shape == (b, n, hk, d // 64)
-    assert v8t.shape == (b, hk, d, n) and vs.shape == (b, hk, d, n // 64)
-    qdq = q8.view(torch.float8_e4m3fn).float() * torch.exp2(
-        qs.float() - 127).repeat_interleave(64, dim=-1)
+    # outputs are PADDED to the kernel's tile alignment (q 256, kv 128)
+    nqp, nkp = 256, 128
+    assert q8.shape == (b, nqp, h, d) and qs.shape == (b, nqp, h, d // 64)
+    assert ks.shape == (b, nkp, hk, d // 64)
+    assert v8t.shape == (b, hk, d, nkp) and vs.shape == (b, hk, d, nkp // 64)
+    assert (q8[:, n:] == 0).all() and (v8t[..., n:] == 0).all()
+    qdq = (q8.view(torch.float8_e4m3fn).float() * torch.exp2(
+        qs.float() - 127).repeat_interleave(64, dim=-1))[:, :n]
     rel = (qdq - q.float()).abs().max() / q.float().abs().max()
     assert rel < 0.07, f"q roundtrip rel {rel}"
-    vdq = v8t.view(torch.float8_e4m3fn).float() * torch.exp2(
-        vs.float() - 127).repeat_interleave(64, dim=-1)
+    vdq = (v8t.view(torch.float8_e4m3fn).float() * torch.exp2(
+        vs.float() - 127).repeat_interleave(64, dim=-1))[..., :n]
     vref = v.permute(0, 2, 3, 1).float()
     assert (vdq - vref).abs().max() / vref.abs().max() < 0.07
     # e4m3 range respected: no inf/nan bytes (0x7f/0xff are nan in e4m3fn)
