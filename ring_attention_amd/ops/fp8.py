@@ -17,9 +17,10 @@ bits); use the bf16 path when training.
 Causal (``causal=True``) uses the bf16 kernels' mirrored paired-tile
 load balance; GQA follows the framework-wide ``qh % hk`` pairing.
 
-v0 scope (kernel asserts): no mask/bias/window, d in {64, 128},
-nq % 256 == 0, nk % 128 == 0 (causal: nq == nk), single shard (no ring
-pass).
+Scope: forward only (use the bf16 path for training); no bias/key-pad
+mask/window; d in {64, 128}.  Ragged lengths are handled by padding the
+quantized buffers (zero bytes — a valid e4m3, never NaN) while the kernel
+masks at the true kv length.
 """
 
 from __future__ import annotations
@@ -89,7 +90,7 @@ def flash_attn_fp8(
     sm_scale: float | None = None,
     causal: bool = False,
 ) -> tuple[Tensor, Tensor]:
-    """MX-FP8 non-causal attention forward on (b, n, h, d) tensors.
+    """MX-FP8 attention forward on (b, n, h, d) tensors.
 
     Returns (out bf16 (b, n, h, d), lse fp32 (b, h, n)).  Quantizes
     internally; pass pre-quantized operands via flash_attn_fp8_quantized
