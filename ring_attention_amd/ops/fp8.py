@@ -107,9 +107,49 @@ def flash_attn_fp8(
 
 def flash_attn_fp8_quantized(q8, k8, v8t, qs, ks, vs, sm_scale: float,
                              causal: bool = False, nk_true: int = 0):
-    out, lse = hip_ext.require().attn_fwd_fp8(q8, k8, v8t, qs, ks, vs,
-                                              sm_scale, causal, nk_true)
-    return out, lse
+    if q8.is_cuda:
+        out, lse = hip_ext.require().attn_fwd_fp8(q8, k8, v8t, qs, ks, vs,
+                                                  sm_scale, causal, nk_true)
+        return out, lse
+    return _eager_fp8(q8, k8, v8t, qs, ks, vs, sm_scale, causal, nk_true)
+
+
+def _dequant(x8: Tensor, scales: Tensor) -> Tensor:
+    return (x8.view(torch.float8_e4m3fn).float()
+            * torch.exp2(scales.float() - 127.0).repeat_interleave(64, dim=-1))
+
+
+def _eager_fp8(q8, k8, v8t, qs, ks, vs, sm_scale, causal, nk_true):
+    """CPU fallback: dequantized eager attention with the kernel's masking
+    semantics (true-length + causal).  Approximates the GPU path to within
+    its P-quantization error (the fallback keeps P in fp32) — it exists so
+    the fp8 wrapper/ring logic runs and tests anywhere, like the rest of
+    the framework's CPU fallbacks."""
+    b, nq, h, d = q8.shape
+    hk, nk = k8.shape[2], k8.shape[1]
+    if nk_true <= 0:
+        nk_true = nk
+    qd = _dequant(q8, qs)                                  # (b, nq, h, d)
+    kd = _dequant(k8, ks)                                  # (b, nk, hk, d)
+    # v8t (b, hk, d, nk), vs (b, hk, d, nk//64): dequant along the last dim
+    vd = (v8t.view(torch.float8_e4m3fn).float()
+          * torch.exp2(vs.float() - 127.0).repeat_interleave(64, dim=-1))
+    vd = vd.permute(0, 3, 1, 2)                            # (b, nk, hk, d)
+    groups = h // hk
+    if groups > 1:
+        kd = kd.repeat(1, 1, groups, 1)                    # qh pairs qh % hk
+        vd = vd.repeat(1, 1, groups, 1)
+    sim = torch.einsum("bihd,bjhd->bhij", qd, kd) * sm_scale
+    jpos = torch.arange(nk, device=sim.device)
+    sim = sim.masked_fill((jpos >= nk_true)[None, None, None, :], float("-inf"))
+    if causal:
+        ipos = torch.arange(nq, device=sim.device)
+        sim = sim.masked_fill((jpos[None, :] > ipos[:, None])[None, None],
+                              float("-inf"))
+    lse = sim.logsumexp(dim=-1)                            # (b, h, nq)
+    out = torch.einsum("bhij,bjhd->bihd", torch.softmax(sim, dim=-1), vd)
+    out = torch.nan_to_num(out)                            # all-masked rows
+    return out.to(torch.bfloat16), lse.float()
 
 
 @torch.no_grad()
