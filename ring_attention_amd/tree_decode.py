@@ -130,10 +130,20 @@ def tree_attn_decode_fp8(
     its own pre-quantized shard (there is no in-function sharding: a serving
     cache lives pre-sharded next to its rank).
     """
-    from .ops import hip_ext
     dtype = q.dtype
     d = q.shape[-1]
-    outs, lses = hip_ext.require().decode_partial_fp8(
-        q.to(torch.bfloat16).contiguous(), k8, v8, ks, vs, d ** -0.5)
-    local_out, lse = hip_ext.require().decode_merge(outs, lses)
+    if q.is_cuda:
+        from .ops import hip_ext
+        outs, lses = hip_ext.require().decode_partial_fp8(
+            q.to(torch.bfloat16).contiguous(), k8, v8, ks, vs, d ** -0.5)
+        local_out, lse = hip_ext.require().decode_merge(outs, lses)
+    else:
+        # CPU fallback: dequantize the cache and run the eager partial (the
+        # fp8 serving paths test and run anywhere, like the rest of the
+        # framework's CPU fallbacks)
+        kd = (k8.view(torch.float8_e4m3fn).float()
+              * torch.exp2(ks.float() - 127.0).unsqueeze(-1))
+        vd = (v8.view(torch.float8_e4m3fn).float()
+              * torch.exp2(vs.float() - 127.0).unsqueeze(-1))
+        local_out, lse = _local_decode_partial(q, kd, vd)
     return _merge_across_ranks(local_out, lse, dtype, eps)

@@ -64,3 +64,41 @@ def _ring_case(rank, world, causal):
 def test_fp8_ring_gloo(world, causal):
     # uint8 shard bundles over the real gloo ring transport
     run_distributed(world, _ring_case, causal)
+
+
+def _decode_case(rank, world):
+    from ring_attention_amd.ops.fp8 import quantize_kv_cache
+    from ring_attention_amd.tree_decode import tree_attn_decode, tree_attn_decode_fp8
+    b, h, n, d = 1, 4, 256, 64
+    torch.manual_seed(37)
+    q = torch.randn(b, h, 1, d)
+    k = torch.randn(b, h, n, d)
+    v = torch.randn(b, h, n, d)
+    ref = tree_attn_decode(q, k, v, shard_kv_seq=False) if world == 1 else None
+    ks_ = k.chunk(world, dim=-2)[rank]
+    vs_ = v.chunk(world, dim=-2)[rank]
+    cache = quantize_kv_cache(ks_, vs_)
+    out = tree_attn_decode_fp8(q, *cache)
+    return out
+
+
+def test_fp8_decode_cpu_single():
+    from ring_attention_amd.ops.fp8 import quantize_kv_cache
+    from ring_attention_amd.tree_decode import tree_attn_decode, tree_attn_decode_fp8
+    b, h, n, d = 2, 4, 512, 64
+    torch.manual_seed(37)
+    q = torch.randn(b, h, 1, d)
+    k = torch.randn(b, h, n, d)
+    v = torch.randn(b, h, n, d)
+    ref = tree_attn_decode(q, k, v, shard_kv_seq=False)
+    out = tree_attn_decode_fp8(q, *quantize_kv_cache(k, v))
+    rel = ((out.float() - ref.float()).abs().max()
+           / (ref.float().abs().max() + 1e-6)).item()
+    assert rel < 5e-2, f"fp8 decode cpu rel {rel}"
+
+
+def test_fp8_decode_gloo_w2():
+    # sharded fp8 cache + the 2-round collective merge, CPU fallback partial
+    results = run_distributed(2, _decode_case)
+    a, b_ = results
+    assert (a - b_).abs().max().item() < 1e-6  # all ranks return the merged out
