@@ -49,8 +49,9 @@ static constexpr int FP8_WAVES = 8;
 static constexpr int FP8_QROWS_WAVE = 32;
 static constexpr int FP8_QROWS_WG = FP8_WAVES * FP8_QROWS_WAVE;   // 256
 static constexpr int FP8_NTHREADS = FP8_WAVES * 64;               // 512
-static constexpr int FP8_KVBLK = 128;                             // 256 measured
-static constexpr int FP8_NBLK = FP8_KVBLK / 32;                   // negative (profiles/README.md)
+// KVBLK 256 measured NEGATIVE and wrong (profiles/README.md) — keep 128
+static constexpr int FP8_KVBLK = 128;
+static constexpr int FP8_NBLK = FP8_KVBLK / 32;                   // 4
 
 // 16-byte-chunk XOR swizzle within a row (CH chunks per row), same
 // both-sides rule as the bf16 kernels
