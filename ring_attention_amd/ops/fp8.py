@@ -96,13 +96,24 @@ def flash_attn_fp8(
     internally; pass pre-quantized operands via flash_attn_fp8_quantized
     to amortize quantization across decode steps.
     """
+    d = q.shape[-1]
+    sm = sm_scale if sm_scale is not None else d ** -0.5
+    if d % 64 != 0:
+        # exact zero-pad to the next kernel dim ({64, 128}, like the bf16
+        # path's _pad_head_dim): padded q/k dims contribute 0 to scores,
+        # padded v dims are sliced off below
+        kd = 64 if d < 64 else 128
+        assert d <= 128, "fp8 path: head dim <= 128"
+        import torch.nn.functional as F
+        q = F.pad(q, (0, kd - d))
+        k = F.pad(k, (0, kd - d))
+        v = F.pad(v, (0, kd - d))
     q8, k8, v8t, qs, ks, vs = quantize_fp8(q, k, v)
     out, lse = flash_attn_fp8_quantized(
-        q8, k8, v8t, qs, ks, vs,
-        sm_scale if sm_scale is not None else q.shape[-1] ** -0.5,
+        q8, k8, v8t, qs, ks, vs, sm,
         causal=causal, nk_true=k.shape[1])
     nq = q.shape[1]
-    return out[:, :nq], lse[..., :nq]
+    return out[:, :nq, :, :d], lse[..., :nq]
 
 
 def flash_attn_fp8_quantized(q8, k8, v8t, qs, ks, vs, sm_scale: float,

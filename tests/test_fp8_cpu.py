@@ -102,3 +102,19 @@ def test_fp8_decode_gloo_w2():
     results = run_distributed(2, _decode_case)
     a, b_ = results
     assert (a - b_).abs().max().item() < 1e-6  # all ranks return the merged out
+
+
+@pytest.mark.parametrize("d", [32, 40, 96])
+def test_fp8_head_dim_padding(d):
+    # any head dim <= 128 via exact zero-pad (bf16-path parity)
+    b, n, h = 1, 256, 2
+    torch.manual_seed(43)
+    q = torch.randn(b, n, h, d, dtype=torch.bfloat16)
+    k = torch.randn(b, n, h, d, dtype=torch.bfloat16)
+    v = torch.randn(b, n, h, d, dtype=torch.bfloat16)
+    out, lse = flash_attn_fp8(q, k, v)
+    assert out.shape == (b, n, h, d)
+    sim = torch.einsum("bihd,bjhd->bhij", q.float(), k.float()) * d ** -0.5
+    ref = torch.einsum("bhij,bjhd->bihd", sim.softmax(-1), v.float())
+    rel = ((out.float() - ref).abs().mean() / ref.abs().mean()).item()
+    assert rel < 0.07, f"fp8 d={d} rel {rel}"
