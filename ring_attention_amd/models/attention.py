@@ -65,6 +65,7 @@ class RingAttention(nn.Module):
         softclamp_qk_sim: bool = False,
         softclamp_value: float = 50.0,
         use_hip_kernel: bool | None = None,
+        fp8_inference: bool = False,
     ):
         super().__init__()
         assert heads % num_grouped_query_heads == 0, (
@@ -86,6 +87,11 @@ class RingAttention(nn.Module):
         self.softclamp_qk_sim = softclamp_qk_sim
         self.softclamp_value = softclamp_value
         self.use_hip_kernel = use_hip_kernel  # None = auto (on GPU)
+        # MX-FP8 inference: no-grad forwards route to the e4m3 serving path
+        # (ops/fp8.py) when its scope applies — non-striped, no mask/bias/
+        # softclamp/lookback.  Training steps and out-of-scope calls keep
+        # the bf16 path automatically.
+        self.fp8_inference = fp8_inference
 
         assert not (striped_ring_attn and not causal), "striped ring attention requires causal"
         self.auto_shard_seq = auto_shard_seq if auto_shard_seq is not None else ring_attn
@@ -167,6 +173,17 @@ class RingAttention(nn.Module):
         return out
 
     def _flash(self, q, k, v, mask, ring_reduce: bool, ring_size: int) -> Tensor:
+        if (self.fp8_inference and not torch.is_grad_enabled()
+                and mask is None and not self.striped_ring_attn
+                and not self.softclamp_qk_sim
+                and self.max_lookback_seq_len is None):
+            from ..ops.fp8 import flash_attn_fp8, ring_flash_attn_fp8
+            if ring_reduce:
+                out, _ = ring_flash_attn_fp8(
+                    q, k, v, ring_size=ring_size, causal=self.causal)
+            else:
+                out, _ = flash_attn_fp8(q, k, v, causal=self.causal)
+            return out
         use_hip = self.use_hip_kernel if self.use_hip_kernel is not None else q.is_cuda
         if use_hip and q.is_cuda:
             from ..ops.ring_flash_hip import ring_flash_attn_hip_

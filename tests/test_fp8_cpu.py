@@ -118,3 +118,28 @@ def test_fp8_head_dim_padding(d):
     ref = torch.einsum("bhij,bjhd->bihd", sim.softmax(-1), v.float())
     rel = ((out.float() - ref).abs().mean() / ref.abs().mean()).item()
     assert rel < 0.07, f"fp8 d={d} rel {rel}"
+
+
+def test_ring_attention_module_fp8_inference():
+    # L3 integration: fp8_inference routes no-grad forwards to the e4m3
+    # path (eager fallback on CPU), training steps keep the bf16/fp32 path
+    from ring_attention_amd.models.attention import RingAttention
+    torch.manual_seed(47)
+    attn = RingAttention(dim=128, dim_head=64, heads=2, causal=True,
+                         fp8_inference=True).to(torch.bfloat16)
+    x = torch.randn(1, 256, 128, dtype=torch.bfloat16)
+    with torch.no_grad():
+        out_fp8 = attn(x)
+    attn.fp8_inference = False
+    with torch.no_grad():
+        ref = attn(x)
+    rel = ((out_fp8.float() - ref.float()).abs().mean()
+           / (ref.float().abs().mean() + 1e-9)).item()
+    assert out_fp8.shape == ref.shape
+    assert rel < 0.2, f"fp8 module path rel {rel}"   # quantization floor
+    # with grad enabled the fp8 path must NOT engage (it has no backward)
+    x2 = torch.randn(1, 256, 128, dtype=torch.bfloat16, requires_grad=True)
+    out = attn_train = RingAttention(dim=128, dim_head=64, heads=2, causal=True,
+                                     fp8_inference=True).to(torch.bfloat16)(x2)
+    out.sum().backward()
+    assert x2.grad is not None
