@@ -35,6 +35,7 @@ __all__ = [
     "flash_attn_fp8",
     "ring_flash_attn_fp8",
     "quantize_fp8",
+    "quantize_kv_cache",
     "RingAttention",
     "RingTransformer",
     "RingRotaryEmbedding",
@@ -55,7 +56,8 @@ def __getattr__(name):
                 "flash_attn_offset"):
         from .ops import ring_flash_hip
         return getattr(ring_flash_hip, name)
-    if name in ("flash_attn_fp8", "ring_flash_attn_fp8", "quantize_fp8"):
+    if name in ("flash_attn_fp8", "ring_flash_attn_fp8", "quantize_fp8",
+                "quantize_kv_cache"):
         from .ops import fp8
         return getattr(fp8, name)
     raise AttributeError(f"module {__name__!r} has no attribute {name!r}")
