@@ -25,3 +25,11 @@ echo "== fwd-only 262144 =="
 timeout 300 python bench.py --steps 5 --warmup 2 --fwd-only --seq-per-gpu 262144 2>/dev/null | tail -1
 echo "== causal 131072 =="
 timeout 300 python bench.py --steps 5 --warmup 2 --causal --seq-per-gpu 131072 2>/dev/null | tail -1
+# MX-FP8 rows (beyond-reference serving path)
+run --fp8
+run --fp8 --seq-per-gpu 32768
+run --fp8 --causal
+run --fp8 --causal --seq-per-gpu 32768
+run --fp8 --heads 32 --kv-heads 4 --causal
+run --fp8 --d-head 128
+run --config 5 --fp8
