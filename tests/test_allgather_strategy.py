@@ -72,6 +72,12 @@ def test_allgather_vs_ring_w2_lookback():
     run_distributed(2, _case, True, False, 1, False, 13)
 
 
+def test_allgather_vs_ring_w4_gqa_striped_lookback():
+    # the full composition at world 4 — closest rehearsal of the 8-GPU
+    # bench-adjacent shapes: striped causal + GQA + token-exact lookback
+    run_distributed(4, _case, True, True, 2, False, 24)
+
+
 def _helper_case(rank, world, striped):
     """gather_global_order / scatter_chunks_of_global round-trip on gloo."""
     import torch
