@@ -64,6 +64,7 @@ class RingTransformer(nn.Module):
         ignore_index: int = -1,
         force_regular_attn: bool = False,
         use_hip_kernel: bool | None = None,
+        fp8_inference: bool = False,
     ):
         super().__init__()
         self.ring_attn = ring_attn
@@ -99,6 +100,7 @@ class RingTransformer(nn.Module):
                 force_regular_attn=force_regular_attn,
                 rotary_embed=False,    # shared rotary passed in per forward
                 use_hip_kernel=use_hip_kernel,
+                fp8_inference=fp8_inference,
             )
             ff = FeedForward(dim=dim, mult=ff_mult)
             self.layers.append(nn.ModuleList([attn, ff]))

@@ -143,3 +143,23 @@ def test_ring_attention_module_fp8_inference():
                                      fp8_inference=True).to(torch.bfloat16)(x2)
     out.sum().backward()
     assert x2.grad is not None
+
+
+def test_ring_transformer_fp8_inference():
+    # L4 passthrough: fp8_inference reaches every layer's attention; logits
+    # track the bf16 model closely (attention quantization washes through
+    # the projections)
+    from ring_attention_amd import RingTransformer
+    torch.manual_seed(3)
+    kw = dict(num_tokens=64, dim=128, depth=2, causal=True, dim_head=64,
+              heads=2, bucket_size=256, ring_seq_size=512)
+    m = RingTransformer(**kw).bfloat16()
+    m8 = RingTransformer(**kw, fp8_inference=True).bfloat16()
+    m8.load_state_dict(m.state_dict())
+    ids = torch.randint(0, 64, (1, 256))
+    with torch.no_grad():
+        ref = m(ids)
+        out = m8(ids)
+    rel = ((out.float() - ref.float()).abs().mean()
+           / (ref.float().abs().mean() + 1e-9)).item()
+    assert rel < 0.05, f"transformer fp8 rel {rel}"
