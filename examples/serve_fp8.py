@@ -1,0 +1,87 @@
+"""End-to-end MX-FP8 serving session on one shard: prefill + decode.
+
+Demonstrates the full fp8 story (all beyond the bf16/fp16 reference):
+
+  1. PREFILL: causal `flash_attn_fp8` over the prompt — e4m3 block-scaled
+     MFMA at ~2x the bf16 matrix rate (666-763 TF measured at 8-32k).
+  2. CACHE WRITE: `quantize_kv_cache` — the KV cache lives in e4m3 with one
+     e8m0 scale per row (half the HBM of bf16, quartered vs fp32).
+  3. DECODE LOOP: `tree_attn_decode_fp8` streams the 8-bit cache every
+     token (128k KV: 50 us/step vs 101 us bf16 on MI355X).
+
+Run:  python examples/serve_fp8.py [--prompt-len 4096] [--steps 16]
+On CPU the same calls run the dequantized eager fallbacks, so the example
+works anywhere; on an MI355X they run the HIP kernels.
+
+In a real multi-GPU server the prefill would be `ring_flash_attn_fp8`
+(8-bit shards on the xGMI wire) and the decode merge the 2-round RCCL
+combine — see examples/serve_tree_decode.py --fp8 for the sharded decode.
+"""
+
+from __future__ import annotations
+
+import argparse
+import os
+import sys
+import time
+
+import torch
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+from ring_attention_amd import flash_attn_fp8, quantize_kv_cache, tree_attn_decode_fp8
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--prompt-len", type=int, default=4096)
+    ap.add_argument("--heads", type=int, default=8)
+    ap.add_argument("--kv-heads", type=int, default=None)
+    ap.add_argument("--d-head", type=int, default=64)
+    ap.add_argument("--steps", type=int, default=16, help="tokens to decode")
+    args = ap.parse_args()
+
+    device = "cuda" if torch.cuda.is_available() else "cpu"
+    dtype = torch.bfloat16
+    b, n, h, d = 1, args.prompt_len, args.heads, args.d_head
+    hk = args.kv_heads or h
+
+    # ---- synthetic prompt activations (in service: the model's q/k/v)
+    torch.manual_seed(0)
+    q = torch.randn(b, n, h, d, device=device, dtype=dtype)
+    k = torch.randn(b, n, hk, d, device=device, dtype=dtype)
+    v = torch.randn(b, n, hk, d, device=device, dtype=dtype)
+
+    # ---- 1. causal fp8 prefill
+    t0 = time.perf_counter()
+    ctx, _lse = flash_attn_fp8(q, k, v, causal=True)
+    if device == "cuda":
+        torch.cuda.synchronize()
+    t_prefill = time.perf_counter() - t0
+
+    # ---- 2. cache write: e4m3 + per-row e8m0 (decode layout (b, hk, n, d))
+    cache = quantize_kv_cache(k.permute(0, 2, 1, 3).contiguous(),
+                              v.permute(0, 2, 1, 3).contiguous())
+    cache_bytes = sum(t.numel() for t in cache)
+
+    # ---- 3. decode loop over the 8-bit cache
+    qd = ctx[:, -1:].permute(0, 2, 1, 3).contiguous()     # (b, h, 1, d)
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        out = tree_attn_decode_fp8(qd, *cache)
+        qd = out                       # in service: lm head -> next token
+    if device == "cuda":
+        torch.cuda.synchronize()
+    us_tok = (time.perf_counter() - t0) / args.steps * 1e6
+
+    bf16_bytes = 2 * b * hk * n * d * 2
+    print(f"device {device}  prompt {n}  heads {h}q/{hk}kv  d {d}\n"
+          f"prefill (causal fp8): {t_prefill*1e3:.1f} ms\n"
+          f"kv cache: {cache_bytes/1e6:.1f} MB e4m3 "
+          f"(bf16 would be {bf16_bytes/1e6:.1f} MB)\n"
+          f"decode: {us_tok:.0f} us/token over the 8-bit cache\n"
+          f"out[0,0,0,:4] = {out[0,0,0,:4].float().tolist()}")
+
+
+if __name__ == "__main__":
+    main()
