@@ -25,9 +25,10 @@ from torch import Tensor
 from .parallel import get_rank, get_world_size, is_distributed
 
 
-def _local_decode_partial(q: Tensor, k: Tensor, v: Tensor) -> tuple[Tensor, Tensor]:
+def _local_decode_partial(q: Tensor, k: Tensor, v: Tensor,
+                          use_hip_kernel: bool | None = None) -> tuple[Tensor, Tensor]:
     """Local flash-decode partial: (out fp32 (b,h,nq,dv), lse fp32 (b,h,nq,1))."""
-    if q.is_cuda:
+    if q.is_cuda and use_hip_kernel is not False:
         from .ops import hip_ext
         if hip_ext.available():
             d = q.shape[-1]
@@ -87,7 +88,7 @@ def tree_attn_decode(
         k, v = (ks[rank], vs[rank]) if rank < len(ks) else (None, None)
 
     if v is not None:
-        local_out, lse = _local_decode_partial(q, k, v)
+        local_out, lse = _local_decode_partial(q, k, v, use_hip_kernel)
     else:
         # seq shorter than world: this rank holds nothing
         local_out = q.new_zeros((b, h, nq, dim_v), dtype=torch.float32)
