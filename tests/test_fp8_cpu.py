@@ -163,3 +163,20 @@ def test_ring_transformer_fp8_inference():
     rel = ((out.float() - ref.float()).abs().mean()
            / (ref.float().abs().mean() + 1e-9)).item()
     assert rel < 0.05, f"transformer fp8 rel {rel}"
+
+
+@pytest.mark.parametrize("nq,nk", [(64, 64), (256, 1024), (128, 2000)])
+def test_fp8_cross_length(nq, nk):
+    # nq != nk (chunked prefill / cross-attention shapes) and tiny lengths
+    # far below one tile — the wrapper pads each side independently
+    b, h, d = 1, 2, 64
+    torch.manual_seed(53)
+    q = torch.randn(b, nq, h, d, dtype=torch.bfloat16)
+    k = torch.randn(b, nk, h, d, dtype=torch.bfloat16)
+    v = torch.randn(b, nk, h, d, dtype=torch.bfloat16)
+    out, lse = flash_attn_fp8(q, k, v)
+    assert out.shape == (b, nq, h, d) and lse.shape == (b, h, nq)
+    sim = torch.einsum("bihd,bjhd->bhij", q.float(), k.float()) * d ** -0.5
+    ref = torch.einsum("bhij,bjhd->bihd", sim.softmax(-1), v.float())
+    rel = ((out.float() - ref).abs().mean() / ref.abs().mean()).item()
+    assert rel < 0.07, f"fp8 nq={nq} nk={nk} rel {rel}"
