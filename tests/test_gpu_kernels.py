@@ -691,3 +691,20 @@ def test_flash_attn_fp8_ragged(n, causal):
     rel = (o - ref).abs().mean().item() / ref.abs().mean().item()
     assert rel < 0.09, f"fp8 ragged rel err {rel}"
     assert (lse.cpu() - ref_lse).abs().max().item() < 0.06
+
+
+@pytest.mark.gpu
+def test_flash_attn_fp8_cross_length():
+    # nq != nk on the HIP kernel (chunked-prefill shape)
+    from ring_attention_amd.ops.fp8 import flash_attn_fp8
+    b, nq, nk, h, d = 1, 256, 1024, 2, 64
+    torch.manual_seed(53)
+    q = torch.randn(b, nq, h, d, device="cuda", dtype=torch.bfloat16)
+    k = torch.randn(b, nk, h, d, device="cuda", dtype=torch.bfloat16)
+    v = torch.randn(b, nk, h, d, device="cuda", dtype=torch.bfloat16)
+    out, lse = flash_attn_fp8(q, k, v)
+    assert out.shape == (b, nq, h, d) and lse.shape == (b, h, nq)
+    sim = torch.einsum("bihd,bjhd->bhij", q.float().cpu(), k.float().cpu()) * d ** -0.5
+    ref = torch.einsum("bhij,bjhd->bihd", sim.softmax(-1), v.float().cpu())
+    rel = ((out.float().cpu() - ref).abs().mean() / ref.abs().mean()).item()
+    assert rel < 0.09, f"fp8 cross-length rel {rel}"
