@@ -116,11 +116,13 @@ __global__ __launch_bounds__(256) void decode_partial_kernel(DecodeParams p) {
 
 void launch_decode_partial(const DecodeParams& p, int head_dim, hipStream_t stream) {
     int waves = p.b * p.h * p.nq;
-    // enough chunks to put ~1 wave per CU at this (b*h): each wave streams
-    // its KV slice; host merges the per-chunk partials
-    long target = 1024;
+    // kv-split fallback when the caller did not set p.chunks (the binding
+    // always does): mirror its GPU-swept heuristic — ~512 keys per wave,
+    // total waves capped near 8192, S capped at 256 (the chunk merge grows
+    // with S).  See bindings.cpp decode_partial.
     long chunks = p.chunks > 0 ? p.chunks
-                 : max(1L, min((long)(p.n / 1024 + 1), target / max(waves / 4, 1)));
+                 : max(1L, min(min((long)(p.n / 512 + 1), 256L),
+                               8192L / max((long)waves, 1L)));
     dim3 grid((waves + 3) / 4, (unsigned)chunks);
     dim3 block(256);
     if (head_dim == 64) {
@@ -242,9 +244,10 @@ __global__ __launch_bounds__(256) void decode_partial_fp8_kernel(DecodeParams p)
 
 void launch_decode_partial_fp8(const DecodeParams& p, int head_dim, hipStream_t stream) {
     int waves = p.b * p.h * p.nq;
-    long target = 1024;
+    // same swept kv-split fallback as launch_decode_partial
     long chunks = p.chunks > 0 ? p.chunks
-                 : max(1L, min((long)(p.n / 1024 + 1), target / max(waves / 4, 1)));
+                 : max(1L, min(min((long)(p.n / 512 + 1), 256L),
+                               8192L / max((long)waves, 1L)));
     dim3 grid((waves + 3) / 4, (unsigned)chunks);
     dim3 block(256);
     if (head_dim == 64) {
